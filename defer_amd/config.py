@@ -1,0 +1,56 @@
+"""Pipeline configuration.
+
+The reference hardcodes every constant (ports 5000/5001/5002, chunk size
+512000, queue depths 1000/10 — dispatcher.py:18,24, node.py:111,114,
+test/test.py:39-40). Here the equivalent knobs are one dataclass, so every
+config named in BASELINE.json is expressible.
+"""
+
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+
+@dataclass
+class PipelineConfig:
+    # --- partitioning -----------------------------------------------------
+    # Names of cut layers (the reference's `partition_layers`,
+    # dispatcher.py:107; e.g. ["add_2", "add_4", ...] for ResNet50,
+    # test/test.py:18). Empty list -> single stage. If None and
+    # num_stages > 1, cuts are chosen by the cost-model auto-partitioner.
+    partition_layers: Optional[List[str]] = None
+    num_stages: int = 1
+
+    # --- execution --------------------------------------------------------
+    device: str = "cuda"          # "cuda" (MI355X) or "cpu" (tests/plumbing)
+    dtype: str = "bf16"           # compute dtype on GPU; "fp32" on CPU
+    batch_size: int = 64          # images per pipeline item (micro-batch)
+    use_hip_graphs: bool = True   # capture each stage's forward in a hipGraph
+
+    # --- inter-stage relay (the data plane) -------------------------------
+    # Codec for boundary activations. The reference compresses with
+    # lz4(zfp(x)) (dispatcher.py:81-84). "none" ships raw bf16 over xGMI;
+    # "zfp" ships fixed-rate ZFP blocks; "zfp+lz4" adds the LZ4 stage.
+    compression: str = "none"     # "none" | "zfp" | "zfp+lz4"
+    zfp_rate_bits: int = 8        # fixed-rate bits per value (ZFP)
+
+    # Depth of the per-stage device-resident activation ring buffers (the
+    # reference's Queue(1000) backpressure buffer, node.py:114 — here sized
+    # in micro-batches; 288 GB HBM per GPU means depth is cheap).
+    ring_depth: int = 4
+
+    # --- distributed ------------------------------------------------------
+    backend: str = "nccl"         # "nccl" (RCCL over xGMI) or "gloo" (CPU)
+    # Whether the last stage sends results back to rank 0 (the reference's
+    # node(N-1) -> dispatcher return hop, dispatcher.py:55,103).
+    return_results: bool = True
+
+    # --- observability ----------------------------------------------------
+    log_stage_stats: bool = False  # per-stage imgs/s, bytes relayed, ratios
+
+    extra: dict = field(default_factory=dict)
+
+    def torch_dtype(self):
+        import torch
+
+        return {"bf16": torch.bfloat16, "fp16": torch.float16,
+                "fp32": torch.float32}[self.dtype]

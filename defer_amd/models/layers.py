@@ -1,0 +1,103 @@
+"""Layer modules for defer_amd models.
+
+Each layer is a thin nn.Module over `defer_amd.ops` (HIP kernels on GPU,
+fp32 PyTorch reference on CPU). These are the in-repo replacements for the
+Keras layer zoo the reference executes via model.predict (node.py:106):
+Conv2D/BatchNorm/ReLU -> ConvBNAct (BN folded at init), Add(+ReLU) ->
+AddAct, MaxPooling2D -> MaxPool, GlobalAveragePooling2D -> GlobalAvgPool,
+Dense -> Dense, softmax -> Softmax.
+
+Activations are NHWC; conv weights are [R, S, Cin, Cout] (RSCK), chosen so
+the implicit-GEMM K dimension (r, s, c) is contiguous per (r, s) slice and
+the B-matrix [K_gemm, Cout] is dense row-major for the MFMA kernel.
+"""
+
+import math
+
+import torch
+import torch.nn as nn
+
+from defer_amd import ops
+
+
+class ConvBNAct(nn.Module):
+    """Conv2d + folded inference BatchNorm + optional ReLU, fused.
+
+    BN folding: y = conv(x) * scale + bias with scale = gamma/sqrt(var+eps),
+    bias = beta - mean*scale. Random init keeps BN-ish statistics so
+    activations stay O(1) through deep stacks (bench uses random-init
+    weights per BASELINE.json).
+    """
+
+    def __init__(self, cin, cout, kernel=3, stride=1, padding=None,
+                 act="relu", bn=True):
+        super().__init__()
+        self.cin, self.cout = cin, cout
+        self.kernel, self.stride = kernel, stride
+        self.padding = (kernel // 2) if padding is None else padding
+        self.act = act
+        fan_in = cin * kernel * kernel
+        w = torch.randn(kernel, kernel, cin, cout) * math.sqrt(2.0 / fan_in)
+        self.weight = nn.Parameter(w)
+        self.scale = nn.Parameter(torch.ones(cout)) if bn else None
+        self.bias = nn.Parameter(torch.zeros(cout))
+
+    def forward(self, x, residual=None):
+        return ops.conv2d_bn_act(
+            x, self.weight.to(x.dtype), self.scale, self.bias,
+            stride=self.stride, padding=self.padding, act=self.act,
+            residual=residual)
+
+    def extra_repr(self):
+        return (f"{self.cin}->{self.cout} k{self.kernel} s{self.stride} "
+                f"p{self.padding} act={self.act}")
+
+    def flops_per_pixel(self):
+        return 2 * self.kernel * self.kernel * self.cin * self.cout
+
+
+class AddAct(nn.Module):
+    """Residual add + activation — the reference's `add_N` cut layers
+    (test/test.py:18) with the ReLU that follows them fused in."""
+
+    def __init__(self, act="relu"):
+        super().__init__()
+        self.act = act
+
+    def forward(self, a, b):
+        return ops.add_act(a, b, act=self.act)
+
+
+class MaxPool(nn.Module):
+    def __init__(self, kernel=3, stride=2, padding=1):
+        super().__init__()
+        self.kernel, self.stride, self.padding = kernel, stride, padding
+
+    def forward(self, x):
+        return ops.maxpool2d(x, self.kernel, self.stride, self.padding)
+
+
+class GlobalAvgPool(nn.Module):
+    def forward(self, x):
+        return ops.global_avg_pool(x)
+
+
+class Dense(nn.Module):
+    def __init__(self, cin, cout, bias=True, act="none"):
+        super().__init__()
+        self.cin, self.cout = cin, cout
+        self.act = act
+        self.weight = nn.Parameter(
+            torch.randn(cin, cout) * math.sqrt(1.0 / cin))
+        self.bias = nn.Parameter(torch.zeros(cout)) if bias else None
+
+    def forward(self, x):
+        y = ops.linear(x, self.weight.to(x.dtype), self.bias)
+        if self.act == "relu":
+            y = ops.relu(y)
+        return y
+
+
+class Softmax(nn.Module):
+    def forward(self, x):
+        return ops.softmax(x)

@@ -1,0 +1,368 @@
+"""Pipeline runtime: stage executors, the DEFER orchestrator, and the
+distributed (one process per GPU, RCCL/xGMI) pipeline.
+
+Reference behavior being rebuilt:
+  - `DEFER.run_defer(model, partition_layers, input_stream, output_stream)`
+    partitions the model, distributes stages to compute nodes, feeds inputs
+    and serves results (dispatcher.py:107-115). Here compute nodes are GPUs
+    (threads+streams in-process, or ranks of a torch.distributed job).
+  - Each compute node runs rx -> compute -> tx with bounded queues
+    (node.py:80-108,114). Here: device-resident rings + RCCL p2p on the
+    NCCL side stream, overlapping the compute stream.
+  - Readiness uses real events/barriers, not sleeps (fixes reference bug
+    B4: dispatcher.py:112, node.py:32-33,95-96).
+"""
+
+import queue
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Callable, List, Optional
+
+import torch
+import torch.distributed as dist
+
+from defer_amd.config import PipelineConfig
+from defer_amd.graph import GraphModel, LayerGraph
+from defer_amd.parallel.comm import Codec, P2PRing
+from defer_amd.parallel.partitioner import (as_graph_model, auto_partition,
+                                            partition_model)
+
+
+# --------------------------------------------------------------------------
+# stage execution
+# --------------------------------------------------------------------------
+
+@dataclass
+class StageStats:
+    items: int = 0
+    images: int = 0
+    compute_s: float = 0.0
+    bytes_in: int = 0
+    bytes_out: int = 0
+
+
+class StageExecutor:
+    """Runs one stage model on one device (the reference Node's
+    model.predict hot loop, node.py:103-108, minus the sockets).
+
+    Weights are pre-cast: conv/dense weights to the compute dtype, folded
+    BN scale/bias kept fp32 for the HIP kernel epilogues. Optionally
+    captures the stage forward into a hipGraph (torch.cuda.CUDAGraph) to
+    amortize launch overhead for small batches.
+    """
+
+    def __init__(self, stage: GraphModel, device, dtype: torch.dtype,
+                 use_graph: bool = False):
+        self.model = stage
+        self.device = torch.device(device)
+        self.dtype = dtype
+        self.use_graph = use_graph and self.device.type == "cuda"
+        stage.to(self.device)
+        if self.device.type == "cuda":
+            for p in stage.parameters():
+                if p.dim() >= 2:  # conv RSCK / dense weights
+                    p.data = p.data.to(dtype).contiguous()
+                else:             # BN scale/bias, conv bias: fp32 epilogue
+                    p.data = p.data.float()
+        self._graph = None
+        self._static_in = None
+        self._static_out = None
+
+    def _capture(self, x: torch.Tensor):
+        self._static_in = x.clone()
+        torch.cuda.synchronize()
+        # warm once on a side stream (allocator warm-up) before capture
+        s = torch.cuda.Stream()
+        with torch.cuda.stream(s):
+            self.model(self._static_in)
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._static_out = self.model(self._static_in)
+        self._graph = g
+
+    def run(self, x: torch.Tensor) -> torch.Tensor:
+        if not self.use_graph:
+            with torch.no_grad():
+                return self.model(x)
+        if self._graph is None:
+            self._capture(x)
+        self._static_in.copy_(x)
+        self._graph.replay()
+        return self._static_out
+
+    def output_shape(self, input_shape) -> tuple:
+        was = self.use_graph
+        self.use_graph = False
+        try:
+            with torch.no_grad():
+                x = torch.zeros(*input_shape, device=self.device,
+                                dtype=self.dtype
+                                if self.device.type == "cuda"
+                                else torch.float32)
+                return tuple(self.run(x).shape)
+        finally:
+            self.use_graph = was
+
+
+# --------------------------------------------------------------------------
+# in-process DEFER orchestrator (API parity; threads + queues + streams)
+# --------------------------------------------------------------------------
+
+class DEFER:
+    """API-shape parity with the reference orchestrator
+    (dispatcher.py:21,107): compute nodes are torch devices ("cuda:0",
+    "cuda:1", ..., or "cpu" for the plumbing config); stages are mapped
+    one-to-one onto them and chained by device-to-device copies (xGMI
+    peer copies on GPU). `None` on the input stream shuts the pipeline
+    down cleanly and run_defer returns after draining (the reference runs
+    forever, dispatcher.py:115)."""
+
+    def __init__(self, computeNodes: List, config: Optional[PipelineConfig] = None):
+        self.computeNodes = list(computeNodes)
+        self.cfg = config or PipelineConfig()
+        self.stats: List[StageStats] = []
+
+    def run_defer(self, model, partition_layers: Optional[List[str]],
+                  input_stream: "queue.Queue", output_stream: "queue.Queue"):
+        cfg = self.cfg
+        n = len(self.computeNodes)
+        if partition_layers is None:
+            partition_layers, stages = auto_partition(model, n)
+        else:
+            stages = partition_model(model, partition_layers)
+        if len(stages) != n:
+            raise ValueError(
+                f"{len(stages)} stages for {n} compute nodes")
+        dtype = cfg.torch_dtype()
+        execs = [StageExecutor(s, dev, dtype, cfg.use_hip_graphs)
+                 for s, dev in zip(stages, self.computeNodes)]
+        self.stats = [StageStats() for _ in range(n)]
+        qs = [input_stream] + [queue.Queue(cfg.ring_depth)
+                               for _ in range(n)]
+        ready = [threading.Event() for _ in range(n)]
+
+        def worker(i: int):
+            ex = execs[i]
+            dev = ex.device
+            if dev.type == "cuda":
+                torch.cuda.set_device(dev)
+            ready[i].set()  # real readiness event, not sleep (fixes B4)
+            st = self.stats[i]
+            while True:
+                x = qs[i].get()
+                if x is None:
+                    qs[i + 1].put(None)
+                    return
+                if dev.type == "cuda":
+                    x = x.to(dev, dtype, non_blocking=True)
+                t0 = time.perf_counter()
+                with torch.no_grad():
+                    y = ex.run(x)
+                st.compute_s += time.perf_counter() - t0
+                st.items += 1
+                st.images += x.shape[0] if x.dim() > 1 else 1
+                qs[i + 1].put(y)
+
+        threads = [threading.Thread(target=worker, args=(i,), daemon=True)
+                   for i in range(n)]
+        for t in threads:
+            t.start()
+        for e in ready:
+            e.wait()  # all stages up before feeding (reference ACK 0x06,
+        #             node.py:42, dispatcher.py:64-65)
+
+        # result server: drain final queue into output_stream
+        while True:
+            y = qs[n].get()
+            if y is None:
+                break
+            output_stream.put(y.float().cpu() if y.is_cuda else y)
+        for t in threads:
+            t.join()
+
+
+# --------------------------------------------------------------------------
+# distributed pipeline (one process per GPU over RCCL) — the bench path
+# --------------------------------------------------------------------------
+
+class DistPipeline:
+    """Rank r runs stage r; boundary activations hop r -> r+1 with RCCL
+    send/recv over xGMI (reference: node_i -> node_{i+1}:5000 relay,
+    node.py:108/89); the last rank returns results to rank 0
+    (dispatcher.py:55,103) when cfg.return_results.
+
+    Requires torch.distributed initialized. world_size == num stages.
+    """
+
+    def __init__(self, model, cfg: PipelineConfig, batch_shape,
+                 device: Optional[torch.device] = None):
+        assert dist.is_initialized()
+        self.cfg = cfg
+        self.rank = dist.get_rank()
+        self.world = dist.get_world_size()
+        self.device = device or (
+            torch.device("cuda", self.rank % torch.cuda.device_count())
+            if cfg.device == "cuda" else torch.device("cpu"))
+        self.dtype = (cfg.torch_dtype() if self.device.type == "cuda"
+                      else torch.float32)
+        self.batch_shape = tuple(batch_shape)
+
+        gm = as_graph_model(model)
+        if self.world == 1:
+            cuts, stages = [], [gm]
+        elif cfg.partition_layers is not None:
+            cuts = list(cfg.partition_layers)
+            stages = partition_model(gm, cuts)
+        else:
+            cuts, stages = auto_partition(
+                gm, self.world,
+                input_shape=(1,) + self.batch_shape[1:],
+                bytes_per_elem=self._wire_bytes_per_elem())
+        if len(stages) != self.world:
+            raise ValueError(f"{len(stages)} stages != world {self.world}")
+        self.cuts = cuts
+        self.stage = StageExecutor(stages[self.rank], self.device,
+                                   self.dtype, cfg.use_hip_graphs)
+
+        # boundary shapes: trace every stage's output shape on rank0's
+        # model copy is wasteful; instead each rank traces its own stage
+        # given its input shape, then shapes are chained via all_gather.
+        self.in_shape, self.out_shape = self._boundary_shapes(stages)
+
+        self.recv_ring = None
+        self.send_ring = None
+        self.result_ring = None
+        if self.world > 1:
+            if self.rank > 0:
+                self.in_codec = Codec(cfg, self.in_shape, self.dtype,
+                                      self.device)
+                self.recv_ring = P2PRing(self.in_codec, cfg.ring_depth)
+            if self.rank < self.world - 1:
+                self.out_codec = Codec(cfg, self.out_shape, self.dtype,
+                                       self.device)
+                self.send_ring = P2PRing(self.out_codec, cfg.ring_depth)
+            if cfg.return_results:
+                # logits hop last -> 0, never compressed (tiny)
+                res_cfg = PipelineConfig(compression="none")
+                if self.rank == 0:
+                    final_shape = self._final_shape(stages)
+                    self.res_codec = Codec(res_cfg, final_shape, self.dtype,
+                                           self.device)
+                    self.result_ring = P2PRing(self.res_codec,
+                                               cfg.ring_depth)
+                elif self.rank == self.world - 1:
+                    self.res_codec = Codec(res_cfg, self.out_shape,
+                                           self.dtype, self.device)
+                    self.result_ring = P2PRing(self.res_codec,
+                                               cfg.ring_depth)
+        self.stats = StageStats()
+
+    def _wire_bytes_per_elem(self):
+        if self.cfg.compression.startswith("zfp"):
+            return self.cfg.zfp_rate_bits / 8.0
+        return 2.0 if self.cfg.dtype == "bf16" else 4.0
+
+    def _boundary_shapes(self, stages):
+        """Chain output shapes across stages via a batch-1 CPU shape trace
+        (all ranks compute all; cheap, avoids a collective). Batch dim is
+        then restored — every layer here is batch-pointwise."""
+        B = self.batch_shape[0]
+        shape = (1,) + tuple(self.batch_shape[1:])
+        self._chain_shapes = []
+        in_shape = out_shape = None
+        for i, s in enumerate(stages):
+            x = torch.zeros(*shape)
+            with torch.no_grad():
+                y = s.graph.forward(x)
+            if i == self.rank:
+                in_shape = (B,) + tuple(shape[1:])
+                out_shape = (B,) + tuple(y.shape[1:])
+            shape = tuple(y.shape)
+            self._chain_shapes.append((B,) + tuple(y.shape[1:]))
+        return in_shape, out_shape
+
+    def _final_shape(self, stages):
+        return self._chain_shapes[-1]
+
+    # ------------------------------------------------------------------ run
+    def run(self, steps: int,
+            feed: Optional[Callable[[int], torch.Tensor]] = None,
+            collect: Optional[Callable[[int, torch.Tensor], None]] = None):
+        """Process `steps` items through the pipeline. rank0 calls
+        feed(k) for each item; the final output lands at rank 0 (if
+        return_results) or at the last rank, passed to collect(k, y)."""
+        cfg = self.cfg
+        r, W = self.rank, self.world
+        D = cfg.ring_depth
+
+        if W == 1:
+            for k in range(steps):
+                y = self.stage.run(feed(k))
+                if collect:
+                    collect(k, y)
+            return
+
+        nxt, prv = r + 1, r - 1
+        last = W - 1
+
+        # pre-post recvs (reverse-order ready before senders start)
+        if r > 0:
+            for k in range(min(D, steps)):
+                w = dist.irecv(self.recv_ring.slot(k), src=prv)
+                self.recv_ring.set_work(k, w)
+        if r == 0 and cfg.return_results:
+            for k in range(min(D, steps)):
+                w = dist.irecv(self.result_ring.slot(k), src=last)
+                self.result_ring.set_work(k, w)
+
+        for k in range(steps):
+            # ---- obtain input
+            if r == 0:
+                x = feed(k)
+            else:
+                self.recv_ring.wait(k)
+                wire = self.recv_ring.slot(k)
+                x = self.in_codec.decode(wire)
+            # ---- compute
+            y = self.stage.run(x)
+            # ---- repost recv. Safe to reuse the slot: the irecv is posted
+            # AFTER the compute consuming it was enqueued, and
+            # ProcessGroupNCCL orders the recv after the current stream's
+            # already-enqueued work (CPU/gloo compute is synchronous).
+            if r > 0 and k + D < steps:
+                w = dist.irecv(self.recv_ring.slot(k + D), src=prv)
+                self.recv_ring.set_work(k + D, w)
+            # ---- forward result
+            if r < last:
+                self.send_ring.wait(k)  # slot free (send k-D done)
+                wire = self.out_codec.encode(y, out=self.send_ring.slot(k))
+                w = dist.isend(wire, dst=nxt)
+                self.send_ring.set_work(k, w)
+            elif cfg.return_results:
+                self.result_ring.wait(k)
+                buf = self.result_ring.slot(k)
+                buf.copy_(y.reshape(-1))
+                w = dist.isend(buf, dst=0)
+                self.result_ring.set_work(k, w)
+            elif collect:
+                collect(k, y)
+            # ---- rank0 result collection
+            if r == 0 and cfg.return_results:
+                self.result_ring.wait(k)
+                if collect:
+                    collect(k, self.result_ring.slot(k).view(
+                        self.res_codec.shape))
+                if k + D < steps:
+                    w = dist.irecv(self.result_ring.slot(k + D), src=last)
+                    self.result_ring.set_work(k + D, w)
+            self.stats.items += 1
+
+        # drain outstanding sends
+        for ring in (self.send_ring, self.result_ring):
+            if ring is not None:
+                for w in ring.works:
+                    if w is not None:
+                        w.wait()

@@ -1,0 +1,56 @@
+"""The DEFER orchestrator API (reference shape: dispatcher.py:21,107) on
+CPU compute nodes — BASELINE.json config 1 plumbing."""
+
+import queue
+import threading
+
+import torch
+
+from defer_amd import DEFER, PipelineConfig
+from defer_amd.models import resnet50
+
+
+def _run_defer(model, cuts, nodes, items, cfg=None):
+    eng = DEFER(nodes, config=cfg or PipelineConfig(device="cpu",
+                                                    dtype="fp32"))
+    in_q, out_q = queue.Queue(10), queue.Queue(10)
+    t = threading.Thread(target=eng.run_defer,
+                         args=(model, cuts, in_q, out_q))
+    t.start()
+    for x in items:
+        in_q.put(x)
+    in_q.put(None)
+    outs = [out_q.get(timeout=120) for _ in items]
+    t.join(timeout=120)
+    assert not t.is_alive()
+    return outs, eng
+
+
+def test_defer_two_stage_cpu_matches_whole_model():
+    m = resnet50()
+    xs = [torch.randn(1, 64, 64, 3) for _ in range(3)]
+    with torch.no_grad():
+        want = [m(x) for x in xs]
+    outs, eng = _run_defer(m, ["add_8"], ["cpu", "cpu"], xs)
+    for o, w in zip(outs, want):
+        assert torch.equal(o, w)
+    assert eng.stats[0].items == 3 and eng.stats[1].items == 3
+
+
+def test_defer_auto_partition_four_nodes():
+    m = resnet50()
+    xs = [torch.randn(1, 64, 64, 3) for _ in range(2)]
+    with torch.no_grad():
+        want = [m(x) for x in xs]
+    outs, _ = _run_defer(m, None, ["cpu"] * 4, xs)
+    for o, w in zip(outs, want):
+        assert torch.equal(o, w)
+
+
+def test_defer_single_node():
+    m = resnet50()
+    xs = [torch.randn(1, 64, 64, 3)]
+    with torch.no_grad():
+        want = m(xs[0])
+    outs, _ = _run_defer(m, [], ["cpu"], xs)
+    assert torch.equal(outs[0], want)

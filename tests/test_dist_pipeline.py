@@ -1,0 +1,81 @@
+"""Distributed pipeline over gloo (CPU, world_size 2/3) — the same code
+path bench.py drives with RCCL on 8x MI355X (SURVEY.md §2.3 relay table).
+"""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from defer_amd.config import PipelineConfig
+from defer_amd.models import resnet50
+
+
+def _worker(rank, world, port, q, cuts, steps, compression):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        torch.manual_seed(0)
+        from defer_amd.parallel.pipeline import DistPipeline
+
+        model = resnet50()
+        cfg = PipelineConfig(device="cpu", dtype="fp32",
+                             partition_layers=cuts, ring_depth=2,
+                             compression=compression, backend="gloo")
+        B = 1
+        pipe = DistPipeline(model, cfg, (B, 64, 64, 3))
+
+        torch.manual_seed(1234)
+        inputs = [torch.randn(B, 64, 64, 3) for _ in range(steps)]
+        results = {}
+
+        pipe.run(steps, feed=lambda k: inputs[k],
+                 collect=lambda k, y: results.__setitem__(k, y.clone()))
+
+        if rank == 0:
+            with torch.no_grad():
+                want = [model(x) for x in inputs]
+            for k in range(steps):
+                assert results[k].shape == want[k].shape
+                err = (results[k] - want[k]).abs().max().item()
+                q.put(("err", k, err))
+            q.put(("done", rank, None))
+    finally:
+        dist.destroy_process_group()
+
+
+def _run(world, cuts, steps=4, compression="none", tol=0.0):
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = 29501 + world + (7 if compression != "none" else 0)
+    procs = [ctx.Process(target=_worker,
+                         args=(r, world, port, q, cuts, steps, compression))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    msgs = []
+    for p in procs:
+        p.join(timeout=300)
+        assert p.exitcode == 0, f"worker failed: {p.exitcode}"
+    while not q.empty():
+        msgs.append(q.get())
+    errs = [m[2] for m in msgs if m[0] == "err"]
+    assert len(errs) == steps
+    for e in errs:
+        assert e <= tol, f"pipeline output mismatch: {e}"
+
+
+def test_dist_pipeline_two_stage():
+    _run(2, ["add_8"])
+
+
+def test_dist_pipeline_three_stage_unbalanced():
+    _run(3, ["add_4", "add_12"], steps=5)
+
+
+@pytest.mark.slow
+def test_dist_pipeline_auto_cuts():
+    _run(2, None)

@@ -1,0 +1,71 @@
+"""Model zoo: ResNet50 / VGG19 shapes and partition parity (the
+reference's 8-stage cut list, test/test.py:18)."""
+
+import torch
+
+from defer_amd.models import DEFER_8STAGE_CUTS, resnet50, vgg19
+from defer_amd.parallel.partitioner import (auto_partition, node_costs,
+                                            partition_model)
+
+
+def test_resnet50_shapes():
+    m = resnet50()
+    x = torch.randn(2, 64, 64, 3)  # fully-conv until GAP: small input ok
+    with torch.no_grad():
+        y = m(x)
+    assert y.shape == (2, 1000)
+    assert torch.allclose(y.float().sum(dim=-1), torch.ones(2), atol=1e-4)
+
+
+def test_resnet50_defer_cut_list_is_valid():
+    m = resnet50()
+    cuts = set(m.graph.valid_cut_points())
+    for c in DEFER_8STAGE_CUTS:
+        assert c in cuts
+    stages = partition_model(m, DEFER_8STAGE_CUTS)
+    assert len(stages) == 8
+
+
+def test_resnet50_partition_equivalence():
+    m = resnet50()
+    x = torch.randn(1, 64, 64, 3)
+    with torch.no_grad():
+        want = m(x)
+        z = x
+        for s in partition_model(m, DEFER_8STAGE_CUTS):
+            z = s(z)
+    assert torch.equal(z, want)  # identical op order -> bitwise
+
+
+def test_vgg19_shapes_and_partition():
+    m = vgg19(num_classes=10)
+    x = torch.randn(1, 64, 64, 3)
+    with torch.no_grad():
+        want = m(x)
+    assert want.shape == (1, 10)
+    cuts, stages = auto_partition(m, 4, input_shape=(1, 64, 64, 3))
+    assert len(stages) == 4
+    with torch.no_grad():
+        z = x
+        for s in stages:
+            z = s(z)
+    assert torch.equal(z, want)
+
+
+def test_auto_partition_resnet_8():
+    m = resnet50()
+    cuts, stages = auto_partition(m, 8)
+    assert len(cuts) == 7 and len(stages) == 8
+    # bottleneck sanity: no stage is empty of compute
+    fl, _ = node_costs(m.graph)
+    for s in stages:
+        assert sum(fl.get(n.name, 0.0) for n in s.graph.nodes) > 0
+
+
+def test_node_costs_total_flops():
+    m = resnet50()
+    fl, ob = node_costs(m.graph)
+    total = sum(fl.values())
+    # ResNet50 @224 is ~8.2 GFLOP (2*MAC) per image
+    assert 7.5e9 < total < 9.0e9
+    assert ob["add_1"] == 56 * 56 * 256 * 2.0
