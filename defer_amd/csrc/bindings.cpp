@@ -1,0 +1,205 @@
+// Torch bindings for the defer_amd gfx950 kernel library.
+//
+// Host-only TU: checks device/dtype/contiguity and calls the launchers in
+// the .hip TUs on the current HIP stream, so ops compose with torch.cuda
+// streams and hipGraph capture.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "kernels.h"
+
+namespace {
+
+using at::Tensor;
+using defer_hip::ConvParams;
+
+hipStream_t cur_stream() {
+    return at::cuda::getCurrentHIPStream().stream();
+}
+
+const void* bptr(const Tensor& t) { return t.data_ptr(); }
+void* bptr_mut(Tensor& t) { return t.data_ptr(); }
+
+void check_bf16(const Tensor& t, const char* name) {
+    TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+    TORCH_CHECK(t.scalar_type() == at::kBFloat16, name, " must be bf16");
+    TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// 16B zero source for OOB global_load_lds lanes, one per process
+const void* zero_buf() {
+    static Tensor z = at::zeros(
+        {16}, at::TensorOptions().dtype(at::kBFloat16).device(at::kCUDA));
+    return bptr(z);
+}
+
+const float* fptr_opt(const c10::optional<Tensor>& t, const char* name) {
+    if (!t) return nullptr;
+    TORCH_CHECK(t->scalar_type() == at::kFloat, name, " must be fp32");
+    TORCH_CHECK(t->is_contiguous(), name, " must be contiguous");
+    return t->data_ptr<float>();
+}
+
+Tensor conv2d_bn_act(Tensor x, Tensor w, c10::optional<Tensor> scale,
+                     c10::optional<Tensor> bias, c10::optional<Tensor> res,
+                     int64_t stride, int64_t pad, bool relu) {
+    check_bf16(x, "x");
+    check_bf16(w, "w");
+    TORCH_CHECK(x.dim() == 4, "x must be NHWC");
+    TORCH_CHECK(w.dim() == 4, "w must be OHWI");
+    int NB = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
+    int Cout = w.size(0), R = w.size(1), S = w.size(2);
+    TORCH_CHECK(w.size(3) == Cin, "w Cin mismatch");
+    int OH = (int)((H + 2 * pad - R) / stride + 1);
+    int OW = (int)((W + 2 * pad - S) / stride + 1);
+    long M = (long)NB * OH * OW;
+    TORCH_CHECK(M < (1LL << 31), "M too large");
+    auto out = at::empty({NB, OH, OW, Cout}, x.options());
+    if (res) check_bf16(*res, "res");
+    hipStream_t s = cur_stream();
+
+    ConvParams p{};
+    p.scale = fptr_opt(scale, "scale");
+    p.bias = fptr_opt(bias, "bias");
+    p.res = res ? bptr(*res) : nullptr;
+    p.out = bptr_mut(out);
+    p.zbuf = zero_buf();
+    p.M = (int)M;
+    p.Cout = Cout;
+
+    if (Cin % 8 != 0) {
+        // stem path: explicit im2col + GEMM (Cin too thin to gather 16B)
+        int K = R * S * Cin;
+        int Kpad = (K + 63) / 64 * 64;
+        auto a = at::empty({M, Kpad}, x.options());
+        defer_hip::launch_im2col(bptr(x), bptr_mut(a), NB, H, W, Cin, OH,
+                                 OW, R, S, (int)stride, (int)pad, Kpad, s);
+        auto wp = at::empty({Cout, Kpad}, w.options());
+        defer_hip::launch_padk(bptr(w), bptr_mut(wp), Cout, K, Kpad, s);
+        p.x = bptr(a);
+        p.w = bptr(wp);
+        p.K = Kpad;
+        p.NB = (int)M; p.H = 1; p.W = 1; p.Cin = Kpad;
+        p.OH = 1; p.OW = 1; p.R = 1; p.S = 1; p.stride = 1; p.pad = 0;
+        defer_hip::launch_conv_igemm(p, relu, (bool)res, true, s);
+        return out;
+    }
+
+    p.x = bptr(x);
+    p.w = bptr(w);
+    p.K = R * S * Cin;
+    p.NB = NB; p.H = H; p.W = W; p.Cin = Cin;
+    p.OH = OH; p.OW = OW; p.R = R; p.S = S;
+    p.stride = (int)stride; p.pad = (int)pad;
+    bool gemm_mode = (R == 1 && S == 1 && stride == 1 && pad == 0);
+    if (gemm_mode) {
+        p.NB = (int)M; p.H = 1; p.W = 1; p.Cin = p.K;
+        p.OH = 1; p.OW = 1;
+    }
+    defer_hip::launch_conv_igemm(p, relu, (bool)res, gemm_mode, s);
+    return out;
+}
+
+Tensor linear(Tensor x, Tensor w, c10::optional<Tensor> bias) {
+    check_bf16(x, "x");
+    check_bf16(w, "w");
+    TORCH_CHECK(x.dim() == 2 && w.dim() == 2, "linear wants 2-D");
+    int M = x.size(0), K = x.size(1), N = w.size(0);
+    TORCH_CHECK(w.size(1) == K, "K mismatch");
+    TORCH_CHECK(K % 8 == 0, "K must be a multiple of 8");
+    auto out = at::empty({M, N}, x.options());
+    c10::optional<Tensor> bias_f;
+    if (bias) bias_f = bias->to(at::kFloat).contiguous();
+    ConvParams p{};
+    p.x = bptr(x); p.w = bptr(w);
+    p.scale = nullptr;
+    p.bias = fptr_opt(bias_f, "bias");
+    p.res = nullptr; p.out = bptr_mut(out); p.zbuf = zero_buf();
+    p.M = M; p.K = K; p.Cout = N;
+    p.NB = M; p.H = 1; p.W = 1; p.Cin = K;
+    p.OH = 1; p.OW = 1; p.R = 1; p.S = 1; p.stride = 1; p.pad = 0;
+    defer_hip::launch_conv_igemm(p, false, false, true, cur_stream());
+    return out;
+}
+
+Tensor bn_act(Tensor x, Tensor scale, Tensor bias, bool relu) {
+    check_bf16(x, "x");
+    int C = x.size(-1);
+    TORCH_CHECK(C % 8 == 0, "C must be a multiple of 8");
+    auto out = at::empty_like(x);
+    auto sf = scale.to(at::kFloat).contiguous();
+    auto bf = bias.to(at::kFloat).contiguous();
+    defer_hip::launch_bn_act(bptr(x), sf.data_ptr<float>(),
+                             bf.data_ptr<float>(), bptr_mut(out),
+                             x.numel() / 8, C / 8, relu, cur_stream());
+    return out;
+}
+
+Tensor add_act(Tensor a, Tensor b, bool relu) {
+    check_bf16(a, "a");
+    check_bf16(b, "b");
+    TORCH_CHECK(a.sizes() == b.sizes(), "shape mismatch");
+    TORCH_CHECK(a.numel() % 8 == 0, "numel must be a multiple of 8");
+    auto out = at::empty_like(a);
+    defer_hip::launch_add_act(bptr(a), bptr(b), bptr_mut(out),
+                              a.numel() / 8, relu, cur_stream());
+    return out;
+}
+
+Tensor relu(Tensor x) {
+    check_bf16(x, "x");
+    TORCH_CHECK(x.numel() % 8 == 0, "numel must be a multiple of 8");
+    auto out = at::empty_like(x);
+    defer_hip::launch_relu(bptr(x), bptr_mut(out), x.numel() / 8,
+                           cur_stream());
+    return out;
+}
+
+Tensor softmax(Tensor x) {
+    check_bf16(x, "x");
+    int cols = x.size(-1);
+    long rows = x.numel() / cols;
+    auto out = at::empty_like(x);
+    defer_hip::launch_softmax(bptr(x), bptr_mut(out), (int)rows, cols,
+                              cur_stream());
+    return out;
+}
+
+Tensor maxpool2d(Tensor x, int64_t kernel, int64_t stride, int64_t pad) {
+    check_bf16(x, "x");
+    TORCH_CHECK(x.dim() == 4, "x must be NHWC");
+    int NB = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+    TORCH_CHECK(C % 8 == 0, "C must be a multiple of 8");
+    int OH = (int)((H + 2 * pad - kernel) / stride + 1);
+    int OW = (int)((W + 2 * pad - kernel) / stride + 1);
+    auto out = at::empty({NB, OH, OW, C}, x.options());
+    defer_hip::launch_maxpool(bptr(x), bptr_mut(out), NB, H, W, C, OH, OW,
+                              (int)kernel, (int)stride, (int)pad,
+                              cur_stream());
+    return out;
+}
+
+Tensor global_avg_pool(Tensor x) {
+    check_bf16(x, "x");
+    TORCH_CHECK(x.dim() == 4, "x must be NHWC");
+    int NB = x.size(0), HW = x.size(1) * x.size(2), C = x.size(3);
+    TORCH_CHECK(C % 8 == 0, "C must be a multiple of 8");
+    auto out = at::empty({NB, C}, x.options());
+    defer_hip::launch_gap(bptr(x), bptr_mut(out), NB, HW, C, cur_stream());
+    return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("conv2d_bn_act", &conv2d_bn_act, py::arg("x"), py::arg("w"),
+          py::arg("scale"), py::arg("bias"), py::arg("res"),
+          py::arg("stride"), py::arg("pad"), py::arg("relu"));
+    m.def("linear", &linear);
+    m.def("bn_act", &bn_act);
+    m.def("add_act", &add_act);
+    m.def("relu", &relu);
+    m.def("softmax", &softmax);
+    m.def("maxpool2d", &maxpool2d);
+    m.def("global_avg_pool", &global_avg_pool);
+}

@@ -1,0 +1,44 @@
+// Host-visible launcher API for the defer_amd gfx950 kernel library.
+// Implemented in the .hip TUs; called from the torch bindings.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+namespace defer_hip {
+
+// bf16 pointers are passed as void* across the host boundary.
+struct ConvParams {
+    const void* x;      // [NB, H, W, Cin] bf16 (or [M][K] in gemm mode)
+    const void* w;      // [Cout][K] bf16 (OHWI flattened)
+    const float* scale; // [Cout] or null
+    const float* bias;  // [Cout] or null
+    const void* res;    // [M][Cout] bf16 or null
+    void* out;          // [M][Cout] bf16
+    const void* zbuf;   // >=16 B zeros
+    int M, K, Cout;
+    int NB, H, W, Cin;
+    int OH, OW, R, S, stride, pad;
+};
+
+void launch_conv_igemm(const ConvParams& p, bool relu, bool has_res,
+                       bool gemm_mode, hipStream_t s);
+void launch_im2col(const void* x, void* out, int NB, int H, int W, int Cin,
+                   int OH, int OW, int R, int S, int stride, int pad,
+                   int Kpad, hipStream_t s);
+void launch_padk(const void* w, void* out, int Cout, int K, int Kpad,
+                 hipStream_t s);
+
+void launch_bn_act(const void* x, const float* scale, const float* bias,
+                   void* y, long total8, int c8, bool relu, hipStream_t s);
+void launch_add_act(const void* a, const void* b, void* y, long total8,
+                    bool relu, hipStream_t s);
+void launch_relu(const void* x, void* y, long total8, hipStream_t s);
+void launch_softmax(const void* x, void* y, int rows, int cols,
+                    hipStream_t s);
+void launch_maxpool(const void* x, void* y, int NB, int H, int W, int C,
+                    int OH, int OW, int k, int stride, int pad,
+                    hipStream_t s);
+void launch_gap(const void* x, void* y, int NB, int HW, int C,
+                hipStream_t s);
+
+}  // namespace defer_hip

@@ -7,9 +7,10 @@ Conv2D/BatchNorm/ReLU -> ConvBNAct (BN folded at init), Add(+ReLU) ->
 AddAct, MaxPooling2D -> MaxPool, GlobalAveragePooling2D -> GlobalAvgPool,
 Dense -> Dense, softmax -> Softmax.
 
-Activations are NHWC; conv weights are [R, S, Cin, Cout] (RSCK), chosen so
-the implicit-GEMM K dimension (r, s, c) is contiguous per (r, s) slice and
-the B-matrix [K_gemm, Cout] is dense row-major for the MFMA kernel.
+Activations are NHWC; conv weights are [Cout, R, S, Cin] (OHWI), chosen so
+the implicit-GEMM K dimension (r, s, c) is contiguous per output channel:
+the MFMA kernel's B^T fragment read is then a dense row read, and the
+global->LDS staging of a [BN][BK] weight tile is fully coalesced.
 """
 
 import math
@@ -37,7 +38,7 @@ class ConvBNAct(nn.Module):
         self.padding = (kernel // 2) if padding is None else padding
         self.act = act
         fan_in = cin * kernel * kernel
-        w = torch.randn(kernel, kernel, cin, cout) * math.sqrt(2.0 / fan_in)
+        w = torch.randn(cout, kernel, kernel, cin) * math.sqrt(2.0 / fan_in)
         self.weight = nn.Parameter(w)
         self.scale = nn.Parameter(torch.ones(cout)) if bn else None
         self.bias = nn.Parameter(torch.zeros(cout))
@@ -88,7 +89,7 @@ class Dense(nn.Module):
         self.cin, self.cout = cin, cout
         self.act = act
         self.weight = nn.Parameter(
-            torch.randn(cin, cout) * math.sqrt(1.0 / cin))
+            torch.randn(cout, cin) * math.sqrt(1.0 / cin))
         self.bias = nn.Parameter(torch.zeros(cout)) if bias else None
 
     def forward(self, x):

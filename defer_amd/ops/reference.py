@@ -19,7 +19,7 @@ import torch.nn.functional as F
 
 def conv2d_bn_act(
     x: torch.Tensor,            # [N, H, W, C]   activation
-    w: torch.Tensor,            # [R, S, C, K]   weights
+    w: torch.Tensor,            # [K, R, S, C]   weights (OHWI)
     scale: Optional[torch.Tensor],  # [K] folded BN scale (None -> 1)
     bias: Optional[torch.Tensor],   # [K] folded BN shift / conv bias
     stride: int = 1,
@@ -32,7 +32,7 @@ def conv2d_bn_act(
     Mirrors what the reference gets from Keras Conv2D + BatchNormalization +
     Add + ReLU layers executed inside model.predict (node.py:106)."""
     xf = x.permute(0, 3, 1, 2).float()                 # NHWC -> NCHW
-    wf = w.permute(3, 2, 0, 1).float()                 # RSCK -> KCRS
+    wf = w.permute(0, 3, 1, 2).float()                 # OHWI -> OIHW
     y = F.conv2d(xf, wf, bias=None, stride=stride, padding=padding)
     if scale is not None:
         y = y * scale.float().view(1, -1, 1, 1)
@@ -82,8 +82,8 @@ def global_avg_pool(x):
 
 
 def linear(x, w, bias=None):
-    """[N, Cin] @ [Cin, Cout] + bias (classifier head)."""
-    y = x.float() @ w.float()
+    """[N, Cin] @ [Cout, Cin]^T + bias (classifier head)."""
+    y = x.float() @ w.float().t()
     if bias is not None:
         y = y + bias.float()
     return y.to(x.dtype)
