@@ -224,13 +224,10 @@ class DistPipeline:
         if len(stages) != self.world:
             raise ValueError(f"{len(stages)} stages != world {self.world}")
         self.cuts = cuts
+        # shape-trace on CPU BEFORE weights move to the GPU
+        self.in_shape, self.out_shape = self._boundary_shapes(stages)
         self.stage = StageExecutor(stages[self.rank], self.device,
                                    self.dtype, cfg.use_hip_graphs)
-
-        # boundary shapes: trace every stage's output shape on rank0's
-        # model copy is wasteful; instead each rank traces its own stage
-        # given its input shape, then shapes are chained via all_gather.
-        self.in_shape, self.out_shape = self._boundary_shapes(stages)
 
         self.recv_ring = None
         self.send_ring = None
