@@ -68,20 +68,23 @@ Tensor conv2d_bn_act(Tensor x, Tensor w, c10::optional<Tensor> scale,
     p.Cout = Cout;
 
     if (Cin % 8 != 0) {
-        // stem path: explicit im2col + GEMM (Cin too thin to gather 16B)
-        int K = R * S * Cin;
-        int Kpad = (K + 63) / 64 * 64;
-        auto a = at::empty({M, Kpad}, x.options());
-        defer_hip::launch_im2col(bptr(x), bptr_mut(a), NB, H, W, Cin, OH,
-                                 OW, R, S, (int)stride, (int)pad, Kpad, s);
-        auto wp = at::empty({Cout, Kpad}, w.options());
-        defer_hip::launch_padk(bptr(w), bptr_mut(wp), Cout, K, Kpad, s);
-        p.x = bptr(a);
+        // stem path (Cin=3): zero-pad channels to 8 and take the regular
+        // implicit-GEMM path (K grows R*S*Cin -> R*S*8; the stem is ~2% of
+        // ResNet50 FLOPs, and padding beats an explicit im2col by ~4x).
+        int C8 = (Cin + 7) / 8 * 8;
+        auto xp = at::empty({NB, H, W, C8}, x.options());
+        defer_hip::launch_pad_channels(bptr(x), bptr_mut(xp),
+                                       (long)NB * H * W, Cin, C8, s);
+        auto wp = at::empty({Cout, R, S, C8}, w.options());
+        defer_hip::launch_pad_channels(bptr(w), bptr_mut(wp),
+                                       (long)Cout * R * S, Cin, C8, s);
+        p.x = bptr(xp);
         p.w = bptr(wp);
-        p.K = Kpad;
-        p.NB = (int)M; p.H = 1; p.W = 1; p.Cin = Kpad;
-        p.OH = 1; p.OW = 1; p.R = 1; p.S = 1; p.stride = 1; p.pad = 0;
-        defer_hip::launch_conv_igemm(p, relu, (bool)res, true, s);
+        p.K = R * S * C8;
+        p.NB = NB; p.H = H; p.W = W; p.Cin = C8;
+        p.OH = OH; p.OW = OW; p.R = R; p.S = S;
+        p.stride = (int)stride; p.pad = (int)pad;
+        defer_hip::launch_conv_igemm(p, relu, (bool)res, false, s);
         return out;
     }
 

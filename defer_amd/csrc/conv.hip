@@ -6,17 +6,21 @@
 //   A row m = input patch of pixel m (gathered on the fly),
 //   B = weights, stored OHWI [Cout][R][S][Cin] so B^T rows are contiguous.
 //
-// Structure (cdna_hip_programming.md §5: the "step-3" / minimum 2-phase
-// shape): 128x64 output tile, BK=64, 4 waves (each 32x64), double-buffered
-// LDS staged by global_load_lds width 16 (lane-linear dest, XOR swizzle
+// Structure (cdna_hip_programming.md §5, "minimum 2-phase" + m-tile loop):
+// 128x64 output tile, BK=64, 4 waves (each 32x64), double-buffered LDS
+// staged by global_load_lds width 16 (lane-linear dest; XOR swizzle
 // applied to the *source* chunk index and the read address — rule 21),
-// one vmcnt(0)+barrier per K-tile. Epilogue fuses folded-BN scale/bias,
-// residual add and ReLU (the Keras Conv2D+BN+Add+ReLU stack the reference
-// executes via model.predict, /root/reference/src/node.py:106).
+// one vmcnt(0)+barrier per K-tile. Each block walks multiple m-tiles
+// (grid-stride) so the pipeline never drains between tiles and — when the
+// whole K fits one tile (1x1 convs over <=64 input channels) — the weight
+// tile is staged once and kept resident in LDS (B_PERSIST).
 //
-// The same kernel is the dense/GEMM path: R=S=1, H=W=1, Cin=K gives
-// out[M,N] = x[M,K] @ w[N,K]^T (+bias, act) — used for the classifier
-// head and the im2col'd stem conv.
+// Epilogue fuses folded-BN scale/bias, residual add and ReLU (the Keras
+// Conv2D+BN+Add+ReLU stack the reference executes via model.predict,
+// /root/reference/src/node.py:106).
+//
+// GEMM mode (R=S=1, H=W=1, Cin=K): out[M,N] = x[M,K] @ w[N,K]^T — used
+// for 1x1/s1 convs, the dense classifier head, and any explicit GEMM.
 #include "common.h"
 #include "kernels.h"
 
@@ -26,20 +30,15 @@ using defer_hip::ConvParams;
 #define BN 64
 #define BK 64
 #define NTHREADS 256
-// chunks are 16-byte (8 bf16) units; A tile = BM*BK bf16 = 1024 chunks,
-// B tile = BN*BK = 512 chunks. Per wave: A 4 glds issues, B 2.
-#define A_CHUNKS (BM * BK / 8)
-#define B_CHUNKS (BN * BK / 8)
-#define KCH (BK / 8)          // chunks per row (8)
+#define KCH (BK / 8)          // 16-B chunks per tile row (8)
 
 // XOR swizzle: logical (row, k8) lives at physical k8p = k8 ^ (row & 7).
-// Read side applies the same XOR on the byte address.
 __device__ __forceinline__ int swz(int row, int k8) {
     return k8 ^ (row & 7);
 }
 
-// async 16B global->LDS for one lane's chunk (wave-uniform LDS base;
-// hardware writes lane i at base + i*16)
+// async 16B global->LDS (wave-uniform LDS base; HW writes lane i at
+// base + i*16)
 __device__ __forceinline__ void glds16(const bf16* src, bf16* lds_base) {
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)src,
@@ -47,57 +46,35 @@ __device__ __forceinline__ void glds16(const bf16* src, bf16* lds_base) {
         16, 0, 0);
 }
 
-template <int ACT, bool HAS_RES, bool GEMM_MODE>
-__global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
+template <int ACT, bool HAS_RES, bool GEMM_MODE, bool B_PERSIST>
+__global__ __launch_bounds__(NTHREADS, 3) void conv_igemm_kernel(
     ConvParams p) {
     const bf16* __restrict__ X = (const bf16*)p.x;
     const bf16* __restrict__ Wt = (const bf16*)p.w;
     const bf16* __restrict__ Z = (const bf16*)p.zbuf;
     const bf16* __restrict__ RES = (const bf16*)p.res;
     bf16* __restrict__ OUT = (bf16*)p.out;
-    __shared__ __attribute__((aligned(16))) bf16 lds[2 * (BM + BN) * BK];
-    bf16* A0 = lds;                         // [BM][BK] x2
-    bf16* B0 = lds + 2 * BM * BK;           // [BN][BK] x2
+    // LDS: A double-buffer + B (single when persistent, double otherwise)
+    __shared__ __attribute__((aligned(16)))
+    bf16 lds[(2 * BM + (B_PERSIST ? 1 : 2) * BN) * BK];
+    bf16* A0 = lds;
+    bf16* B0 = lds + 2 * BM * BK;
 
     const int tid = threadIdx.x;
     const int wave = tid / WAVE;
     const int lane = tid % WAVE;
-    const int m0 = blockIdx.x * BM;
     const int n0 = blockIdx.y * BN;
+    const int mtiles = (p.M + BM - 1) / BM;
+    const int nk = (p.K + BK - 1) / BK;
 
-    // ---- per-thread A-source precomputation (constant across K-tiles).
-    // This wave's glds lanes cover physical chunks (wave*256 + i*64 + lane)
-    // of the A tile; chunk -> (row m, physical k8) -> logical k8.
-    int a_row[4];        // tile-local row of each of my 4 A chunks
-    int a_k8[4];         // logical k8 (after inverse swizzle)
-    const bf16* a_base[4];  // pixel base pointer (at ih0, iw0, c=0)
-    int a_ih0[4], a_iw0[4];
-    bool a_mvalid[4];
+    // ---- per-thread chunk geometry (constant): 4 A chunks, 2 B chunks
+    int a_row[4], a_k8[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-        int chunk = wave * 256 + i * 64 + lane;   // physical chunk index
-        int row = chunk / KCH;
-        int k8p = chunk % KCH;
-        a_row[i] = row;
-        a_k8[i] = swz(row, k8p);                  // logical k8
-        int m = m0 + row;
-        a_mvalid[i] = (m < p.M);
-        int mm = a_mvalid[i] ? m : 0;
-        if (GEMM_MODE) {
-            a_base[i] = X + (long)mm * p.K;
-            a_ih0[i] = 0; a_iw0[i] = 0;
-        } else {
-            int ow = mm % p.OW;
-            int t = mm / p.OW;
-            int oh = t % p.OH;
-            int nb = t / p.OH;
-            int ih0 = oh * p.stride - p.pad;
-            int iw0 = ow * p.stride - p.pad;
-            a_ih0[i] = ih0; a_iw0[i] = iw0;
-            a_base[i] = X + (((long)nb * p.H + ih0) * p.W + iw0) * p.Cin;
-        }
+        int chunk = wave * 256 + i * 64 + lane;
+        a_row[i] = chunk / KCH;
+        a_k8[i] = swz(a_row[i], chunk % KCH);
     }
-    // B chunks: physical chunks (wave*128 + i*64 + lane)
     int b_row[2], b_k8[2];
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
@@ -106,36 +83,40 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
         b_k8[i] = swz(b_row[i], chunk % KCH);
     }
 
-    const int nk = (p.K + BK - 1) / BK;
-
-    // ---- staging: issue glds for K-tile kt into buffer buf (0/1)
-    auto stage = [&](int kt, int buf) {
+    auto stage_a = [&](int mt, int kt, int buf) {
         const int k0 = kt * BK;
+        const int m0 = mt * BM;
         bf16* A = A0 + buf * BM * BK;
-        bf16* B = B0 + buf * BN * BK;
-        // A gather
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
             int k = k0 + a_k8[i] * 8;
+            int m = m0 + a_row[i];
             const bf16* src = Z;
-            if (k < p.K && a_mvalid[i]) {
+            if (k < p.K && m < p.M) {
                 if (GEMM_MODE) {
-                    src = a_base[i] + k;
+                    src = X + (long)m * p.K + k;
                 } else {
+                    int ow = m % p.OW;
+                    int t = m / p.OW;
+                    int oh = t % p.OH;
+                    int nb = t / p.OH;
                     int c = k % p.Cin;
                     int rs = k / p.Cin;
                     int r = rs / p.S;
                     int s = rs % p.S;
-                    int ih = a_ih0[i] + r;
-                    int iw = a_iw0[i] + s;
+                    int ih = oh * p.stride - p.pad + r;
+                    int iw = ow * p.stride - p.pad + s;
                     if (ih >= 0 && ih < p.H && iw >= 0 && iw < p.W)
-                        src = a_base[i] + ((long)r * p.W + s) * p.Cin + c;
+                        src = X + (((long)nb * p.H + ih) * p.W + iw)
+                                    * p.Cin + c;
                 }
             }
-            // wave-uniform LDS base for this glds issue
             glds16(src, A + (wave * 256 + i * 64) * 8);
         }
-        // B weights: row n = n0 + b_row, contiguous in k
+    };
+    auto stage_b = [&](int kt, int buf) {
+        const int k0 = kt * BK;
+        bf16* B = B0 + buf * BN * BK;
 #pragma unroll
         for (int i = 0; i < 2; ++i) {
             int k = k0 + b_k8[i] * 8;
@@ -147,123 +128,109 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
         }
     };
 
-    // ---- MFMA compute on buffer buf
     f32x4 acc[2][4];
-#pragma unroll
-    for (int mi = 0; mi < 2; ++mi)
-#pragma unroll
-        for (int ni = 0; ni < 4; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
-
     const int lo16 = lane & 15;
-    const int hi4 = lane >> 4;   // 0..3
-    auto compute = [&](int buf) {
-        bf16* A = A0 + buf * BM * BK;
-        bf16* B = B0 + buf * BN * BK;
+    const int hi4 = lane >> 4;
+    auto compute = [&](int abuf, int bbuf) {
+        bf16* A = A0 + abuf * BM * BK;
+        bf16* B = B0 + bbuf * BN * BK;
 #pragma unroll
         for (int ks = 0; ks < BK / 32; ++ks) {
-            const int kk = ks * 32 + hi4 * 8;   // bf16 index in row
-            const int k8 = kk / 8;
-            bf16x8 af[2], bf[4];
+            bf16x8 af[2], bfr[4];
 #pragma unroll
             for (int mi = 0; mi < 2; ++mi) {
                 int row = wave * 32 + mi * 16 + lo16;
                 af[mi] = *reinterpret_cast<bf16x8*>(
-                    A + row * BK + swz(row, k8) * 8);
+                    A + row * BK + swz(row, ks * 4 + hi4) * 8);
             }
 #pragma unroll
             for (int ni = 0; ni < 4; ++ni) {
                 int row = ni * 16 + lo16;
-                bf[ni] = *reinterpret_cast<bf16x8*>(
-                    B + row * BK + swz(row, k8) * 8);
+                bfr[ni] = *reinterpret_cast<bf16x8*>(
+                    B + row * BK + swz(row, ks * 4 + hi4) * 8);
             }
 #pragma unroll
             for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
                 for (int ni = 0; ni < 4; ++ni)
                     acc[mi][ni] =
-                        MFMA_BF16_16x16x32(af[mi], bf[ni], acc[mi][ni]);
+                        MFMA_BF16_16x16x32(af[mi], bfr[ni], acc[mi][ni]);
         }
     };
 
-    // ---- main loop: minimum 2-phase (stage t+1 before compute t)
-    stage(0, 0);
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
-    int cur = 0;
-    for (int kt = 0; kt < nk - 1; ++kt) {
-        stage(kt + 1, cur ^ 1);
-        compute(cur);
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        __syncthreads();
-        cur ^= 1;
-    }
-    compute(cur);
-
-    // ---- epilogue: scale/bias + residual + act, bf16 stores
+    auto epilogue = [&](int mt) {
+        const int m0 = mt * BM;
 #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) {
-        int n = n0 + ni * 16 + lo16;
-        if (n >= p.Cout) continue;
-        float sc = p.scale ? p.scale[n] : 1.0f;
-        float bi = p.bias ? p.bias[n] : 0.0f;
+        for (int ni = 0; ni < 4; ++ni) {
+            int n = n0 + ni * 16 + lo16;
+            if (n >= p.Cout) continue;
+            float sc = p.scale ? p.scale[n] : 1.0f;
+            float bi = p.bias ? p.bias[n] : 0.0f;
 #pragma unroll
-        for (int mi = 0; mi < 2; ++mi) {
+            for (int mi = 0; mi < 2; ++mi) {
 #pragma unroll
-            for (int e = 0; e < 4; ++e) {
-                int m = m0 + wave * 32 + mi * 16 + hi4 * 4 + e;
-                if (m >= p.M) continue;
-                float v = acc[mi][ni][e] * sc + bi;
-                if (HAS_RES) v += bf2f(RES[(long)m * p.Cout + n]);
-                OUT[(long)m * p.Cout + n] = f2bf(apply_act(v, ACT));
+                for (int e = 0; e < 4; ++e) {
+                    int m = m0 + wave * 32 + mi * 16 + hi4 * 4 + e;
+                    if (m >= p.M) continue;
+                    float v = acc[mi][ni][e] * sc + bi;
+                    if (HAS_RES) v += bf2f(RES[(long)m * p.Cout + n]);
+                    OUT[(long)m * p.Cout + n] = f2bf(apply_act(v, ACT));
+                }
             }
         }
+    };
+
+    // ---- flattened (m-tile, k-tile) pipeline; stage cursor runs one
+    // iteration ahead of the compute cursor.
+    int s_mt = blockIdx.x, s_kt = 0;
+    stage_a(s_mt, s_kt, 0);
+    stage_b(s_kt, 0);
+    // advance stage cursor
+    if (++s_kt == nk) { s_kt = 0; s_mt += gridDim.x; }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    int cur = 0;
+    for (int mt = blockIdx.x; mt < mtiles; mt += gridDim.x) {
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+            for (int ni = 0; ni < 4; ++ni)
+                acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+        for (int kt = 0; kt < nk; ++kt) {
+            if (s_mt < mtiles) {
+                stage_a(s_mt, s_kt, cur ^ 1);
+                if (!B_PERSIST) stage_b(s_kt, cur ^ 1);
+                if (++s_kt == nk) { s_kt = 0; s_mt += gridDim.x; }
+            }
+            compute(cur, B_PERSIST ? 0 : cur);
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __syncthreads();
+            cur ^= 1;
+        }
+        epilogue(mt);
     }
 }
 
 // ---------------------------------------------------------------------------
-// im2col for the stem conv (Cin=3: channel chunks are too thin for the
-// implicit gather). Produces [M][K_pad] bf16, zero-padded beyond
-// K = R*S*Cin, which then runs through the GEMM mode of the kernel above.
-__global__ void im2col_kernel(const bf16* __restrict__ x, bf16* __restrict__ out,
-                              int NB, int H, int W, int Cin,
-                              int OH, int OW, int R, int S,
-                              int stride, int pad, int Kpad) {
-    long M = (long)NB * OH * OW;
-    int K = R * S * Cin;
-    long total = M * Kpad;
+// Channel pad, NHWC: [rows, C] -> [rows, C8] zero-padded (stem Cin=3 -> 8;
+// also pads OHWI weights viewed as [Cout*R*S, Cin]).
+__global__ void pad_channels_kernel(const bf16* __restrict__ x,
+                                    bf16* __restrict__ y, long rows, int C,
+                                    int C8) {
+    long total = rows * (C8 / 8);
     long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
     long gs = (long)gridDim.x * blockDim.x;
     for (long i = i0; i < total; i += gs) {
-        int k = (int)(i % Kpad);
-        long m = i / Kpad;
-        bf16 v = (bf16)0.f;
-        if (k < K) {
-            int c = k % Cin;
-            int rs = k / Cin;
-            int r = rs / S, s = rs % S;
-            int ow = (int)(m % OW);
-            long t = m / OW;
-            int oh = (int)(t % OH);
-            int nb = (int)(t / OH);
-            int ih = oh * stride - pad + r;
-            int iw = ow * stride - pad + s;
-            if (ih >= 0 && ih < H && iw >= 0 && iw < W)
-                v = x[(((long)nb * H + ih) * W + iw) * Cin + c];
+        int cb = (int)(i % (C8 / 8));
+        long row = i / (C8 / 8);
+        bf16x8 v;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            int c = cb * 8 + j;
+            v[j] = (c < C) ? x[row * C + c] : (bf16)0.f;
         }
-        out[i] = v;
-    }
-}
-
-// Zero-pad a weight tensor's K dim: [Cout][K] -> [Cout][Kpad]
-__global__ void padk_kernel(const bf16* __restrict__ w, bf16* __restrict__ out,
-                            int Cout, int K, int Kpad) {
-    long total = (long)Cout * Kpad;
-    long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    long gs = (long)gridDim.x * blockDim.x;
-    for (long i = i0; i < total; i += gs) {
-        int k = (int)(i % Kpad);
-        long n = i / Kpad;
-        out[i] = (k < K) ? w[n * K + k] : (bf16)0.f;
+        store_bf16x8(y + row * C8 + cb * 8, v);
     }
 }
 
@@ -273,22 +240,38 @@ namespace defer_hip {
 
 void launch_conv_igemm(const ConvParams& p, bool relu, bool has_res,
                        bool gemm_mode, hipStream_t s) {
-    dim3 grid((p.M + BM - 1) / BM, (p.Cout + BN - 1) / BN);
-    dim3 block(NTHREADS);
-#define DISPATCH(A, R, G) \
-    hipLaunchKernelGGL((conv_igemm_kernel<A, R, G>), grid, block, 0, s, p)
-    if (gemm_mode) {
-        if (relu) { if (has_res) DISPATCH(ACT_RELU, true, true);
-                    else DISPATCH(ACT_RELU, false, true); }
-        else      { if (has_res) DISPATCH(ACT_NONE, true, true);
-                    else DISPATCH(ACT_NONE, false, true); }
-    } else {
-        if (relu) { if (has_res) DISPATCH(ACT_RELU, true, false);
-                    else DISPATCH(ACT_RELU, false, false); }
-        else      { if (has_res) DISPATCH(ACT_NONE, true, false);
-                    else DISPATCH(ACT_NONE, false, false); }
+    const int mtiles = (p.M + BM - 1) / BM;
+    const int ny = (p.Cout + BN - 1) / BN;
+    // target ~3 blocks/CU x 256 CUs; each block m-loops the rest
+    int gx = mtiles;
+    const int target = 768;
+    if ((long)mtiles * ny > target) {
+        gx = target / ny > 0 ? target / ny : 1;
+        if (gx > mtiles) gx = mtiles;
     }
-#undef DISPATCH
+    dim3 grid(gx, ny);
+    dim3 block(NTHREADS);
+    const int nk = (p.K + BK - 1) / BK;
+    const bool bp = (nk == 1);
+#define DISPATCH4(A, R, G, BP) \
+    hipLaunchKernelGGL((conv_igemm_kernel<A, R, G, BP>), grid, block, 0, \
+                       s, p)
+#define DISPATCH2(A, R)                                        \
+    do {                                                       \
+        if (gemm_mode) {                                       \
+            if (bp) DISPATCH4(A, R, true, true);               \
+            else DISPATCH4(A, R, true, false);                 \
+        } else {                                               \
+            if (bp) DISPATCH4(A, R, false, true);              \
+            else DISPATCH4(A, R, false, false);                \
+        }                                                      \
+    } while (0)
+    if (relu) { if (has_res) DISPATCH2(ACT_RELU, true);
+                else DISPATCH2(ACT_RELU, false); }
+    else      { if (has_res) DISPATCH2(ACT_NONE, true);
+                else DISPATCH2(ACT_NONE, false); }
+#undef DISPATCH2
+#undef DISPATCH4
 }
 
 static int grid1d(long work, int block) {
@@ -296,20 +279,11 @@ static int grid1d(long work, int block) {
     return (int)(g < 2048 ? g : 2048);
 }
 
-void launch_im2col(const void* x, void* out, int NB, int H, int W, int Cin,
-                   int OH, int OW, int R, int S, int stride, int pad,
-                   int Kpad, hipStream_t s) {
-    long M = (long)NB * OH * OW;
-    hipLaunchKernelGGL(im2col_kernel, dim3(grid1d(M * Kpad, 256)),
-                       dim3(256), 0, s, (const bf16*)x, (bf16*)out, NB, H,
-                       W, Cin, OH, OW, R, S, stride, pad, Kpad);
-}
-
-void launch_padk(const void* w, void* out, int Cout, int K, int Kpad,
-                 hipStream_t s) {
-    hipLaunchKernelGGL(padk_kernel, dim3(grid1d((long)Cout * Kpad, 256)),
-                       dim3(256), 0, s, (const bf16*)w, (bf16*)out, Cout, K,
-                       Kpad);
+void launch_pad_channels(const void* x, void* y, long rows, int C, int C8,
+                         hipStream_t s) {
+    hipLaunchKernelGGL(pad_channels_kernel,
+                       dim3(grid1d(rows * (C8 / 8), 256)), dim3(256), 0, s,
+                       (const bf16*)x, (bf16*)y, rows, C, C8);
 }
 
 }  // namespace defer_hip

@@ -22,11 +22,8 @@ struct ConvParams {
 
 void launch_conv_igemm(const ConvParams& p, bool relu, bool has_res,
                        bool gemm_mode, hipStream_t s);
-void launch_im2col(const void* x, void* out, int NB, int H, int W, int Cin,
-                   int OH, int OW, int R, int S, int stride, int pad,
-                   int Kpad, hipStream_t s);
-void launch_padk(const void* w, void* out, int Cout, int K, int Kpad,
-                 hipStream_t s);
+void launch_pad_channels(const void* x, void* y, long rows, int C, int C8,
+                         hipStream_t s);
 
 void launch_bn_act(const void* x, const float* scale, const float* bias,
                    void* y, long total8, int c8, bool relu, hipStream_t s);

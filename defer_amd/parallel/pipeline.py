@@ -53,7 +53,12 @@ class StageExecutor:
     """
 
     def __init__(self, stage: GraphModel, device, dtype: torch.dtype,
-                 use_graph: bool = False):
+                 use_graph: bool = False, fuse: bool = True):
+        if fuse:
+            from defer_amd.parallel.fusion import fuse_residual_adds
+
+            stage = GraphModel(fuse_residual_adds(stage.graph),
+                               name=stage.model_name)
         self.model = stage
         self.device = torch.device(device)
         self.dtype = dtype

@@ -69,3 +69,38 @@ def test_node_costs_total_flops():
     # ResNet50 @224 is ~8.2 GFLOP (2*MAC) per image
     assert 7.5e9 < total < 9.0e9
     assert ob["add_1"] == 56 * 56 * 256 * 2.0
+
+
+def test_fusion_preserves_semantics():
+    from defer_amd.graph import GraphModel
+    from defer_amd.parallel.fusion import fuse_residual_adds
+    from defer_amd.models.layers import ConvBNAct, AddAct
+
+    m = resnet50()
+    x = torch.randn(1, 64, 64, 3)
+    with torch.no_grad():
+        want = m(x)
+    fused = GraphModel(fuse_residual_adds(m.graph))
+    with torch.no_grad():
+        got = fused(x)
+    assert torch.allclose(got, want, atol=1e-5)
+    # all 16 residual adds fused away
+    kinds = [type(n.layer).__name__ for n in fused.graph.nodes]
+    assert "AddAct" not in kinds
+    assert kinds.count("FusedConvAddAct") == 16
+    # cut names still valid
+    assert "add_8" in fused.graph.valid_cut_points()
+
+
+def test_stage_executor_applies_fusion_cpu():
+    from defer_amd.parallel.pipeline import StageExecutor
+    from defer_amd.graph import GraphModel
+
+    m = resnet50()
+    x = torch.randn(1, 64, 64, 3)
+    with torch.no_grad():
+        want = m(x)
+    ex = StageExecutor(GraphModel(m.graph), "cpu", torch.float32)
+    with torch.no_grad():
+        got = ex.run(x)
+    assert torch.allclose(got, want, atol=1e-5)
