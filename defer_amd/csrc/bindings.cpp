@@ -33,8 +33,25 @@ const void* zero_buf() {
     return bptr(z);
 }
 
-const float* fptr_opt(const c10::optional<Tensor>& t, const char* name) {
-    if (!t) return nullptr;
+// scale/bias must be non-null in the kernel (nullable per-element loads
+// de-pipeline the epilogue); substitute cached ones/zeros when absent.
+const float* ones_buf() {
+    static Tensor t = at::ones(
+        {8192}, at::TensorOptions().dtype(at::kFloat).device(at::kCUDA));
+    return t.data_ptr<float>();
+}
+const float* zeros_buf() {
+    static Tensor t = at::zeros(
+        {8192}, at::TensorOptions().dtype(at::kFloat).device(at::kCUDA));
+    return t.data_ptr<float>();
+}
+
+const float* fptr_opt(const c10::optional<Tensor>& t, const char* name,
+                      const float* fallback, int64_t n) {
+    if (!t) {
+        TORCH_CHECK(n <= 8192, "Cout too large for cached scale/bias");
+        return fallback;
+    }
     TORCH_CHECK(t->scalar_type() == at::kFloat, name, " must be fp32");
     TORCH_CHECK(t->is_contiguous(), name, " must be contiguous");
     return t->data_ptr<float>();
@@ -59,8 +76,8 @@ Tensor conv2d_bn_act(Tensor x, Tensor w, c10::optional<Tensor> scale,
     hipStream_t s = cur_stream();
 
     ConvParams p{};
-    p.scale = fptr_opt(scale, "scale");
-    p.bias = fptr_opt(bias, "bias");
+    p.scale = fptr_opt(scale, "scale", ones_buf(), Cout);
+    p.bias = fptr_opt(bias, "bias", zeros_buf(), Cout);
     p.res = res ? bptr(*res) : nullptr;
     p.out = bptr_mut(out);
     p.zbuf = zero_buf();
@@ -115,8 +132,8 @@ Tensor linear(Tensor x, Tensor w, c10::optional<Tensor> bias) {
     if (bias) bias_f = bias->to(at::kFloat).contiguous();
     ConvParams p{};
     p.x = bptr(x); p.w = bptr(w);
-    p.scale = nullptr;
-    p.bias = fptr_opt(bias_f, "bias");
+    p.scale = fptr_opt(c10::nullopt, "scale", ones_buf(), N);
+    p.bias = fptr_opt(bias_f, "bias", zeros_buf(), N);
     p.res = nullptr; p.out = bptr_mut(out); p.zbuf = zero_buf();
     p.M = M; p.K = K; p.Cout = N;
     p.NB = M; p.H = 1; p.W = 1; p.Cin = K;

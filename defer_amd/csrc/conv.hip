@@ -6,14 +6,17 @@
 //   A row m = input patch of pixel m (gathered on the fly),
 //   B = weights, stored OHWI [Cout][R][S][Cin] so B^T rows are contiguous.
 //
-// Structure (cdna_hip_programming.md §5, "minimum 2-phase" + m-tile loop):
-// 128x64 output tile, BK=64, 4 waves (each 32x64), double-buffered LDS
+// Structure (cdna_hip_programming.md §5, T3+T4: counted vmcnt pipeline):
+// 128x64 output tile, BK=64, 4 waves (each 32x64), TRIPLE-buffered LDS
 // staged by global_load_lds width 16 (lane-linear dest; XOR swizzle
-// applied to the *source* chunk index and the read address — rule 21),
-// one vmcnt(0)+barrier per K-tile. Each block walks multiple m-tiles
-// (grid-stride) so the pipeline never drains between tiles and — when the
-// whole K fits one tile (1x1 convs over <=64 input channels) — the weight
-// tile is staged once and kept resident in LDS (B_PERSIST).
+// applied to the *source* chunk index and the read address — rule 21).
+// The stage cursor runs two K-tiles ahead; each iteration waits a COUNTED
+// s_waitcnt vmcnt(OPS) (never 0 mid-loop) and a raw s_barrier, so glds
+// stay in flight across barriers and the ~1 us HBM latency hides under
+// two iterations of MFMA. Each block walks multiple m-tiles (grid-stride)
+// so the pipeline never drains between tiles; when the whole K fits one
+// tile (1x1 convs over <=64 input channels) the weight tile is staged
+// once and kept resident in LDS (B_PERSIST).
 //
 // Epilogue fuses folded-BN scale/bias, residual add and ReLU (the Keras
 // Conv2D+BN+Add+ReLU stack the reference executes via model.predict,
@@ -47,18 +50,18 @@ __device__ __forceinline__ void glds16(const bf16* src, bf16* lds_base) {
 }
 
 template <int ACT, bool HAS_RES, bool GEMM_MODE, bool B_PERSIST>
-__global__ __launch_bounds__(NTHREADS, 3) void conv_igemm_kernel(
+__global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
     ConvParams p) {
     const bf16* __restrict__ X = (const bf16*)p.x;
     const bf16* __restrict__ Wt = (const bf16*)p.w;
     const bf16* __restrict__ Z = (const bf16*)p.zbuf;
     const bf16* __restrict__ RES = (const bf16*)p.res;
     bf16* __restrict__ OUT = (bf16*)p.out;
-    // LDS: A double-buffer + B (single when persistent, double otherwise)
+    // LDS: A triple-buffer + B (single when persistent, triple otherwise)
     __shared__ __attribute__((aligned(16)))
-    bf16 lds[(2 * BM + (B_PERSIST ? 1 : 2) * BN) * BK];
+    bf16 lds[(3 * BM + (B_PERSIST ? 1 : 3) * BN) * BK];
     bf16* A0 = lds;
-    bf16* B0 = lds + 2 * BM * BK;
+    bf16* B0 = lds + 3 * BM * BK;
 
     const int tid = threadIdx.x;
     const int wave = tid / WAVE;
@@ -81,6 +84,23 @@ __global__ __launch_bounds__(NTHREADS, 3) void conv_igemm_kernel(
         int chunk = wave * 128 + i * 64 + lane;
         b_row[i] = chunk / KCH;
         b_k8[i] = swz(b_row[i], chunk % KCH);
+    }
+
+    const int lo16 = lane & 15;
+    const int hi4 = lane >> 4;
+    // folded-BN scale/bias for this block's 64 output channels, loaded
+    // once (p.scale/p.bias are always non-null; the binding substitutes
+    // cached ones/zeros) — per-element nullable loads inside the epilogue
+    // made hipcc emit one vmcnt(0) per element (de-pipelining trap).
+    const float* __restrict__ SCALE = p.scale;
+    const float* __restrict__ BIAS = p.bias;
+    float sc[4], bi[4];
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+        int n = n0 + ni * 16 + lo16;
+        if (n >= p.Cout) n = p.Cout - 1;
+        sc[ni] = SCALE[n];
+        bi[ni] = BIAS[n];
     }
 
     auto stage_a = [&](int mt, int kt, int buf) {
@@ -129,8 +149,6 @@ __global__ __launch_bounds__(NTHREADS, 3) void conv_igemm_kernel(
     };
 
     f32x4 acc[2][4];
-    const int lo16 = lane & 15;
-    const int hi4 = lane >> 4;
     auto compute = [&](int abuf, int bbuf) {
         bf16* A = A0 + abuf * BM * BK;
         bf16* B = B0 + bbuf * BN * BK;
@@ -158,57 +176,143 @@ __global__ __launch_bounds__(NTHREADS, 3) void conv_igemm_kernel(
         }
     };
 
-    auto epilogue = [&](int mt) {
-        const int m0 = mt * BM;
+    auto epilogue = [&](int mt, int abuf) {
+        // Coalesced epilogue via an LDS transpose bounce through the A
+        // buffer just consumed (free until stage((it+2)%3), which differs
+        // from abuf): the MFMA C-fragment layout is column-scattered, so
+        // direct stores are 32 scalar 2-B ops per lane whose completion
+        // the next counted vmcnt would drain. Instead: acc*scale+bias ->
+        // XOR-swizzled f32 LDS tile (two 64-row halves), then every wave
+        // reads rows back contiguously and does 8-B residual loads +
+        // 8-B bf16 stores.
+        float* scratch = (float*)(A0 + abuf * BM * BK);   // 16 KB
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();       // all waves done with A[abuf]
 #pragma unroll
-        for (int ni = 0; ni < 4; ++ni) {
-            int n = n0 + ni * 16 + lo16;
-            if (n >= p.Cout) continue;
-            float sc = p.scale ? p.scale[n] : 1.0f;
-            float bi = p.bias ? p.bias[n] : 0.0f;
+        for (int h = 0; h < 2; ++h) {
+            if ((wave >> 1) == h) {
+                const int rbase = (wave & 1) * 32;
 #pragma unroll
-            for (int mi = 0; mi < 2; ++mi) {
+                for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
-                for (int e = 0; e < 4; ++e) {
-                    int m = m0 + wave * 32 + mi * 16 + hi4 * 4 + e;
-                    if (m >= p.M) continue;
-                    float v = acc[mi][ni][e] * sc + bi;
-                    if (HAS_RES) v += bf2f(RES[(long)m * p.Cout + n]);
-                    OUT[(long)m * p.Cout + n] = f2bf(apply_act(v, ACT));
+                    for (int ni = 0; ni < 4; ++ni)
+#pragma unroll
+                        for (int e = 0; e < 4; ++e) {
+                            int r = rbase + mi * 16 + hi4 * 4 + e;
+                            int c = ni * 16 + lo16;
+                            int cs = c ^ (((r >> 2) & 3) << 4);
+                            scratch[r * 64 + cs] =
+                                acc[mi][ni][e] * sc[ni] + bi[ni];
+                        }
+            }
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+            // readback: 64 rows x 16 chunks of 4 floats, 4 chunks/thread
+            // Interior blocks (the common case) take a branch-free
+            // vector path: batch-issue 4 LDS reads + 4 residual loads,
+            // sched_barrier, then convert+store — so hipcc emits counted
+            // waits instead of one vmcnt(0) per chunk (de-pipelining
+            // trap) and no per-chunk exec-mask dances.
+            const bool interior =
+                (mt * BM + BM <= p.M) && (n0 + BN <= p.Cout);
+            if (interior) {
+                f32x4 v4[4];
+                bf16x4 rv[4];
+                long off[4];
+#pragma unroll
+                for (int i = 0; i < 4; ++i) {
+                    int chunk = tid + i * 256;
+                    int r = chunk >> 4;
+                    int c4 = (chunk & 15) * 4;
+                    int cs = c4 ^ (((r >> 2) & 3) << 4);
+                    off[i] = (long)(mt * BM + h * 64 + r) * p.Cout
+                             + n0 + c4;
+                    v4[i] = *reinterpret_cast<f32x4*>(
+                        scratch + r * 64 + cs);
+                    if (HAS_RES)
+                        rv[i] = *reinterpret_cast<const bf16x4*>(
+                            RES + off[i]);
+                }
+                __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+                for (int i = 0; i < 4; ++i) {
+                    bf16x4 o;
+#pragma unroll
+                    for (int j = 0; j < 4; ++j) {
+                        float v = v4[i][j];
+                        if (HAS_RES) v += bf2f(rv[i][j]);
+                        o[j] = f2bf(apply_act(v, ACT));
+                    }
+                    *reinterpret_cast<bf16x4*>(OUT + off[i]) = o;
+                }
+            } else {
+#pragma unroll
+                for (int i = 0; i < 4; ++i) {
+                    int chunk = tid + i * 256;
+                    int r = chunk >> 4;
+                    int c4 = (chunk & 15) * 4;
+                    int cs = c4 ^ (((r >> 2) & 3) << 4);
+                    int m = mt * BM + h * 64 + r;
+                    int n = n0 + c4;
+                    if (m >= p.M || n >= p.Cout) continue;
+                    f32x4 v4 = *reinterpret_cast<f32x4*>(
+                        scratch + r * 64 + cs);
+                    long off = (long)m * p.Cout + n;
+#pragma unroll
+                    for (int j = 0; j < 4; ++j) {
+                        if (n + j >= p.Cout) continue;
+                        float v = v4[j];
+                        if (HAS_RES) v += bf2f(RES[off + j]);
+                        OUT[off + j] = f2bf(apply_act(v, ACT));
+                    }
                 }
             }
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();   // next half reuses scratch
         }
     };
 
-    // ---- flattened (m-tile, k-tile) pipeline; stage cursor runs one
-    // iteration ahead of the compute cursor.
-    int s_mt = blockIdx.x, s_kt = 0;
-    stage_a(s_mt, s_kt, 0);
-    stage_b(s_kt, 0);
-    // advance stage cursor
-    if (++s_kt == nk) { s_kt = 0; s_mt += gridDim.x; }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
+    // ---- flattened (m-tile, k-tile) pipeline with a 2-tile-deep stage
+    // cursor. Per-wave glds ops per staged tile: 4 A (+2 B unless
+    // persistent). vmcnt waits are counted so prefetches stay in flight.
+    constexpr int OPS = B_PERSIST ? 4 : 6;
 
-    int cur = 0;
+    int s_mt = blockIdx.x, s_kt = 0;
+    auto advance = [&]() {
+        if (++s_kt == nk) { s_kt = 0; s_mt += gridDim.x; }
+    };
+    auto stage = [&](int buf) {
+        stage_a(s_mt, s_kt, buf);
+        if (!B_PERSIST) stage_b(s_kt, buf);
+        advance();
+    };
+
+    if (B_PERSIST) stage_b(0, 0);   // oldest ops: drained by first wait
+    int staged = 0;
+    if (s_mt < mtiles) { stage(0); ++staged; }
+    if (s_mt < mtiles) { stage(1); ++staged; }
+
+    int it = 0;   // computed-iteration counter (tile i lives in buf i%3)
     for (int mt = blockIdx.x; mt < mtiles; mt += gridDim.x) {
 #pragma unroll
         for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
             for (int ni = 0; ni < 4; ++ni)
                 acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
-        for (int kt = 0; kt < nk; ++kt) {
-            if (s_mt < mtiles) {
-                stage_a(s_mt, s_kt, cur ^ 1);
-                if (!B_PERSIST) stage_b(s_kt, cur ^ 1);
-                if (++s_kt == nk) { s_kt = 0; s_mt += gridDim.x; }
-            }
-            compute(cur, B_PERSIST ? 0 : cur);
-            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-            __syncthreads();
-            cur ^= 1;
+        for (int kt = 0; kt < nk; ++kt, ++it) {
+            // tile `it` landed when <= OPS*(tiles-in-flight-behind-it)
+            // ops remain outstanding
+            if (staged - it - 1 >= 1)
+                asm volatile("s_waitcnt vmcnt(%0)" ::"i"(OPS) : "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+            compute(it % 3, B_PERSIST ? 0 : it % 3);
+            if (kt == nk - 1) epilogue(mt, it % 3);
+            // refill: buffer (it+2)%3 was last read at compute(it-1),
+            // which every wave finished before this iteration's barrier
+            if (s_mt < mtiles) { stage((it + 2) % 3); ++staged; }
         }
-        epilogue(mt);
     }
 }
 
