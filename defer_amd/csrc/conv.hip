@@ -6,21 +6,27 @@
 //   A row m = input patch of pixel m (gathered on the fly),
 //   B = weights, stored OHWI [Cout][R][S][Cin] so B^T rows are contiguous.
 //
-// Structure (cdna_hip_programming.md §5, T3+T4: counted vmcnt pipeline):
-// 128x64 output tile, BK=64, 4 waves (each 32x64), TRIPLE-buffered LDS
-// staged by global_load_lds width 16 (lane-linear dest; XOR swizzle
-// applied to the *source* chunk index and the read address — rule 21).
-// The stage cursor runs two K-tiles ahead; each iteration waits a COUNTED
-// s_waitcnt vmcnt(OPS) (never 0 mid-loop) and a raw s_barrier, so glds
-// stay in flight across barriers and the ~1 us HBM latency hides under
-// two iterations of MFMA. Each block walks multiple m-tiles (grid-stride)
-// so the pipeline never drains between tiles; when the whole K fits one
-// tile (1x1 convs over <=64 input channels) the weight tile is staged
-// once and kept resident in LDS (B_PERSIST).
+// Structure (cdna_hip_programming.md §5, T3+T4 counted-vmcnt pipeline):
+// template-parameterized BM x BN output tile, BK=64, 4 waves,
+// DEPTH-buffered LDS staged by global_load_lds width 16 (lane-linear
+// dest; XOR swizzle applied to the *source* chunk index and the read
+// address — rule 21). The stage cursor runs DEPTH-1 K-tiles ahead; each
+// iteration waits a COUNTED s_waitcnt vmcnt(OPS) — never 0 mid-loop —
+// and a raw s_barrier, so glds stay in flight across barriers and HBM
+// latency hides under MFMA. Each block walks multiple m-tiles
+// (grid-stride) so the pipeline never drains between tiles; when the
+// whole K fits one tile, the weight tile is staged once (B_PERSIST).
+//
+// Tile configs (picked per shape by the launcher):
+//   BM128/BN64/D3  — default; small-Cout or modest grids
+//   BM128/BN128/D2 — Cout>=128 with plenty of m-tiles: halves A
+//                    re-staging across n-blocks, 2x MFMA per staged byte
+//   BM64 variants  — small-M shapes (7x7/14x14 batches): 2x the blocks
 //
 // Epilogue fuses folded-BN scale/bias, residual add and ReLU (the Keras
-// Conv2D+BN+Add+ReLU stack the reference executes via model.predict,
-// /root/reference/src/node.py:106).
+// Conv2D+BN+Add+ReLU stack the reference runs via model.predict,
+// /root/reference/src/node.py:106), bounced through the just-freed LDS
+// A-buffer so residual loads and bf16 stores are 8-B coalesced.
 //
 // GEMM mode (R=S=1, H=W=1, Cin=K): out[M,N] = x[M,K] @ w[N,K]^T — used
 // for 1x1/s1 convs, the dense classifier head, and any explicit GEMM.
@@ -29,8 +35,6 @@
 
 using defer_hip::ConvParams;
 
-#define BM 128
-#define BN 64
 #define BK 64
 #define NTHREADS 256
 #define KCH (BK / 8)          // 16-B chunks per tile row (8)
@@ -40,8 +44,6 @@ __device__ __forceinline__ int swz(int row, int k8) {
     return k8 ^ (row & 7);
 }
 
-// async 16B global->LDS (wave-uniform LDS base; HW writes lane i at
-// base + i*16)
 __device__ __forceinline__ void glds16(const bf16* src, bf16* lds_base) {
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)src,
@@ -49,7 +51,8 @@ __device__ __forceinline__ void glds16(const bf16* src, bf16* lds_base) {
         16, 0, 0);
 }
 
-template <int ACT, bool HAS_RES, bool GEMM_MODE, bool B_PERSIST>
+template <int ACT, bool HAS_RES, bool GEMM_MODE, bool B_PERSIST, int BM,
+          int BN, int DEPTH>
 __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
     ConvParams p) {
     const bf16* __restrict__ X = (const bf16*)p.x;
@@ -57,47 +60,57 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
     const bf16* __restrict__ Z = (const bf16*)p.zbuf;
     const bf16* __restrict__ RES = (const bf16*)p.res;
     bf16* __restrict__ OUT = (bf16*)p.out;
-    // LDS: A triple-buffer + B (single when persistent, triple otherwise)
+
+    // wave tiling: WMW waves along m (32 rows each), WNW along n
+    constexpr int WMW = BM / 32;                // 2 or 4
+    constexpr int WNW = 4 / WMW;                // 2 or 1
+    constexpr int WN = BN / WNW;                // wave n-width
+    constexpr int NI = WN / 16;                 // B frags per wave
+    constexpr int ACH = BM / 32;                // A chunks per thread
+    constexpr int BCH = BN / 32;                // B chunks per thread
+    constexpr int OPS = ACH + (B_PERSIST ? 0 : BCH);
+
     __shared__ __attribute__((aligned(16)))
-    bf16 lds[(3 * BM + (B_PERSIST ? 1 : 3) * BN) * BK];
+    bf16 lds[(DEPTH * BM + (B_PERSIST ? 1 : DEPTH) * BN) * BK];
     bf16* A0 = lds;
-    bf16* B0 = lds + 3 * BM * BK;
+    bf16* B0 = lds + DEPTH * BM * BK;
 
     const int tid = threadIdx.x;
     const int wave = tid / WAVE;
     const int lane = tid % WAVE;
+    const int wm = wave / WNW;                  // wave m index
+    const int wn = wave % WNW;                  // wave n index
     const int n0 = blockIdx.y * BN;
     const int mtiles = (p.M + BM - 1) / BM;
     const int nk = (p.K + BK - 1) / BK;
 
-    // ---- per-thread chunk geometry (constant): 4 A chunks, 2 B chunks
-    int a_row[4], a_k8[4];
+    // ---- per-thread chunk geometry (constant across tiles)
+    int a_row[ACH], a_k8[ACH];
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-        int chunk = wave * 256 + i * 64 + lane;
+    for (int i = 0; i < ACH; ++i) {
+        int chunk = wave * (ACH * 64) + i * 64 + lane;
         a_row[i] = chunk / KCH;
         a_k8[i] = swz(a_row[i], chunk % KCH);
     }
-    int b_row[2], b_k8[2];
+    int b_row[BCH], b_k8[BCH];
 #pragma unroll
-    for (int i = 0; i < 2; ++i) {
-        int chunk = wave * 128 + i * 64 + lane;
+    for (int i = 0; i < BCH; ++i) {
+        int chunk = wave * (BCH * 64) + i * 64 + lane;
         b_row[i] = chunk / KCH;
         b_k8[i] = swz(b_row[i], chunk % KCH);
     }
 
     const int lo16 = lane & 15;
     const int hi4 = lane >> 4;
-    // folded-BN scale/bias for this block's 64 output channels, loaded
-    // once (p.scale/p.bias are always non-null; the binding substitutes
-    // cached ones/zeros) — per-element nullable loads inside the epilogue
-    // made hipcc emit one vmcnt(0) per element (de-pipelining trap).
+    // folded-BN scale/bias for this wave's n-columns, loaded once
+    // (p.scale/p.bias always non-null; nullable per-element loads in the
+    // epilogue made hipcc emit one vmcnt(0) per element).
     const float* __restrict__ SCALE = p.scale;
     const float* __restrict__ BIAS = p.bias;
-    float sc[4], bi[4];
+    float sc[NI], bi[NI];
 #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) {
-        int n = n0 + ni * 16 + lo16;
+    for (int ni = 0; ni < NI; ++ni) {
+        int n = n0 + wn * WN + ni * 16 + lo16;
         if (n >= p.Cout) n = p.Cout - 1;
         sc[ni] = SCALE[n];
         bi[ni] = BIAS[n];
@@ -108,7 +121,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
         const int m0 = mt * BM;
         bf16* A = A0 + buf * BM * BK;
 #pragma unroll
-        for (int i = 0; i < 4; ++i) {
+        for (int i = 0; i < ACH; ++i) {
             int k = k0 + a_k8[i] * 8;
             int m = m0 + a_row[i];
             const bf16* src = Z;
@@ -131,111 +144,101 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
                                     * p.Cin + c;
                 }
             }
-            glds16(src, A + (wave * 256 + i * 64) * 8);
+            glds16(src, A + (wave * (ACH * 64) + i * 64) * 8);
         }
     };
     auto stage_b = [&](int kt, int buf) {
         const int k0 = kt * BK;
         bf16* B = B0 + buf * BN * BK;
 #pragma unroll
-        for (int i = 0; i < 2; ++i) {
+        for (int i = 0; i < BCH; ++i) {
             int k = k0 + b_k8[i] * 8;
             int n = n0 + b_row[i];
             const bf16* src = (k < p.K && n < p.Cout)
                                   ? Wt + (long)n * p.K + k
                                   : Z;
-            glds16(src, B + (wave * 128 + i * 64) * 8);
+            glds16(src, B + (wave * (BCH * 64) + i * 64) * 8);
         }
     };
 
-    f32x4 acc[2][4];
+    f32x4 acc[2][NI];
     auto compute = [&](int abuf, int bbuf) {
         bf16* A = A0 + abuf * BM * BK;
         bf16* B = B0 + bbuf * BN * BK;
 #pragma unroll
         for (int ks = 0; ks < BK / 32; ++ks) {
-            bf16x8 af[2], bfr[4];
+            bf16x8 af[2], bfr[NI];
 #pragma unroll
             for (int mi = 0; mi < 2; ++mi) {
-                int row = wave * 32 + mi * 16 + lo16;
+                int row = wm * 32 + mi * 16 + lo16;
                 af[mi] = *reinterpret_cast<bf16x8*>(
                     A + row * BK + swz(row, ks * 4 + hi4) * 8);
             }
 #pragma unroll
-            for (int ni = 0; ni < 4; ++ni) {
-                int row = ni * 16 + lo16;
+            for (int ni = 0; ni < NI; ++ni) {
+                int row = wn * WN + ni * 16 + lo16;
                 bfr[ni] = *reinterpret_cast<bf16x8*>(
                     B + row * BK + swz(row, ks * 4 + hi4) * 8);
             }
 #pragma unroll
             for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
-                for (int ni = 0; ni < 4; ++ni)
+                for (int ni = 0; ni < NI; ++ni)
                     acc[mi][ni] =
                         MFMA_BF16_16x16x32(af[mi], bfr[ni], acc[mi][ni]);
         }
     };
 
+    // ---- epilogue: LDS transpose bounce through the A buffer just
+    // consumed; RH output rows per round (RH*BN f32 = one A buffer).
+    constexpr int RH = (BM * BK * 2 / 4) / BN;   // 64/32/16 rows per round
+    constexpr int ROUNDS = BM / RH;
+    constexpr int CPR = RH * BN / 4 / NTHREADS;  // 16-B chunks per thread
     auto epilogue = [&](int mt, int abuf) {
-        // Coalesced epilogue via an LDS transpose bounce through the A
-        // buffer just consumed (free until stage((it+2)%3), which differs
-        // from abuf): the MFMA C-fragment layout is column-scattered, so
-        // direct stores are 32 scalar 2-B ops per lane whose completion
-        // the next counted vmcnt would drain. Instead: acc*scale+bias ->
-        // XOR-swizzled f32 LDS tile (two 64-row halves), then every wave
-        // reads rows back contiguously and does 8-B residual loads +
-        // 8-B bf16 stores.
-        float* scratch = (float*)(A0 + abuf * BM * BK);   // 16 KB
+        float* scratch = (float*)(A0 + abuf * BM * BK);
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();       // all waves done with A[abuf]
 #pragma unroll
-        for (int h = 0; h < 2; ++h) {
-            if ((wave >> 1) == h) {
-                const int rbase = (wave & 1) * 32;
+        for (int h = 0; h < ROUNDS; ++h) {
+            const int r0 = h * RH;           // tile-local first row
 #pragma unroll
-                for (int mi = 0; mi < 2; ++mi)
+            for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
-                    for (int ni = 0; ni < 4; ++ni)
+                for (int ni = 0; ni < NI; ++ni)
 #pragma unroll
-                        for (int e = 0; e < 4; ++e) {
-                            int r = rbase + mi * 16 + hi4 * 4 + e;
-                            int c = ni * 16 + lo16;
-                            int cs = c ^ (((r >> 2) & 3) << 4);
-                            scratch[r * 64 + cs] =
-                                acc[mi][ni][e] * sc[ni] + bi[ni];
-                        }
-            }
+                    for (int e = 0; e < 4; ++e) {
+                        int r = wm * 32 + mi * 16 + hi4 * 4 + e;
+                        if (r < r0 || r >= r0 + RH) continue;
+                        int c = wn * WN + ni * 16 + lo16;
+                        int rl = r - r0;
+                        int cs = c ^ (((rl >> 2) & 3) << 4);
+                        scratch[rl * BN + cs] =
+                            acc[mi][ni][e] * sc[ni] + bi[ni];
+                    }
             asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
             __builtin_amdgcn_s_barrier();
-            // readback: 64 rows x 16 chunks of 4 floats, 4 chunks/thread
-            // Interior blocks (the common case) take a branch-free
-            // vector path: batch-issue 4 LDS reads + 4 residual loads,
-            // sched_barrier, then convert+store — so hipcc emits counted
-            // waits instead of one vmcnt(0) per chunk (de-pipelining
-            // trap) and no per-chunk exec-mask dances.
             const bool interior =
                 (mt * BM + BM <= p.M) && (n0 + BN <= p.Cout);
             if (interior) {
-                f32x4 v4[4];
-                bf16x4 rv[4];
-                long off[4];
+                f32x4 v4[CPR];
+                bf16x4 rv[CPR];
+                long off[CPR];
 #pragma unroll
-                for (int i = 0; i < 4; ++i) {
-                    int chunk = tid + i * 256;
-                    int r = chunk >> 4;
-                    int c4 = (chunk & 15) * 4;
-                    int cs = c4 ^ (((r >> 2) & 3) << 4);
-                    off[i] = (long)(mt * BM + h * 64 + r) * p.Cout
-                             + n0 + c4;
+                for (int i = 0; i < CPR; ++i) {
+                    int chunk = tid + i * NTHREADS;
+                    int rl = chunk / (BN / 4);
+                    int c4 = (chunk % (BN / 4)) * 4;
+                    int cs = c4 ^ (((rl >> 2) & 3) << 4);
+                    off[i] = (long)(mt * BM + r0 + rl) * p.Cout + n0 + c4;
                     v4[i] = *reinterpret_cast<f32x4*>(
-                        scratch + r * 64 + cs);
+                        scratch + rl * BN + cs);
                     if (HAS_RES)
                         rv[i] = *reinterpret_cast<const bf16x4*>(
                             RES + off[i]);
                 }
                 __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
-                for (int i = 0; i < 4; ++i) {
+                for (int i = 0; i < CPR; ++i) {
                     bf16x4 o;
 #pragma unroll
                     for (int j = 0; j < 4; ++j) {
@@ -247,16 +250,16 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
                 }
             } else {
 #pragma unroll
-                for (int i = 0; i < 4; ++i) {
-                    int chunk = tid + i * 256;
-                    int r = chunk >> 4;
-                    int c4 = (chunk & 15) * 4;
-                    int cs = c4 ^ (((r >> 2) & 3) << 4);
-                    int m = mt * BM + h * 64 + r;
+                for (int i = 0; i < CPR; ++i) {
+                    int chunk = tid + i * NTHREADS;
+                    int rl = chunk / (BN / 4);
+                    int c4 = (chunk % (BN / 4)) * 4;
+                    int cs = c4 ^ (((rl >> 2) & 3) << 4);
+                    int m = mt * BM + r0 + rl;
                     int n = n0 + c4;
                     if (m >= p.M || n >= p.Cout) continue;
                     f32x4 v4 = *reinterpret_cast<f32x4*>(
-                        scratch + r * 64 + cs);
+                        scratch + rl * BN + cs);
                     long off = (long)m * p.Cout + n;
 #pragma unroll
                     for (int j = 0; j < 4; ++j) {
@@ -268,15 +271,14 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
                 }
             }
             asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-            __builtin_amdgcn_s_barrier();   // next half reuses scratch
+            __builtin_amdgcn_s_barrier();   // next round reuses scratch
         }
     };
 
-    // ---- flattened (m-tile, k-tile) pipeline with a 2-tile-deep stage
-    // cursor. Per-wave glds ops per staged tile: 4 A (+2 B unless
-    // persistent). vmcnt waits are counted so prefetches stay in flight.
-    constexpr int OPS = B_PERSIST ? 4 : 6;
-
+    // ---- flattened (m-tile, k-tile) pipeline; the stage at the END of
+    // each iteration keeps the cursor DEPTH-1 tiles ahead of compute
+    // (the refilled buffer was last read at compute(it-(DEPTH-1)), which
+    // every wave finished before this iteration's barrier).
     int s_mt = blockIdx.x, s_kt = 0;
     auto advance = [&]() {
         if (++s_kt == nk) { s_kt = 0; s_mt += gridDim.x; }
@@ -289,29 +291,31 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
 
     if (B_PERSIST) stage_b(0, 0);   // oldest ops: drained by first wait
     int staged = 0;
-    if (s_mt < mtiles) { stage(0); ++staged; }
-    if (s_mt < mtiles) { stage(1); ++staged; }
+#pragma unroll
+    for (int d = 0; d < DEPTH - 1; ++d)
+        if (s_mt < mtiles) { stage(d); ++staged; }
 
-    int it = 0;   // computed-iteration counter (tile i lives in buf i%3)
+    int it = 0;   // computed-iteration counter; tile i -> buffer i%DEPTH
     for (int mt = blockIdx.x; mt < mtiles; mt += gridDim.x) {
 #pragma unroll
         for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
-            for (int ni = 0; ni < 4; ++ni)
+            for (int ni = 0; ni < NI; ++ni)
                 acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
         for (int kt = 0; kt < nk; ++kt, ++it) {
-            // tile `it` landed when <= OPS*(tiles-in-flight-behind-it)
-            // ops remain outstanding
+            // tile `it` landed once <= OPS*(tiles staged after it)
+            // remain outstanding
             if (staged - it - 1 >= 1)
                 asm volatile("s_waitcnt vmcnt(%0)" ::"i"(OPS) : "memory");
             else
                 asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
             __builtin_amdgcn_s_barrier();
-            compute(it % 3, B_PERSIST ? 0 : it % 3);
-            if (kt == nk - 1) epilogue(mt, it % 3);
-            // refill: buffer (it+2)%3 was last read at compute(it-1),
-            // which every wave finished before this iteration's barrier
-            if (s_mt < mtiles) { stage((it + 2) % 3); ++staged; }
+            compute(it % DEPTH, B_PERSIST ? 0 : it % DEPTH);
+            if (kt == nk - 1) epilogue(mt, it % DEPTH);
+            if (s_mt < mtiles) {
+                stage((it + DEPTH - 1) % DEPTH);
+                ++staged;
+            }
         }
     }
 }
@@ -344,9 +348,26 @@ namespace defer_hip {
 
 void launch_conv_igemm(const ConvParams& p, bool relu, bool has_res,
                        bool gemm_mode, hipStream_t s) {
-    const int mtiles = (p.M + BM - 1) / BM;
-    const int ny = (p.Cout + BN - 1) / BN;
-    // target ~3 blocks/CU x 256 CUs; each block m-loops the rest
+    const int nk = (p.K + BK - 1) / BK;
+    const bool bp = (nk == 1);
+    // tile selection. BN128/D2 halves A re-staging and doubles MFMA per
+    // staged byte but its 2-deep pipeline waits vmcnt(0) per K-tile, so
+    // it only pays on deep-K (3x3) shapes; memory-bound small-K shapes
+    // keep the 3-deep BN64 pipeline. BM=64 for small-M shapes (more
+    // blocks on the 256 CUs).
+    const int mt128 = (p.M + 127) / 128;
+    const int mt64 = (p.M + 63) / 64;
+    int BMsel, BNsel;
+    const bool deepK = p.K >= 1024;
+    if (deepK && p.Cout >= 128) {
+        BNsel = 128;
+        BMsel = (mt128 * (p.Cout / 128) >= 384) ? 128 : 64;
+    } else {
+        BNsel = 64;
+        BMsel = (mt128 * ((p.Cout + 63) / 64) >= 384) ? 128 : 64;
+    }
+    const int mtiles = (p.M + BMsel - 1) / BMsel;
+    const int ny = (p.Cout + BNsel - 1) / BNsel;
     int gx = mtiles;
     const int target = 768;
     if ((long)mtiles * ny > target) {
@@ -355,27 +376,74 @@ void launch_conv_igemm(const ConvParams& p, bool relu, bool has_res,
     }
     dim3 grid(gx, ny);
     dim3 block(NTHREADS);
-    const int nk = (p.K + BK - 1) / BK;
-    const bool bp = (nk == 1);
-#define DISPATCH4(A, R, G, BP) \
-    hipLaunchKernelGGL((conv_igemm_kernel<A, R, G, BP>), grid, block, 0, \
-                       s, p)
-#define DISPATCH2(A, R)                                        \
-    do {                                                       \
-        if (gemm_mode) {                                       \
-            if (bp) DISPATCH4(A, R, true, true);               \
-            else DISPATCH4(A, R, true, false);                 \
-        } else {                                               \
-            if (bp) DISPATCH4(A, R, false, true);              \
-            else DISPATCH4(A, R, false, false);                \
-        }                                                      \
+
+#define DISPATCH_TILE(A, R, G, BP, BMv, BNv, D)                           \
+    hipLaunchKernelGGL(                                                   \
+        (conv_igemm_kernel<A, R, G, BP, BMv, BNv, D>), grid, block, 0, s, \
+        p)
+#define DISPATCH_BOOLS(BMv, BNv, D)                                       \
+    do {                                                                  \
+        if (relu) {                                                       \
+            if (has_res) {                                                \
+                if (gemm_mode) {                                          \
+                    if (bp) DISPATCH_TILE(ACT_RELU, true, true, true,     \
+                                          BMv, BNv, D);                   \
+                    else DISPATCH_TILE(ACT_RELU, true, true, false, BMv,  \
+                                       BNv, D);                           \
+                } else {                                                  \
+                    if (bp) DISPATCH_TILE(ACT_RELU, true, false, true,    \
+                                          BMv, BNv, D);                   \
+                    else DISPATCH_TILE(ACT_RELU, true, false, false, BMv, \
+                                       BNv, D);                           \
+                }                                                         \
+            } else {                                                      \
+                if (gemm_mode) {                                          \
+                    if (bp) DISPATCH_TILE(ACT_RELU, false, true, true,    \
+                                          BMv, BNv, D);                   \
+                    else DISPATCH_TILE(ACT_RELU, false, true, false, BMv, \
+                                       BNv, D);                           \
+                } else {                                                  \
+                    if (bp) DISPATCH_TILE(ACT_RELU, false, false, true,   \
+                                          BMv, BNv, D);                   \
+                    else DISPATCH_TILE(ACT_RELU, false, false, false,     \
+                                       BMv, BNv, D);                      \
+                }                                                         \
+            }                                                             \
+        } else {                                                          \
+            if (has_res) {                                                \
+                if (gemm_mode) {                                          \
+                    if (bp) DISPATCH_TILE(ACT_NONE, true, true, true,     \
+                                          BMv, BNv, D);                   \
+                    else DISPATCH_TILE(ACT_NONE, true, true, false, BMv,  \
+                                       BNv, D);                           \
+                } else {                                                  \
+                    if (bp) DISPATCH_TILE(ACT_NONE, true, false, true,    \
+                                          BMv, BNv, D);                   \
+                    else DISPATCH_TILE(ACT_NONE, true, false, false, BMv, \
+                                       BNv, D);                           \
+                }                                                         \
+            } else {                                                      \
+                if (gemm_mode) {                                          \
+                    if (bp) DISPATCH_TILE(ACT_NONE, false, true, true,    \
+                                          BMv, BNv, D);                   \
+                    else DISPATCH_TILE(ACT_NONE, false, true, false, BMv, \
+                                       BNv, D);                           \
+                } else {                                                  \
+                    if (bp) DISPATCH_TILE(ACT_NONE, false, false, true,   \
+                                          BMv, BNv, D);                   \
+                    else DISPATCH_TILE(ACT_NONE, false, false, false,     \
+                                       BMv, BNv, D);                      \
+                }                                                         \
+            }                                                             \
+        }                                                                 \
     } while (0)
-    if (relu) { if (has_res) DISPATCH2(ACT_RELU, true);
-                else DISPATCH2(ACT_RELU, false); }
-    else      { if (has_res) DISPATCH2(ACT_NONE, true);
-                else DISPATCH2(ACT_NONE, false); }
-#undef DISPATCH2
-#undef DISPATCH4
+
+    if (BMsel == 128 && BNsel == 64) DISPATCH_BOOLS(128, 64, 3);
+    else if (BMsel == 128 && BNsel == 128) DISPATCH_BOOLS(128, 128, 2);
+    else if (BMsel == 64 && BNsel == 128) DISPATCH_BOOLS(64, 128, 2);
+    else DISPATCH_BOOLS(64, 64, 3);
+#undef DISPATCH_BOOLS
+#undef DISPATCH_TILE
 }
 
 static int grid1d(long work, int block) {
