@@ -209,6 +209,51 @@ Tensor global_avg_pool(Tensor x) {
     return out;
 }
 
+std::tuple<int64_t, int64_t, int64_t> field_shape(const Tensor& t) {
+    auto sz = t.sizes();
+    int nd = sz.size();
+    if (nd == 1) return {1, 1, sz[0]};
+    if (nd == 2) return {1, sz[0], sz[1]};
+    int64_t d0 = 1;
+    for (int i = 0; i + 2 < nd; ++i) d0 *= sz[i];
+    return {d0, sz[nd - 2], sz[nd - 1]};
+}
+
+Tensor zfp_encode(Tensor x, int64_t rate, c10::optional<Tensor> out) {
+    TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x must be GPU contig");
+    bool bf16 = x.scalar_type() == at::kBFloat16;
+    TORCH_CHECK(bf16 || x.scalar_type() == at::kFloat, "bf16/f32 only");
+    auto [d0, d1, d2] = field_shape(x);
+    long nb = ((d0 + 3) / 4) * ((d1 + 3) / 4) * ((d2 + 3) / 4);
+    long bytes = nb * rate * 8;
+    Tensor o;
+    if (out) {
+        TORCH_CHECK(out->numel() == bytes && out->is_contiguous()
+                    && out->scalar_type() == at::kByte, "bad out buffer");
+        o = *out;
+    } else {
+        o = at::empty({bytes},
+                      x.options().dtype(at::kByte));
+    }
+    defer_hip::launch_zfp_encode(bptr(x), bptr_mut(o), bf16, (int)d0,
+                                 (int)d1, (int)d2, (int)rate,
+                                 cur_stream());
+    return o;
+}
+
+Tensor zfp_decode(Tensor wire, std::vector<int64_t> shape, int64_t rate,
+                  bool bf16_out) {
+    TORCH_CHECK(wire.is_cuda() && wire.is_contiguous()
+                && wire.scalar_type() == at::kByte, "bad wire");
+    auto o = at::empty(shape, wire.options().dtype(
+                                  bf16_out ? at::kBFloat16 : at::kFloat));
+    auto [d0, d1, d2] = field_shape(o);
+    defer_hip::launch_zfp_decode(bptr(wire), bptr_mut(o), bf16_out,
+                                 (int)d0, (int)d1, (int)d2, (int)rate,
+                                 cur_stream());
+    return o;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -222,4 +267,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("softmax", &softmax);
     m.def("maxpool2d", &maxpool2d);
     m.def("global_avg_pool", &global_avg_pool);
+    m.def("zfp_encode", &zfp_encode, py::arg("x"), py::arg("rate"),
+          py::arg("out") = py::none());
+    m.def("zfp_decode", &zfp_decode);
 }

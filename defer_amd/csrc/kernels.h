@@ -38,4 +38,10 @@ void launch_maxpool(const void* x, void* y, int NB, int H, int W, int C,
 void launch_gap(const void* x, void* y, int NB, int HW, int C,
                 hipStream_t s);
 
+// fixed-rate ZFP-style codec (see csrc/codec.hip / ops/zfp_ref.py)
+void launch_zfp_encode(const void* x, void* out, bool bf16_in, int d0,
+                       int d1, int d2, int rate, hipStream_t s);
+void launch_zfp_decode(const void* wire, void* y, bool bf16_out, int d0,
+                       int d1, int d2, int rate, hipStream_t s);
+
 }  // namespace defer_hip
