@@ -24,7 +24,8 @@ def _worker(rank, world, port, q, cuts, steps, compression):
         model = resnet50()
         cfg = PipelineConfig(device="cpu", dtype="fp32",
                              partition_layers=cuts, ring_depth=2,
-                             compression=compression, backend="gloo")
+                             compression=compression, zfp_rate_bits=14,
+                             backend="gloo")
         B = 1
         pipe = DistPipeline(model, cfg, (B, 64, 64, 3))
 
@@ -79,3 +80,10 @@ def test_dist_pipeline_three_stage_unbalanced():
 @pytest.mark.slow
 def test_dist_pipeline_auto_cuts():
     _run(2, None)
+
+
+def test_dist_pipeline_with_zfp_compression():
+    """Config 4 analogue on CPU: 2-stage pipeline with the ZFP relay.
+    Lossy codec -> compare against the uncompressed model with tolerance.
+    """
+    _run(2, ["add_8"], steps=2, compression="zfp", tol=0.05)
