@@ -96,26 +96,45 @@ struct BitWriter {
     }
 };
 
+// Word-window bit reader: keeps up to 64 stream bits in `win`; take(k)
+// returns the next k bits LSB-first (zeros past the end of the stream,
+// like the reference's exhausted reads).
 struct BitReader {
     const u32* in;
-    u32 cur;
-    int navail;
-    int budget;
+    u64 win;
+    int nwin;        // bits in window
+    int left;        // stream bits not yet in window
 
     __device__ void init(const u32* p, int bits) {
-        in = p; budget = bits; cur = *in++; navail = 32;
+        in = p; win = 0; nwin = 0; left = bits;
+        refill();
     }
-    // returns bit (0 past end); ok flag via pointer
-    __device__ int get1(bool* ok) {
-        if (budget <= 0) { *ok = false; return 0; }
-        --budget;
-        if (navail == 0) { cur = *in++; navail = 32; }
-        int b = cur & 1;
-        cur >>= 1;
-        --navail;
-        *ok = true;
-        return b;
+    __device__ void refill() {
+        while (nwin <= 32 && left > 0) {
+            u32 w = *in++;
+            int t = left < 32 ? left : 32;
+            u64 ww = (t < 32) ? (w & ((1u << t) - 1u)) : w;
+            win |= ww << nwin;
+            nwin += t;
+            left -= t;
+        }
     }
+    __device__ u64 take(int k) {   // k <= 64; zeros past stream end
+        u64 v = 0;
+        int got = 0;
+        while (got < k) {
+            refill();
+            if (nwin == 0) break;
+            int t = (k - got) < nwin ? (k - got) : nwin;
+            u64 mask = (t >= 64) ? ~0ull : ((1ull << t) - 1ull);
+            v |= (win & mask) << got;
+            win = (t >= 64) ? 0 : (win >> t);
+            nwin -= t;
+            got += t;
+        }
+        return v;
+    }
+    __device__ bool empty() const { return nwin == 0 && left == 0; }
 };
 
 // ---------------------------------------------------------------------------
@@ -181,20 +200,20 @@ __global__ void zfp_encode_kernel(const void* __restrict__ xv,
                 wr.put_bits(x & ((n < 64) ? ((1ull << n) - 1ull)
                                           : ~0ull), n);
                 x >>= n;
-                // group tests
+                // group tests, run-wise: each run is tz zeros then a
+                // one = the word (1 << tz) over tz+1 bits (bit-identical
+                // to per-bit emission; put_bits truncates at the budget
+                // exactly like per-bit puts would)
                 while (n < 64) {
-                    int has = (x != 0);
                     if (wr.budget <= 0) break;
+                    int has = (x != 0);
                     wr.put1(has);
                     if (!has) break;
-                    while (true) {
-                        int b = (int)(x & 1);
-                        if (wr.budget <= 0) { n = 64; break; }
-                        wr.put1(b);
-                        x >>= 1;
-                        ++n;
-                        if (b || n == 64) break;
-                    }
+                    int tz = __builtin_ctzll(x);
+                    int emit = tz + 1;
+                    wr.put_bits(1ull << tz, emit);
+                    x = (emit >= 64) ? 0 : (x >> emit);
+                    n += emit;
                 }
             }
             // all lanes need n for the next plane's ballot bookkeeping?
@@ -233,44 +252,61 @@ __global__ void zfp_decode_kernel(const u32* __restrict__ wire,
         long idx = ((long)gi * d1 + gj) * d2 + gk;
 
         const u32* bin = wire + blk * wpb;
-        // lane 0 parses; plane words broadcast to all lanes
+        // lane 0 parses word-wise; plane words broadcast to all lanes.
+        // Truncation endgame differs from the per-bit reference only in
+        // zero bits / internal n, so reconstruction is identical.
         BitReader rd;
         rd.init(bin, rate * 64);
         u32 hdr = 0;
-        if (lane == 0) {
-            for (int i = 0; i < HDR_BITS; ++i) {
-                bool ok;
-                hdr |= (u32)rd.get1(&ok) << i;
-            }
-        }
+        if (lane == 0) hdr = (u32)rd.take(HDR_BITS);
         hdr = __shfl(hdr, 0);
         float outv = 0.f;
         if (hdr >> 15) {
             int emax = (int)(hdr & 0x1FFu) - 256;
             u32 u = 0;
             int n = 0;
+            int done = 0;
             for (int p = PLANES - 1; p >= 0; --p) {
-                int bud = __shfl(rd.budget, 0);
-                if (bud <= 0) break;
                 u64 x = 0;
                 if (lane == 0) {
-                    for (int i = 0; i < n; ++i) {
-                        bool ok;
-                        x |= (u64)rd.get1(&ok) << i;
-                    }
-                    while (n < 64) {
-                        bool ok;
-                        int b = rd.get1(&ok);
-                        if (!ok || !b) break;
-                        while (true) {
-                            bool ok2;
-                            int b2 = rd.get1(&ok2);
-                            x |= (u64)b2 << n;
-                            ++n;
-                            if (b2 || n == 64 || !ok2) break;
+                    if (rd.empty()) {
+                        done = 1;
+                    } else {
+                        x = rd.take(n);               // significant prefix
+                        while (n < 64) {              // group runs
+                            if (rd.empty()) break;
+                            if (!rd.take(1)) break;   // group test
+                            // scan zeros (may span window refills) until
+                            // the run's 1 bit
+                            bool found = false;
+                            while (n < 64) {
+                                rd.refill();
+                                int lim = 64 - n;
+                                int avail =
+                                    rd.nwin < lim ? rd.nwin : lim;
+                                if (avail == 0) break;  // stream end
+                                u64 w = rd.win &
+                                        ((avail >= 64)
+                                             ? ~0ull
+                                             : ((1ull << avail) - 1ull));
+                                int tz = w ? __builtin_ctzll(w) : avail;
+                                if (tz < avail) {
+                                    rd.take(tz + 1);
+                                    x |= 1ull << (n + tz);
+                                    n += tz + 1;
+                                    found = true;
+                                    break;
+                                }
+                                rd.take(avail);       // all zeros
+                                n += avail;
+                                if (rd.empty()) break;
+                            }
+                            if (!found) break;
                         }
                     }
                 }
+                done = __shfl(done, 0);
+                if (done) break;
                 x = __shfl(x, 0);
                 n = __shfl(n, 0);
                 u |= (u32)((x >> lane) & 1) << p;
