@@ -40,6 +40,8 @@ def parse_args():
     ap.add_argument("--device", default="cuda", choices=["cuda", "cpu"])
     ap.add_argument("--ring-depth", type=int, default=4)
     ap.add_argument("--no-return-results", action="store_true")
+    ap.add_argument("--stats", action="store_true",
+                    help="print per-stage stats to stderr")
     return ap.parse_args()
 
 
@@ -132,6 +134,14 @@ def main():
     images = args.steps * B
     ips = images / el
 
+    if args.stats:
+        import sys
+        st = pipe.stats
+        print(f"[stage {pipe.rank}] items={st.items} images={st.images} "
+              f"wire_in={st.bytes_in/1e6:.1f}MB wire_out="
+              f"{st.bytes_out/1e6:.1f}MB "
+              f"({st.bytes_out/max(st.items,1)/1e6:.2f} MB/item)",
+              file=sys.stderr)
     if pipe.rank == 0:
         out = {
             "metric": "images/sec (whole node) ResNet50 pipeline"

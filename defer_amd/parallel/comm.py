@@ -54,6 +54,12 @@ class Codec:
                              dtype=self.dtype)
 
 
+def dtype_bytes(dt) -> int:
+    import torch as _t
+
+    return _t.empty(0, dtype=dt).element_size()
+
+
 class P2PRing:
     """Ring of preallocated recv (or send) buffers with outstanding-work
     tracking — the device-resident analogue of the reference's bounded

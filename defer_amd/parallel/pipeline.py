@@ -361,6 +361,14 @@ class DistPipeline:
                     w = dist.irecv(self.result_ring.slot(k + D), src=last)
                     self.result_ring.set_work(k + D, w)
             self.stats.items += 1
+            self.stats.images += self.batch_shape[0]
+            from defer_amd.parallel.comm import dtype_bytes
+            if r > 0:
+                self.stats.bytes_in += (self.in_codec.wire_numel
+                                        * dtype_bytes(self.in_codec.wire_dtype))
+            if r < last:
+                self.stats.bytes_out += (self.out_codec.wire_numel
+                                         * dtype_bytes(self.out_codec.wire_dtype))
 
         # drain outstanding sends
         for ring in (self.send_ring, self.result_ring):

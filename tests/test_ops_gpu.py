@@ -166,3 +166,49 @@ def test_resnet50_full_forward_vs_cpu():
     assert (cos > 0.98).all(), f"cosine {cos}"
     top_match = (got.argmax(-1) == want.argmax(-1)).float().mean()
     assert top_match >= 0.75, f"top1 agreement {top_match}"
+
+
+@requires_gpu
+def test_defer_threaded_multistage_on_gpu():
+    """4-stage DEFER pipeline chained on one GPU (stage handoff via
+    device tensors) matches the single-stage GPU forward."""
+    import queue
+    import threading
+
+    from defer_amd import DEFER, PipelineConfig
+    from defer_amd.graph import GraphModel
+    from defer_amd.models import resnet50
+    from defer_amd.parallel.pipeline import StageExecutor
+
+    torch.manual_seed(0)
+    m = resnet50()
+    ex = StageExecutor(GraphModel(m.graph), DEV, torch.bfloat16)
+    xs = [torch.randn(2, 224, 224, 3) for _ in range(3)]
+    with torch.no_grad():
+        want = [ex.run(x.to(DEV, torch.bfloat16)).float().cpu()
+                for x in xs]
+
+    eng = DEFER(["cuda:0"] * 4,
+                config=PipelineConfig(device="cuda", dtype="bf16"))
+    in_q, out_q = queue.Queue(8), queue.Queue(8)
+    t = threading.Thread(target=eng.run_defer,
+                         args=(resnet50_copy_like(m), None, in_q, out_q))
+    t.start()
+    for x in xs:
+        in_q.put(x.to(DEV, torch.bfloat16))
+    in_q.put(None)
+    outs = [out_q.get(timeout=180) for _ in xs]
+    t.join(timeout=60)
+    for o, w in zip(outs, want):
+        assert o.shape == w.shape
+        assert (o - w).abs().max() < 1e-2
+
+
+def resnet50_copy_like(m):
+    """Fresh resnet50 with the same weights (same seed construction)."""
+    import torch as _t
+
+    from defer_amd.models import resnet50 as _r
+
+    _t.manual_seed(0)
+    return _r()
