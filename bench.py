@@ -127,9 +127,11 @@ def main():
     barrier_sync()
     t1 = time.perf_counter()
 
-    elapsed = torch.tensor([t1 - t0], dtype=torch.float64)
+    elapsed = torch.tensor([t1 - t0], dtype=torch.float64,
+                           device=dev if args.device == "cuda" else "cpu")
     if dist_mode:
-        dist.all_reduce(elapsed, op=dist.ReduceOp.MAX)
+        dist.all_reduce(elapsed, op=dist.ReduceOp.MAX)  # NCCL needs GPU
+    elapsed = elapsed.cpu()
     el = elapsed.item()
     images = args.steps * B
     ips = images / el

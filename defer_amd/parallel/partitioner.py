@@ -24,8 +24,10 @@ from defer_amd.graph import GraphModel, LayerGraph, from_torch
 # hop, full duplex. Used only as a relative weight in the cut chooser.
 XGMI_LINK_GBPS = 153.0
 # Effective sustained compute for weighting conv FLOPs against hop bytes
-# (relative units; ratio is what matters).
-EFF_TFLOPS = 800.0
+# (relative units; ratio is what matters). Calibrated against the measured
+# whole-model ResNet50 bf16 forward on MI355X (~2.9 ms / 64 images at
+# 8.2 GFLOP/image ~= 180 TF effective — tools/convbench.py).
+EFF_TFLOPS = 180.0
 
 
 def as_graph_model(model) -> GraphModel:

@@ -93,7 +93,13 @@ class StageExecutor:
             with torch.no_grad():
                 return self.model(x)
         if self._graph is None:
-            self._capture(x)
+            try:
+                self._capture(x)
+            except Exception:
+                # capture not supported for this stage: eager fallback
+                self.use_graph = False
+                with torch.no_grad():
+                    return self.model(x)
         self._static_in.copy_(x)
         self._graph.replay()
         return self._static_out
