@@ -254,6 +254,35 @@ Tensor zfp_decode(Tensor wire, std::vector<int64_t> shape, int64_t rate,
     return o;
 }
 
+Tensor lz4_compress(Tensor x) {
+    TORCH_CHECK(x.is_cuda() && x.is_contiguous()
+                && x.scalar_type() == at::kByte, "x must be GPU u8 contig");
+    long n = x.numel();
+    TORCH_CHECK(n > 0, "empty input");
+    auto scratch = at::empty({defer_hip::lz4_scratch_bytes(n)},
+                             x.options());
+    auto o = at::empty({defer_hip::lz4_max_compressed(n)}, x.options());
+    defer_hip::launch_lz4_compress(bptr(x), n, bptr_mut(scratch),
+                                   bptr_mut(o), cur_stream());
+    // total compressed size = header word [2 + nblocks] (syncs the stream;
+    // the hop needs the byte count on the host to post the send anyway)
+    long nb = (n + 4095) / 4096;
+    auto hdr_cpu = o.narrow(0, 4 * (2 + nb), 4).to(at::kCPU);
+    uint32_t total;
+    std::memcpy(&total, hdr_cpu.data_ptr(), 4);
+    return o.narrow(0, 0, 4 * (2 + nb + 1) + (long)total);
+}
+
+Tensor lz4_decompress(Tensor comp, int64_t raw_len) {
+    TORCH_CHECK(comp.is_cuda() && comp.is_contiguous()
+                && comp.scalar_type() == at::kByte, "bad comp buffer");
+    TORCH_CHECK(raw_len > 0, "raw_len must be positive");
+    auto o = at::empty({raw_len}, comp.options());
+    defer_hip::launch_lz4_decompress(bptr(comp), bptr_mut(o), raw_len,
+                                     cur_stream());
+    return o;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -270,4 +299,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("zfp_encode", &zfp_encode, py::arg("x"), py::arg("rate"),
           py::arg("out") = py::none());
     m.def("zfp_decode", &zfp_decode);
+    m.def("lz4_compress", &lz4_compress);
+    m.def("lz4_decompress", &lz4_decompress);
 }

@@ -44,4 +44,12 @@ void launch_zfp_encode(const void* x, void* out, bool bf16_in, int d0,
 void launch_zfp_decode(const void* wire, void* y, bool bf16_out, int d0,
                        int d1, int d2, int rate, hipStream_t s);
 
+// LZ4-style block compressor (see csrc/lz4.hip / ops/lz4_ref.py)
+long lz4_max_compressed(long n);
+long lz4_scratch_bytes(long n);
+void launch_lz4_compress(const void* in, long n, void* scratch, void* out,
+                         hipStream_t s);
+void launch_lz4_decompress(const void* comp, void* out, long raw_len,
+                           hipStream_t s);
+
 }  // namespace defer_hip

@@ -24,7 +24,7 @@ import torch.distributed as dist
 
 from defer_amd.config import PipelineConfig
 from defer_amd.graph import GraphModel, LayerGraph
-from defer_amd.parallel.comm import Codec, P2PRing
+from defer_amd.parallel.comm import Codec, P2PRing, dtype_bytes, make_ring
 from defer_amd.parallel.partitioner import (as_graph_model, auto_partition,
                                             partition_model)
 
@@ -247,11 +247,11 @@ class DistPipeline:
             if self.rank > 0:
                 self.in_codec = Codec(cfg, self.in_shape, self.dtype,
                                       self.device)
-                self.recv_ring = P2PRing(self.in_codec, cfg.ring_depth)
+                self.recv_ring = make_ring(self.in_codec, cfg.ring_depth)
             if self.rank < self.world - 1:
                 self.out_codec = Codec(cfg, self.out_shape, self.dtype,
                                        self.device)
-                self.send_ring = P2PRing(self.out_codec, cfg.ring_depth)
+                self.send_ring = make_ring(self.out_codec, cfg.ring_depth)
             if cfg.return_results:
                 # logits hop last -> 0, never compressed (tiny)
                 res_cfg = PipelineConfig(compression="none")
@@ -318,21 +318,18 @@ class DistPipeline:
 
         # pre-post recvs (reverse-order ready before senders start)
         if r > 0:
-            for k in range(min(D, steps)):
-                w = dist.irecv(self.recv_ring.slot(k), src=prv)
-                self.recv_ring.set_work(k, w)
+            self.recv_ring.prime(steps, prv)
         if r == 0 and cfg.return_results:
-            for k in range(min(D, steps)):
-                w = dist.irecv(self.result_ring.slot(k), src=last)
-                self.result_ring.set_work(k, w)
+            self.result_ring.prime(steps, last)
 
         for k in range(steps):
             # ---- obtain input
             if r == 0:
                 x = feed(k)
             else:
-                self.recv_ring.wait(k)
-                wire = self.recv_ring.slot(k)
+                wire = self.recv_ring.wait_recv(k, steps, prv)
+                self.stats.bytes_in += (wire.numel()
+                                        * dtype_bytes(wire.dtype))
                 x = self.in_codec.decode(wire)
             # ---- compute
             y = self.stage.run(x)
@@ -340,45 +337,29 @@ class DistPipeline:
             # AFTER the compute consuming it was enqueued, and
             # ProcessGroupNCCL orders the recv after the current stream's
             # already-enqueued work (CPU/gloo compute is synchronous).
-            if r > 0 and k + D < steps:
-                w = dist.irecv(self.recv_ring.slot(k + D), src=prv)
-                self.recv_ring.set_work(k + D, w)
+            if r > 0:
+                self.recv_ring.repost(k, steps, prv)
             # ---- forward result
             if r < last:
-                self.send_ring.wait(k)  # slot free (send k-D done)
-                wire = self.out_codec.encode(y, out=self.send_ring.slot(k))
-                w = dist.isend(wire, dst=nxt)
-                self.send_ring.set_work(k, w)
+                self.stats.bytes_out += self.send_ring.send_encoded(
+                    k, y, nxt)
             elif cfg.return_results:
-                self.result_ring.wait(k)
-                buf = self.result_ring.slot(k)
-                buf.copy_(y.reshape(-1))
-                w = dist.isend(buf, dst=0)
-                self.result_ring.set_work(k, w)
+                self.result_ring.send_encoded(k, y, 0)
             elif collect:
                 collect(k, y)
             # ---- rank0 result collection
             if r == 0 and cfg.return_results:
-                self.result_ring.wait(k)
+                buf = self.result_ring.wait_recv(k, steps, last)
                 if collect:
-                    collect(k, self.result_ring.slot(k).view(
-                        self.res_codec.shape))
-                if k + D < steps:
-                    w = dist.irecv(self.result_ring.slot(k + D), src=last)
-                    self.result_ring.set_work(k + D, w)
+                    collect(k, buf.view(self.res_codec.shape))
+                self.result_ring.repost(k, steps, last)
             self.stats.items += 1
             self.stats.images += self.batch_shape[0]
-            from defer_amd.parallel.comm import dtype_bytes
-            if r > 0:
-                self.stats.bytes_in += (self.in_codec.wire_numel
-                                        * dtype_bytes(self.in_codec.wire_dtype))
-            if r < last:
-                self.stats.bytes_out += (self.out_codec.wire_numel
-                                         * dtype_bytes(self.out_codec.wire_dtype))
 
         # drain outstanding sends
         for ring in (self.send_ring, self.result_ring):
             if ring is not None:
-                for w in ring.works:
+                for w in list(ring.works) + list(
+                        getattr(ring, "size_works", [])):
                     if w is not None:
                         w.wait()

@@ -20,7 +20,7 @@ CSRC = ROOT / "defer_amd" / "csrc"
 
 sources = [str(CSRC / f) for f in
            ["bindings.cpp", "conv.hip", "elementwise.hip", "pool.hip",
-            "codec.hip"]
+            "codec.hip", "lz4.hip"]
            if (CSRC / f).exists()]
 
 setup(

@@ -51,7 +51,8 @@ def _worker(rank, world, port, q, cuts, steps, compression):
 def _run(world, cuts, steps=4, compression="none", tol=0.0):
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    port = 29501 + world + (7 if compression != "none" else 0)
+    port = (29501 + world
+            + {"none": 0, "zfp": 7, "zfp+lz4": 11}[compression])
     procs = [ctx.Process(target=_worker,
                          args=(r, world, port, q, cuts, steps, compression))
              for r in range(world)]
@@ -87,3 +88,11 @@ def test_dist_pipeline_with_zfp_compression():
     Lossy codec -> compare against the uncompressed model with tolerance.
     """
     _run(2, ["add_8"], steps=2, compression="zfp", tol=0.05)
+
+
+def test_dist_pipeline_with_zfp_lz4_compression():
+    """Full reference wire codec lz4(zfp(x)) (dispatcher.py:81-84) over a
+    variable-size hop (size message + exact payload, the RCCL analogue of
+    the reference's length-prefixed framing). LZ4 is lossless on top of
+    ZFP, so the tolerance equals the ZFP-only case."""
+    _run(2, ["add_8"], steps=3, compression="zfp+lz4", tol=0.05)
