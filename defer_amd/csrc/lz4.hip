@@ -62,6 +62,9 @@ lz4_compress_kernel(const u8* __restrict__ in, long n,
         // lane 0 scans for the next match (or end of block)
         int litFrom = 0, litLen = 0, moff = 0, mlen = 0, done = 0;
         if (lane == 0) {
+            // skip acceleration (mirrors ops/lz4_ref.py): after every 64
+            // failed probes the scan step grows by one
+            int cnt = 64;
             for (;;) {
                 if (pos >= len - 8) {  // tail: emit final literals
                     litFrom = anchor; litLen = len - anchor;
@@ -83,7 +86,8 @@ lz4_compress_kernel(const u8* __restrict__ in, long n,
                     anchor = pos;
                     break;
                 }
-                ++pos;
+                pos += cnt >> 6;
+                ++cnt;
             }
         }
         litFrom = __shfl(litFrom, 0);

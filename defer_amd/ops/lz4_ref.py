@@ -67,6 +67,9 @@ def _compress_block(buf: bytes) -> bytes:
                     break
                 rem -= 255
 
+    # LZ4-style skip acceleration: after every 64 failed probes the scan
+    # step grows by one, bounding worst-case work on incompressible data.
+    cnt = 64
     while pos < n - 8:
         v = read32(pos)
         h = _hash(v)
@@ -79,8 +82,10 @@ def _compress_block(buf: bytes) -> bytes:
             emit(anchor, pos - anchor, pos - cand, mlen)
             pos += mlen
             anchor = pos
+            cnt = 64
         else:
-            pos += 1
+            pos += cnt >> 6
+            cnt += 1
     emit(anchor, n - anchor, 0, 0)    # final literals-only sequence
     return bytes(out)
 

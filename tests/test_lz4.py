@@ -84,5 +84,6 @@ def test_gpu_roundtrip_on_zfp_payload():
     w2 = codec.lz4_decompress(c, w.numel())
     assert torch.equal(w, w2)
     y = codec.zfp_decode(w2, x.shape, rate, dtype=torch.bfloat16)
-    err = (y.float() - x.float()).abs().max().item()
-    assert err < 0.1, err
+    rel = ((y.float() - x.float()).abs().max()
+           / x.float().abs().max()).item()
+    assert rel < 0.15, rel
