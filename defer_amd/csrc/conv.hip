@@ -17,11 +17,18 @@
 // (grid-stride) so the pipeline never drains between tiles; when the
 // whole K fits one tile, the weight tile is staged once (B_PERSIST).
 //
-// Tile configs (picked per shape by the launcher):
-//   BM128/BN64/D3  — default; small-Cout or modest grids
-//   BM128/BN128/D2 — Cout>=128 with plenty of m-tiles: halves A
-//                    re-staging across n-blocks, 2x MFMA per staged byte
-//   BM64 variants  — small-M shapes (7x7/14x14 batches): 2x the blocks
+// Address math is kept off the k-loop critical path:
+//   - the m -> (batch, oh, ow) decomposition runs once per STAGED m-tile
+//     (magic-multiply division, no v_rcp chains), cached per chunk;
+//   - AMODE_CONV derives (c, r, s) from k with two umulhi magics;
+//   - AMODE_RSC (Cin % 64 == 0, i.e. every non-stem conv) walks K in
+//     (cin64-block, r, s)-major order so one k-tile is a single scalar
+//     (r, s, cb) triple — the per-lane work is two adds and a bounds
+//     check. The B gather applies the same permutation to the OHWI
+//     weights in-flight, so A and B agree on the K order and no host
+//     repack is needed.
+// The fused epilogue runs OUTSIDE the k-loop (between tiles) so the hot
+// loop carries no exec-masked epilogue code.
 //
 // Epilogue fuses folded-BN scale/bias, residual add and ReLU (the Keras
 // Conv2D+BN+Add+ReLU stack the reference runs via model.predict,
@@ -39,9 +46,16 @@ using defer_hip::ConvParams;
 #define NTHREADS 256
 #define KCH (BK / 8)          // 16-B chunks per tile row (8)
 
+enum { AMODE_GEMM = 0, AMODE_CONV = 1, AMODE_RSC = 2 };
+
 // XOR swizzle: logical (row, k8) lives at physical k8p = k8 ^ (row & 7).
 __device__ __forceinline__ int swz(int row, int k8) {
     return k8 ^ (row & 7);
+}
+
+// floor(n / d) for n*d < 2^32, via mul = ceil(2^32 / d) (d > 1).
+__device__ __forceinline__ u32 umagic(u32 n, u32 mul, u32 d) {
+    return d == 1 ? n : __umulhi(n, mul);
 }
 
 __device__ __forceinline__ void glds16(const bf16* src, bf16* lds_base) {
@@ -51,7 +65,7 @@ __device__ __forceinline__ void glds16(const bf16* src, bf16* lds_base) {
         16, 0, 0);
 }
 
-template <int ACT, bool HAS_RES, bool GEMM_MODE, bool B_PERSIST, int BM,
+template <int ACT, bool HAS_RES, int AMODE, bool B_PERSIST, int BM,
           int BN, int DEPTH>
 __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
     ConvParams p) {
@@ -116,47 +130,83 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
         bi[ni] = BIAS[n];
     }
 
-    auto stage_a = [&](int mt, int kt, int buf) {
-        const int k0 = kt * BK;
+    // ---- per-chunk A state for the STAGED m-tile (refreshed when the
+    // stage cursor crosses into a new m-tile — once per nk k-tiles)
+    long a_base[ACH];
+    int a_ihb[ACH], a_iwb[ACH];
+    bool a_mval[ACH];
+    auto a_setup = [&](int mt) {
         const int m0 = mt * BM;
+#pragma unroll
+        for (int i = 0; i < ACH; ++i) {
+            int m = m0 + a_row[i];
+            a_mval[i] = m < p.M;
+            if (m >= p.M) m = p.M - 1;
+            if (AMODE == AMODE_GEMM) {
+                a_base[i] = (long)m * p.K + a_k8[i] * 8;
+            } else {
+                u32 t = umagic(m, p.owmul, p.OW);       // m / OW
+                int ow = m - (int)t * p.OW;
+                u32 nb = umagic(t, p.ohmul, p.OH);      // t / OH
+                int oh = (int)t - (int)nb * p.OH;
+                a_ihb[i] = oh * p.stride - p.pad;
+                a_iwb[i] = ow * p.stride - p.pad;
+                a_base[i] = (long)nb * p.H * p.W * p.Cin;
+            }
+        }
+    };
+
+    // stage cursor scalars (RSC walks (cb, r, s) incrementally — the
+    // k-tile order IS (cb*R*S + r*S + s), no division anywhere)
+    int s_mt = blockIdx.x, s_kt = 0;
+    int s_cb = 0, s_r = 0, s_s = 0;
+
+    auto stage_a = [&](int buf) {
         bf16* A = A0 + buf * BM * BK;
 #pragma unroll
         for (int i = 0; i < ACH; ++i) {
-            int k = k0 + a_k8[i] * 8;
-            int m = m0 + a_row[i];
             const bf16* src = Z;
-            if (k < p.K && m < p.M) {
-                if (GEMM_MODE) {
-                    src = X + (long)m * p.K + k;
-                } else {
-                    int ow = m % p.OW;
-                    int t = m / p.OW;
-                    int oh = t % p.OH;
-                    int nb = t / p.OH;
-                    int c = k % p.Cin;
-                    int rs = k / p.Cin;
-                    int r = rs / p.S;
-                    int s = rs % p.S;
-                    int ih = oh * p.stride - p.pad + r;
-                    int iw = ow * p.stride - p.pad + s;
-                    if (ih >= 0 && ih < p.H && iw >= 0 && iw < p.W)
-                        src = X + (((long)nb * p.H + ih) * p.W + iw)
-                                    * p.Cin + c;
-                }
+            if (AMODE == AMODE_GEMM) {
+                int k = s_kt * BK + a_k8[i] * 8;
+                if (a_mval[i] && k < p.K)
+                    src = X + a_base[i] + (long)s_kt * BK;
+            } else if (AMODE == AMODE_CONV) {
+                int k = s_kt * BK + a_k8[i] * 8;
+                u32 rs = umagic(k, p.cmul, p.Cin);      // k / Cin
+                int c = k - (int)rs * p.Cin;
+                u32 r = umagic(rs, p.smul, p.S);        // rs / S
+                int sI = (int)rs - (int)r * p.S;
+                int ih = a_ihb[i] + (int)r;
+                int iw = a_iwb[i] + sI;
+                if (a_mval[i] && k < p.K && (u32)ih < (u32)p.H &&
+                    (u32)iw < (u32)p.W)
+                    src = X + a_base[i] + ((long)ih * p.W + iw) * p.Cin
+                              + c;
+            } else {  // AMODE_RSC
+                int ih = a_ihb[i] + s_r;
+                int iw = a_iwb[i] + s_s;
+                if (a_mval[i] && (u32)ih < (u32)p.H && (u32)iw < (u32)p.W)
+                    src = X + a_base[i] + ((long)ih * p.W + iw) * p.Cin
+                              + s_cb * 64 + a_k8[i] * 8;
             }
             glds16(src, A + (wave * (ACH * 64) + i * 64) * 8);
         }
     };
-    auto stage_b = [&](int kt, int buf) {
-        const int k0 = kt * BK;
+    auto stage_b = [&](int buf) {
         bf16* B = B0 + buf * BN * BK;
+        // RSC: same K permutation as A, applied to the OHWI weights
+        const long koff = (AMODE == AMODE_RSC)
+                              ? (long)(s_r * p.S + s_s) * p.Cin
+                                    + s_cb * 64
+                              : (long)s_kt * BK;
 #pragma unroll
         for (int i = 0; i < BCH; ++i) {
-            int k = k0 + b_k8[i] * 8;
             int n = n0 + b_row[i];
-            const bf16* src = (k < p.K && n < p.Cout)
-                                  ? Wt + (long)n * p.K + k
-                                  : Z;
+            long k = koff + b_k8[i] * 8;
+            const bf16* src =
+                (n < p.Cout && (AMODE == AMODE_RSC || k < p.K))
+                    ? Wt + (long)n * p.K + k
+                    : Z;
             glds16(src, B + (wave * (BCH * 64) + i * 64) * 8);
         }
     };
@@ -279,17 +329,27 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
     // each iteration keeps the cursor DEPTH-1 tiles ahead of compute
     // (the refilled buffer was last read at compute(it-(DEPTH-1)), which
     // every wave finished before this iteration's barrier).
-    int s_mt = blockIdx.x, s_kt = 0;
     auto advance = [&]() {
-        if (++s_kt == nk) { s_kt = 0; s_mt += gridDim.x; }
+        if (++s_kt == nk) {
+            s_kt = 0;
+            s_mt += gridDim.x;
+            s_cb = 0; s_r = 0; s_s = 0;
+            if (s_mt < mtiles) a_setup(s_mt);
+        } else if (AMODE == AMODE_RSC) {
+            if (++s_s == p.S) {
+                s_s = 0;
+                if (++s_r == p.R) { s_r = 0; ++s_cb; }
+            }
+        }
     };
     auto stage = [&](int buf) {
-        stage_a(s_mt, s_kt, buf);
-        if (!B_PERSIST) stage_b(s_kt, buf);
+        stage_a(buf);
+        if (!B_PERSIST) stage_b(buf);
         advance();
     };
 
-    if (B_PERSIST) stage_b(0, 0);   // oldest ops: drained by first wait
+    if (s_mt < mtiles) a_setup(s_mt);
+    if (B_PERSIST) stage_b(0);      // oldest ops: drained by first wait
     int staged = 0;
 #pragma unroll
     for (int d = 0; d < DEPTH - 1; ++d)
@@ -311,12 +371,15 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
                 asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
             __builtin_amdgcn_s_barrier();
             compute(it % DEPTH, B_PERSIST ? 0 : it % DEPTH);
-            if (kt == nk - 1) epilogue(mt, it % DEPTH);
             if (s_mt < mtiles) {
                 stage((it + DEPTH - 1) % DEPTH);
                 ++staged;
             }
         }
+        // epilogue AFTER the next tiles' stages were issued: their glds
+        // land in buffers (it..it+DEPTH-2)%DEPTH, disjoint from the
+        // scratch buffer (it-1)%DEPTH this writes
+        epilogue(mt, (it - 1) % DEPTH);
     }
 }
 
@@ -346,17 +409,26 @@ __global__ void pad_channels_kernel(const bf16* __restrict__ x,
 // host launchers
 namespace defer_hip {
 
-void launch_conv_igemm(const ConvParams& p, bool relu, bool has_res,
+static u32 magic32(u32 d) {          // ceil(2^32 / d), d > 1
+    return d <= 1 ? 0u : (u32)(((1ull << 32) + d - 1) / d);
+}
+
+void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
                        bool gemm_mode, hipStream_t s) {
+    ConvParams p = p0;
+    p.owmul = magic32((u32)p.OW);
+    p.ohmul = magic32((u32)p.OH);
+    p.cmul = magic32((u32)p.Cin);
+    p.smul = magic32((u32)p.S);
     const int nk = (p.K + BK - 1) / BK;
     const bool bp = (nk == 1);
-    // tile selection. BN128/D2 halves A re-staging and doubles MFMA per
-    // staged byte but its 2-deep pipeline waits vmcnt(0) per K-tile, so
-    // it only pays on deep-K (3x3) shapes; memory-bound small-K shapes
-    // keep the 3-deep BN64 pipeline. BM=64 for small-M shapes (more
-    // blocks on the 256 CUs).
+    const bool rsc = !gemm_mode && !bp && (p.Cin % 64 == 0);
+    const int amode = gemm_mode ? AMODE_GEMM
+                                : (rsc ? AMODE_RSC : AMODE_CONV);
+    // tile selection. BN128 halves A re-staging and doubles MFMA per
+    // staged byte; memory-bound small-K shapes keep the 3-deep BN64
+    // pipeline. BM=64 for small-M shapes (more blocks on the 256 CUs).
     const int mt128 = (p.M + 127) / 128;
-    const int mt64 = (p.M + 63) / 64;
     int BMsel, BNsel;
     const bool deepK = p.K >= 1024;
     if (deepK && p.Cout >= 128) {
@@ -385,54 +457,66 @@ void launch_conv_igemm(const ConvParams& p, bool relu, bool has_res,
     do {                                                                  \
         if (relu) {                                                       \
             if (has_res) {                                                \
-                if (gemm_mode) {                                          \
-                    if (bp) DISPATCH_TILE(ACT_RELU, true, true, true,     \
-                                          BMv, BNv, D);                   \
-                    else DISPATCH_TILE(ACT_RELU, true, true, false, BMv,  \
-                                       BNv, D);                           \
+                if (amode == AMODE_GEMM) {                                \
+                    if (bp) DISPATCH_TILE(ACT_RELU, true, AMODE_GEMM,     \
+                                          true, BMv, BNv, D);             \
+                    else DISPATCH_TILE(ACT_RELU, true, AMODE_GEMM,        \
+                                       false, BMv, BNv, D);               \
+                } else if (amode == AMODE_RSC) {                          \
+                    DISPATCH_TILE(ACT_RELU, true, AMODE_RSC, false, BMv,  \
+                                  BNv, D);                                \
                 } else {                                                  \
-                    if (bp) DISPATCH_TILE(ACT_RELU, true, false, true,    \
-                                          BMv, BNv, D);                   \
-                    else DISPATCH_TILE(ACT_RELU, true, false, false, BMv, \
-                                       BNv, D);                           \
+                    if (bp) DISPATCH_TILE(ACT_RELU, true, AMODE_CONV,     \
+                                          true, BMv, BNv, D);             \
+                    else DISPATCH_TILE(ACT_RELU, true, AMODE_CONV,       \
+                                       false, BMv, BNv, D);               \
                 }                                                         \
             } else {                                                      \
-                if (gemm_mode) {                                          \
-                    if (bp) DISPATCH_TILE(ACT_RELU, false, true, true,    \
-                                          BMv, BNv, D);                   \
-                    else DISPATCH_TILE(ACT_RELU, false, true, false, BMv, \
-                                       BNv, D);                           \
+                if (amode == AMODE_GEMM) {                                \
+                    if (bp) DISPATCH_TILE(ACT_RELU, false, AMODE_GEMM,    \
+                                          true, BMv, BNv, D);             \
+                    else DISPATCH_TILE(ACT_RELU, false, AMODE_GEMM,       \
+                                       false, BMv, BNv, D);               \
+                } else if (amode == AMODE_RSC) {                          \
+                    DISPATCH_TILE(ACT_RELU, false, AMODE_RSC, false,      \
+                                  BMv, BNv, D);                           \
                 } else {                                                  \
-                    if (bp) DISPATCH_TILE(ACT_RELU, false, false, true,   \
-                                          BMv, BNv, D);                   \
-                    else DISPATCH_TILE(ACT_RELU, false, false, false,     \
-                                       BMv, BNv, D);                      \
+                    if (bp) DISPATCH_TILE(ACT_RELU, false, AMODE_CONV,    \
+                                          true, BMv, BNv, D);             \
+                    else DISPATCH_TILE(ACT_RELU, false, AMODE_CONV,       \
+                                       false, BMv, BNv, D);               \
                 }                                                         \
             }                                                             \
         } else {                                                          \
             if (has_res) {                                                \
-                if (gemm_mode) {                                          \
-                    if (bp) DISPATCH_TILE(ACT_NONE, true, true, true,     \
-                                          BMv, BNv, D);                   \
-                    else DISPATCH_TILE(ACT_NONE, true, true, false, BMv,  \
-                                       BNv, D);                           \
+                if (amode == AMODE_GEMM) {                                \
+                    if (bp) DISPATCH_TILE(ACT_NONE, true, AMODE_GEMM,     \
+                                          true, BMv, BNv, D);             \
+                    else DISPATCH_TILE(ACT_NONE, true, AMODE_GEMM,        \
+                                       false, BMv, BNv, D);               \
+                } else if (amode == AMODE_RSC) {                          \
+                    DISPATCH_TILE(ACT_NONE, true, AMODE_RSC, false, BMv,  \
+                                  BNv, D);                                \
                 } else {                                                  \
-                    if (bp) DISPATCH_TILE(ACT_NONE, true, false, true,    \
-                                          BMv, BNv, D);                   \
-                    else DISPATCH_TILE(ACT_NONE, true, false, false, BMv, \
-                                       BNv, D);                           \
+                    if (bp) DISPATCH_TILE(ACT_NONE, true, AMODE_CONV,     \
+                                          true, BMv, BNv, D);             \
+                    else DISPATCH_TILE(ACT_NONE, true, AMODE_CONV,        \
+                                       false, BMv, BNv, D);               \
                 }                                                         \
             } else {                                                      \
-                if (gemm_mode) {                                          \
-                    if (bp) DISPATCH_TILE(ACT_NONE, false, true, true,    \
-                                          BMv, BNv, D);                   \
-                    else DISPATCH_TILE(ACT_NONE, false, true, false, BMv, \
-                                       BNv, D);                           \
+                if (amode == AMODE_GEMM) {                                \
+                    if (bp) DISPATCH_TILE(ACT_NONE, false, AMODE_GEMM,    \
+                                          true, BMv, BNv, D);             \
+                    else DISPATCH_TILE(ACT_NONE, false, AMODE_GEMM,       \
+                                       false, BMv, BNv, D);               \
+                } else if (amode == AMODE_RSC) {                          \
+                    DISPATCH_TILE(ACT_NONE, false, AMODE_RSC, false,      \
+                                  BMv, BNv, D);                           \
                 } else {                                                  \
-                    if (bp) DISPATCH_TILE(ACT_NONE, false, false, true,   \
-                                          BMv, BNv, D);                   \
-                    else DISPATCH_TILE(ACT_NONE, false, false, false,     \
-                                       BMv, BNv, D);                      \
+                    if (bp) DISPATCH_TILE(ACT_NONE, false, AMODE_CONV,    \
+                                          true, BMv, BNv, D);             \
+                    else DISPATCH_TILE(ACT_NONE, false, AMODE_CONV,       \
+                                       false, BMv, BNv, D);               \
                 }                                                         \
             }                                                             \
         }                                                                 \
@@ -440,7 +524,7 @@ void launch_conv_igemm(const ConvParams& p, bool relu, bool has_res,
 
     if (BMsel == 128 && BNsel == 64) DISPATCH_BOOLS(128, 64, 3);
     else if (BMsel == 128 && BNsel == 128) DISPATCH_BOOLS(128, 128, 2);
-    else if (BMsel == 64 && BNsel == 128) DISPATCH_BOOLS(64, 128, 2);
+    else if (BMsel == 64 && BNsel == 128) DISPATCH_BOOLS(64, 128, 3);
     else DISPATCH_BOOLS(64, 64, 3);
 #undef DISPATCH_BOOLS
 #undef DISPATCH_TILE

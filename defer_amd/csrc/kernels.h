@@ -18,6 +18,9 @@ struct ConvParams {
     int M, K, Cout;
     int NB, H, W, Cin;
     int OH, OW, R, S, stride, pad;
+    // magic-multiply reciprocals (filled by launch_conv_igemm):
+    // floor(n/d) = umulhi(n, ceil(2^32/d)) for n*d < 2^32
+    unsigned int owmul, ohmul, cmul, smul;
 };
 
 void launch_conv_igemm(const ConvParams& p, bool relu, bool has_res,
