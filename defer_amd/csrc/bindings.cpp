@@ -85,9 +85,42 @@ Tensor conv2d_bn_act(Tensor x, Tensor w, c10::optional<Tensor> scale,
     p.Cout = Cout;
 
     if (Cin % 8 != 0) {
-        // stem path (Cin=3): zero-pad channels to 8 and take the regular
-        // implicit-GEMM path (K grows R*S*Cin -> R*S*8; the stem is ~2% of
-        // ResNet50 FLOPs, and padding beats an explicit im2col by ~4x).
+        if ((stride * Cin) % 2 == 0) {
+            // STEM path (ResNet 7x7/s2, Cin=3): spatially pre-pad the
+            // raw input and gather K as (r, run)-major where a run is
+            // TR = ceil(S*Cin/8)*8 CONTIGUOUS bf16 of one padded input
+            // row (S*Cin real + a few next-pixel values that multiply
+            // zero weights). K drops from R*S*pad8(Cin) (392) to R*TR
+            // (168) and the gather needs no validity math at all.
+            int TR = (S * Cin + 7) / 8 * 8;
+            int Keff = R * TR;
+            int nk = (Keff + 63) / 64;
+            int rmax = (nk * 64 - 1) / TR;          // last k row touched
+            int PW = std::max(W + 2 * (int)pad,
+                              (OW - 1) * (int)stride
+                                  + (TR + Cin - 1) / Cin);
+            int PH = std::max(H + 2 * (int)pad,
+                              (OH - 1) * (int)stride + rmax + 1);
+            if (((long)PW * Cin) % 2) PW += 1;      // 4-B chunk alignment
+            auto xp = at::empty({NB, PH, PW, Cin}, x.options());
+            defer_hip::launch_pad2d(bptr(x), bptr_mut(xp), NB, H, W, Cin,
+                                    PH, PW, (int)pad, (int)pad, s);
+            auto wp = at::empty({Cout, Keff}, w.options());
+            defer_hip::launch_stem_repack_w(bptr(w), bptr_mut(wp), Cout,
+                                            R, S, Cin, TR, s);
+            p.x = bptr(xp);
+            p.w = bptr(wp);
+            p.K = Keff;
+            p.NB = NB; p.H = PH; p.W = PW; p.Cin = Cin;
+            p.OH = OH; p.OW = OW; p.R = R;
+            p.S = TR;                   // S carries the run length TR
+            p.stride = (int)stride; p.pad = 0;
+            defer_hip::launch_conv_igemm(p, relu, (bool)res, false, true,
+                                         s);
+            return out;
+        }
+        // odd stride*Cin (e.g. 3x3/s1 VGG stem): zero-pad channels to 8
+        // and take the regular implicit-GEMM path.
         int C8 = (Cin + 7) / 8 * 8;
         auto xp = at::empty({NB, H, W, C8}, x.options());
         defer_hip::launch_pad_channels(bptr(x), bptr_mut(xp),
@@ -101,7 +134,7 @@ Tensor conv2d_bn_act(Tensor x, Tensor w, c10::optional<Tensor> scale,
         p.NB = NB; p.H = H; p.W = W; p.Cin = C8;
         p.OH = OH; p.OW = OW; p.R = R; p.S = S;
         p.stride = (int)stride; p.pad = (int)pad;
-        defer_hip::launch_conv_igemm(p, relu, (bool)res, false, s);
+        defer_hip::launch_conv_igemm(p, relu, (bool)res, false, false, s);
         return out;
     }
 
@@ -116,7 +149,7 @@ Tensor conv2d_bn_act(Tensor x, Tensor w, c10::optional<Tensor> scale,
         p.NB = (int)M; p.H = 1; p.W = 1; p.Cin = p.K;
         p.OH = 1; p.OW = 1;
     }
-    defer_hip::launch_conv_igemm(p, relu, (bool)res, gemm_mode, s);
+    defer_hip::launch_conv_igemm(p, relu, (bool)res, gemm_mode, false, s);
     return out;
 }
 
@@ -138,7 +171,7 @@ Tensor linear(Tensor x, Tensor w, c10::optional<Tensor> bias) {
     p.M = M; p.K = K; p.Cout = N;
     p.NB = M; p.H = 1; p.W = 1; p.Cin = K;
     p.OH = 1; p.OW = 1; p.R = 1; p.S = 1; p.stride = 1; p.pad = 0;
-    defer_hip::launch_conv_igemm(p, false, false, true, cur_stream());
+    defer_hip::launch_conv_igemm(p, false, false, true, false, cur_stream());
     return out;
 }
 

@@ -46,7 +46,7 @@ using defer_hip::ConvParams;
 #define NTHREADS 256
 #define KCH (BK / 8)          // 16-B chunks per tile row (8)
 
-enum { AMODE_GEMM = 0, AMODE_CONV = 1, AMODE_RSC = 2 };
+enum { AMODE_GEMM = 0, AMODE_CONV = 1, AMODE_RSC = 2, AMODE_STEM = 3 };
 
 // XOR swizzle: logical (row, k8) lives at physical k8p = k8 ^ (row & 7).
 __device__ __forceinline__ int swz(int row, int k8) {
@@ -144,6 +144,16 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
             if (m >= p.M) m = p.M - 1;
             if (AMODE == AMODE_GEMM) {
                 a_base[i] = (long)m * p.K + a_k8[i] * 8;
+            } else if (AMODE == AMODE_STEM) {
+                // spatially pre-padded input (pad absorbed into coords);
+                // p.H/p.W are the PADDED dims, base points at the
+                // window's first row run
+                u32 t = umagic(m, p.owmul, p.OW);       // m / OW
+                int ow = m - (int)t * p.OW;
+                u32 nb = umagic(t, p.ohmul, p.OH);      // t / OH
+                int oh = (int)t - (int)nb * p.OH;
+                a_base[i] = ((long)nb * p.H + oh * p.stride) * p.W * p.Cin
+                            + (long)(ow * p.stride) * p.Cin;
             } else {
                 u32 t = umagic(m, p.owmul, p.OW);       // m / OW
                 int ow = m - (int)t * p.OW;
@@ -182,6 +192,16 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
                     (u32)iw < (u32)p.W)
                     src = X + a_base[i] + ((long)ih * p.W + iw) * p.Cin
                               + c;
+            } else if (AMODE == AMODE_STEM) {
+                // K walks (r, run) with run = TR bytes of one padded
+                // input row (p.S carries TR; all coords in-bounds by
+                // construction, so no validity math at all)
+                int k = s_kt * BK + a_k8[i] * 8;
+                u32 r = umagic(k, p.smul, p.S);         // k / TR
+                int t = k - (int)r * p.S;
+                if (a_mval[i])
+                    src = X + a_base[i]
+                              + (long)r * p.W * p.Cin + t;
             } else {  // AMODE_RSC
                 int ih = a_ihb[i] + s_r;
                 int iw = a_iwb[i] + s_s;
@@ -406,6 +426,49 @@ __global__ void pad_channels_kernel(const bf16* __restrict__ x,
 }
 
 // ---------------------------------------------------------------------------
+// Spatial zero-pad NHWC: [NB,H,W,C] -> [NB,PH,PW,C], interior at
+// (ph0, pw0). Feeds the STEM gather (pad absorbed into coordinates so the
+// k-loop runs with zero validity math).
+__global__ void pad2d_kernel(const bf16* __restrict__ x,
+                             bf16* __restrict__ y, int NB, int H, int W,
+                             int C, int PH, int PW, int ph0, int pw0) {
+    const int rowlen = PW * C;
+    for (int row = blockIdx.x; row < NB * PH; row += gridDim.x) {
+        int ph = row % PH, nb = row / PH;
+        int ih = ph - ph0;
+        bf16* dst = y + (long)row * rowlen;
+        if ((unsigned)ih >= (unsigned)H) {
+            for (int j = threadIdx.x; j < rowlen; j += blockDim.x)
+                dst[j] = (bf16)0.f;
+        } else {
+            const bf16* src = x + ((long)nb * H + ih) * W * C;
+            const int i0 = pw0 * C, i1 = i0 + W * C;
+            for (int j = threadIdx.x; j < rowlen; j += blockDim.x)
+                dst[j] = (j >= i0 && j < i1) ? src[j - i0] : (bf16)0.f;
+        }
+    }
+}
+
+// Repack OHWI weights for the STEM K-order: wp[n][r*TR + s*C + c] =
+// w[n][r][s][c], zero in the run padding (t >= S*C).
+__global__ void stem_repack_w_kernel(const bf16* __restrict__ w,
+                                     bf16* __restrict__ wp, int Cout,
+                                     int R, int S, int C, int TR) {
+    long total = (long)Cout * R * TR;
+    long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long gs = (long)gridDim.x * blockDim.x;
+    for (long i = i0; i < total; i += gs) {
+        int t = (int)(i % TR);
+        long u = i / TR;
+        int r = (int)(u % R);
+        int n = (int)(u / R);
+        bf16 v = (bf16)0.f;
+        if (t < S * C) v = w[((long)(n * R + r) * S + t / C) * C + t % C];
+        wp[i] = v;
+    }
+}
+
+// ---------------------------------------------------------------------------
 // host launchers
 namespace defer_hip {
 
@@ -414,17 +477,19 @@ static u32 magic32(u32 d) {          // ceil(2^32 / d), d > 1
 }
 
 void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
-                       bool gemm_mode, hipStream_t s) {
+                       bool gemm_mode, bool stem_mode, hipStream_t s) {
     ConvParams p = p0;
     p.owmul = magic32((u32)p.OW);
     p.ohmul = magic32((u32)p.OH);
     p.cmul = magic32((u32)p.Cin);
     p.smul = magic32((u32)p.S);
     const int nk = (p.K + BK - 1) / BK;
-    const bool bp = (nk == 1);
-    const bool rsc = !gemm_mode && !bp && (p.Cin % 64 == 0);
-    const int amode = gemm_mode ? AMODE_GEMM
-                                : (rsc ? AMODE_RSC : AMODE_CONV);
+    const bool bp = (nk == 1) && !stem_mode;
+    const bool rsc = !gemm_mode && !stem_mode && !bp && (p.Cin % 64 == 0);
+    const int amode = stem_mode ? AMODE_STEM
+                                : gemm_mode ? AMODE_GEMM
+                                            : (rsc ? AMODE_RSC
+                                                   : AMODE_CONV);
     // tile selection. BN128 halves A re-staging and doubles MFMA per
     // staged byte; memory-bound small-K shapes keep the 3-deep BN64
     // pipeline. BM=64 for small-M shapes (more blocks on the 256 CUs).
@@ -465,6 +530,9 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
                 } else if (amode == AMODE_RSC) {                          \
                     DISPATCH_TILE(ACT_RELU, true, AMODE_RSC, false, BMv,  \
                                   BNv, D);                                \
+                } else if (amode == AMODE_STEM) {                       \
+                    DISPATCH_TILE(ACT_RELU, true, AMODE_STEM, false,     \
+                                  BMv, BNv, D);                       \
                 } else {                                                  \
                     if (bp) DISPATCH_TILE(ACT_RELU, true, AMODE_CONV,     \
                                           true, BMv, BNv, D);             \
@@ -480,6 +548,9 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
                 } else if (amode == AMODE_RSC) {                          \
                     DISPATCH_TILE(ACT_RELU, false, AMODE_RSC, false,      \
                                   BMv, BNv, D);                           \
+                } else if (amode == AMODE_STEM) {                       \
+                    DISPATCH_TILE(ACT_RELU, false, AMODE_STEM, false,     \
+                                  BMv, BNv, D);                       \
                 } else {                                                  \
                     if (bp) DISPATCH_TILE(ACT_RELU, false, AMODE_CONV,    \
                                           true, BMv, BNv, D);             \
@@ -497,6 +568,9 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
                 } else if (amode == AMODE_RSC) {                          \
                     DISPATCH_TILE(ACT_NONE, true, AMODE_RSC, false, BMv,  \
                                   BNv, D);                                \
+                } else if (amode == AMODE_STEM) {                       \
+                    DISPATCH_TILE(ACT_NONE, true, AMODE_STEM, false,     \
+                                  BMv, BNv, D);                       \
                 } else {                                                  \
                     if (bp) DISPATCH_TILE(ACT_NONE, true, AMODE_CONV,     \
                                           true, BMv, BNv, D);             \
@@ -512,6 +586,9 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
                 } else if (amode == AMODE_RSC) {                          \
                     DISPATCH_TILE(ACT_NONE, false, AMODE_RSC, false,      \
                                   BMv, BNv, D);                           \
+                } else if (amode == AMODE_STEM) {                       \
+                    DISPATCH_TILE(ACT_NONE, false, AMODE_STEM, false,     \
+                                  BMv, BNv, D);                       \
                 } else {                                                  \
                     if (bp) DISPATCH_TILE(ACT_NONE, false, AMODE_CONV,    \
                                           true, BMv, BNv, D);             \
@@ -540,6 +617,23 @@ void launch_pad_channels(const void* x, void* y, long rows, int C, int C8,
     hipLaunchKernelGGL(pad_channels_kernel,
                        dim3(grid1d(rows * (C8 / 8), 256)), dim3(256), 0, s,
                        (const bf16*)x, (bf16*)y, rows, C, C8);
+}
+
+void launch_pad2d(const void* x, void* y, int NB, int H, int W, int C,
+                  int PH, int PW, int ph0, int pw0, hipStream_t s) {
+    int rows = NB * PH;
+    hipLaunchKernelGGL(pad2d_kernel,
+                       dim3(rows < 2048 ? rows : 2048), dim3(256), 0, s,
+                       (const bf16*)x, (bf16*)y, NB, H, W, C, PH, PW,
+                       ph0, pw0);
+}
+
+void launch_stem_repack_w(const void* w, void* wp, int Cout, int R, int S,
+                          int C, int TR, hipStream_t s) {
+    hipLaunchKernelGGL(stem_repack_w_kernel,
+                       dim3(grid1d((long)Cout * R * TR, 256)), dim3(256),
+                       0, s, (const bf16*)w, (bf16*)wp, Cout, R, S, C,
+                       TR);
 }
 
 }  // namespace defer_hip

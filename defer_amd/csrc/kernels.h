@@ -24,9 +24,14 @@ struct ConvParams {
 };
 
 void launch_conv_igemm(const ConvParams& p, bool relu, bool has_res,
-                       bool gemm_mode, hipStream_t s);
+                       bool gemm_mode, bool stem_mode, hipStream_t s);
 void launch_pad_channels(const void* x, void* y, long rows, int C, int C8,
                          hipStream_t s);
+// STEM path helpers: spatial zero-pad + weight repack to (r, run) K-order
+void launch_pad2d(const void* x, void* y, int NB, int H, int W, int C,
+                  int PH, int PW, int ph0, int pw0, hipStream_t s);
+void launch_stem_repack_w(const void* w, void* wp, int Cout, int R, int S,
+                          int C, int TR, hipStream_t s);
 
 void launch_bn_act(const void* x, const float* scale, const float* bias,
                    void* y, long total8, int c8, bool relu, hipStream_t s);
