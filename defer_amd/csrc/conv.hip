@@ -288,35 +288,42 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
             asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
             __builtin_amdgcn_s_barrier();
             const bool interior =
-                (mt * BM + BM <= p.M) && (n0 + BN <= p.Cout);
+                (mt * BM + BM <= p.M) && (n0 + BN <= p.Cout) &&
+                (p.Cout % 8 == 0);     // 16-B aligned rows
             if (interior) {
-                f32x4 v4[CPR];
-                bf16x4 rv[CPR];
-                long off[CPR];
+                // 8-col chunks: 16-B residual loads and 16-B bf16 stores
+                // (8-B stores made the 1x1 layers store-issue-bound, cf.
+                // MI355X_MICROARCH store-tail note)
+                constexpr int CP8 = RH * BN / 8 / NTHREADS;
+                f32x4 v4[CP8][2];
+                bf16x8 rv[CP8];
+                long off[CP8];
 #pragma unroll
-                for (int i = 0; i < CPR; ++i) {
+                for (int i = 0; i < CP8; ++i) {
                     int chunk = tid + i * NTHREADS;
-                    int rl = chunk / (BN / 4);
-                    int c4 = (chunk % (BN / 4)) * 4;
-                    int cs = c4 ^ (((rl >> 2) & 3) << 4);
-                    off[i] = (long)(mt * BM + r0 + rl) * p.Cout + n0 + c4;
-                    v4[i] = *reinterpret_cast<f32x4*>(
+                    int rl = chunk / (BN / 8);
+                    int c8 = (chunk % (BN / 8)) * 8;
+                    int cs = c8 ^ (((rl >> 2) & 3) << 4);
+                    off[i] = (long)(mt * BM + r0 + rl) * p.Cout + n0 + c8;
+                    v4[i][0] = *reinterpret_cast<f32x4*>(
                         scratch + rl * BN + cs);
+                    v4[i][1] = *reinterpret_cast<f32x4*>(
+                        scratch + rl * BN + cs + 4);
                     if (HAS_RES)
-                        rv[i] = *reinterpret_cast<const bf16x4*>(
+                        rv[i] = *reinterpret_cast<const bf16x8*>(
                             RES + off[i]);
                 }
                 __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
-                for (int i = 0; i < CPR; ++i) {
-                    bf16x4 o;
+                for (int i = 0; i < CP8; ++i) {
+                    bf16x8 o;
 #pragma unroll
-                    for (int j = 0; j < 4; ++j) {
-                        float v = v4[i][j];
+                    for (int j = 0; j < 8; ++j) {
+                        float v = v4[i][j / 4][j % 4];
                         if (HAS_RES) v += bf2f(rv[i][j]);
                         o[j] = f2bf(apply_act(v, ACT));
                     }
-                    *reinterpret_cast<bf16x4*>(OUT + off[i]) = o;
+                    *reinterpret_cast<bf16x8*>(OUT + off[i]) = o;
                 }
             } else {
 #pragma unroll
@@ -502,6 +509,13 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
     } else {
         BNsel = 64;
         BMsel = (mt128 * ((p.Cout + 63) / 64) >= 384) ? 128 : 64;
+    }
+    // underfill guard: 256 CUs want >=256 workgroups; narrow the n-tile
+    // before leaving CUs idle (small-M deep-K shapes, e.g. 7x7x512)
+    if (BNsel == 128 &&
+        (long)((p.M + BMsel - 1) / BMsel) * ((p.Cout + 127) / 128) < 256) {
+        BNsel = 64;
+        BMsel = 64;
     }
     const int mtiles = (p.M + BMsel - 1) / BMsel;
     const int ny = (p.Cout + BNsel - 1) / BNsel;
