@@ -397,11 +397,16 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
             else
                 asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
             __builtin_amdgcn_s_barrier();
-            compute(it % DEPTH, B_PERSIST ? 0 : it % DEPTH);
+            // stage BEFORE compute: buffer (it+DEPTH-1)%DEPTH was last
+            // read by compute(it-1), which every wave finished before
+            // this barrier — issuing the glds here buys them one extra
+            // compute phase of latency cover (DEPTH=2 gets its only
+            // overlap from this)
             if (s_mt < mtiles) {
                 stage((it + DEPTH - 1) % DEPTH);
                 ++staged;
             }
+            compute(it % DEPTH, B_PERSIST ? 0 : it % DEPTH);
         }
         // epilogue AFTER the next tiles' stages were issued: their glds
         // land in buffers (it..it+DEPTH-2)%DEPTH, disjoint from the
