@@ -41,7 +41,11 @@ def parse_args():
     ap.add_argument("--ring-depth", type=int, default=4)
     ap.add_argument("--no-return-results", action="store_true")
     ap.add_argument("--stats", action="store_true",
-                    help="print per-stage stats to stderr")
+                    help="print per-stage stats to stderr (hipEvent "
+                         "compute time, wire bytes)")
+    ap.add_argument("--dump-partition", default=None, metavar="DIR",
+                    help="write per-stage DOT/text partition dumps "
+                         "(plot_model parity, reference node.py:39)")
     return ap.parse_args()
 
 
@@ -88,7 +92,9 @@ def main():
         compression=args.compression, zfp_rate_bits=args.zfp_bits,
         ring_depth=args.ring_depth,
         backend="nccl" if args.device == "cuda" else "gloo",
-        return_results=not args.no_return_results)
+        return_results=not args.no_return_results,
+        log_stage_stats=args.stats,
+        partition_dump_dir=args.dump_partition)
 
     dev = (torch.device("cuda", local_rank) if args.device == "cuda"
            else torch.device("cpu"))
@@ -120,6 +126,7 @@ def main():
     # ---- warmup (fills pipeline, triggers graph capture)
     pipe.run(args.warmup, feed=feed, collect=collect)
     barrier_sync()
+    pipe.reset_stats()
 
     # ---- timed region: exactly --steps items
     t0 = time.perf_counter()
@@ -139,7 +146,10 @@ def main():
     if args.stats:
         import sys
         st = pipe.stats
+        cms = st.compute_ms / max(st.items, 1)
+        busy = st.compute_ms / 1e3 / el * 100 if el > 0 else 0.0
         print(f"[stage {pipe.rank}] items={st.items} images={st.images} "
+              f"compute={cms:.3f}ms/item (dev busy ~{busy:.0f}%) "
               f"wire_in={st.bytes_in/1e6:.1f}MB wire_out="
               f"{st.bytes_out/1e6:.1f}MB "
               f"({st.bytes_out/max(st.items,1)/1e6:.2f} MB/item)",

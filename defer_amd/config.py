@@ -46,6 +46,9 @@ class PipelineConfig:
 
     # --- observability ----------------------------------------------------
     log_stage_stats: bool = False  # per-stage imgs/s, bytes relayed, ratios
+    # When set, write per-stage DOT + text partition dumps here (the
+    # reference writes model_{ip}.png per node, node.py:39).
+    partition_dump_dir: Optional[str] = None
 
     extra: dict = field(default_factory=dict)
 

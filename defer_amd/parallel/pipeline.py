@@ -37,7 +37,8 @@ from defer_amd.parallel.partitioner import (as_graph_model, auto_partition,
 class StageStats:
     items: int = 0
     images: int = 0
-    compute_s: float = 0.0
+    compute_s: float = 0.0    # host wall (threaded DEFER orchestrator)
+    compute_ms: float = 0.0   # device time, hipEvent pairs (DistPipeline)
     bytes_in: int = 0
     bytes_out: int = 0
 
@@ -147,10 +148,16 @@ class DEFER:
         if len(stages) != n:
             raise ValueError(
                 f"{len(stages)} stages for {n} compute nodes")
+        if cfg.partition_dump_dir:
+            from defer_amd.utils.visualize import dump_partition
+
+            dump_partition(stages, cfg.partition_dump_dir,
+                           cut_points=partition_layers)
         dtype = cfg.torch_dtype()
         execs = [StageExecutor(s, dev, dtype, cfg.use_hip_graphs)
                  for s, dev in zip(stages, self.computeNodes)]
         self.stats = [StageStats() for _ in range(n)]
+        self.errors: List = [None] * n
         qs = [input_stream] + [queue.Queue(cfg.ring_depth)
                                for _ in range(n)]
         ready = [threading.Event() for _ in range(n)]
@@ -167,12 +174,23 @@ class DEFER:
                 if x is None:
                     qs[i + 1].put(None)
                     return
-                if dev.type == "cuda":
-                    x = x.to(dev, dtype, non_blocking=True)
-                t0 = time.perf_counter()
-                with torch.no_grad():
-                    y = ex.run(x)
-                st.compute_s += time.perf_counter() - t0
+                try:
+                    if dev.type == "cuda":
+                        x = x.to(dev, dtype, non_blocking=True)
+                    t0 = time.perf_counter()
+                    with torch.no_grad():
+                        y = ex.run(x)
+                    st.compute_s += time.perf_counter() - t0
+                except Exception as e:   # fail-fast per stage: record the
+                    # error, poison downstream, then drain upstream so
+                    # earlier stages never block on a full queue — a dead
+                    # stage must not hang the chain (the reference just
+                    # hangs, SURVEY.md §5 failure detection)
+                    self.errors[i] = e
+                    qs[i + 1].put(None)
+                    while qs[i].get() is not None:
+                        pass
+                    return
                 st.items += 1
                 st.images += x.shape[0] if x.dim() > 1 else 1
                 qs[i + 1].put(y)
@@ -193,6 +211,12 @@ class DEFER:
             output_stream.put(y.float().cpu() if y.is_cuda else y)
         for t in threads:
             t.join()
+        failed = [(i, e) for i, e in enumerate(self.errors)
+                  if e is not None]
+        if failed:
+            i, e = failed[0]
+            raise RuntimeError(
+                f"stage {i} on {self.computeNodes[i]} failed: {e!r}") from e
 
 
 # --------------------------------------------------------------------------
@@ -235,6 +259,11 @@ class DistPipeline:
         if len(stages) != self.world:
             raise ValueError(f"{len(stages)} stages != world {self.world}")
         self.cuts = cuts
+        if cfg.partition_dump_dir and self.rank == 0:
+            from defer_amd.utils.visualize import dump_partition
+
+            dump_partition(stages, cfg.partition_dump_dir,
+                           cut_points=cuts)
         # shape-trace on CPU BEFORE weights move to the GPU
         self.in_shape, self.out_shape = self._boundary_shapes(stages)
         self.stage = StageExecutor(stages[self.rank], self.device,
@@ -267,6 +296,19 @@ class DistPipeline:
                     self.result_ring = P2PRing(self.res_codec,
                                                cfg.ring_depth)
         self.stats = StageStats()
+        self.timer = None
+        if cfg.log_stage_stats:
+            from defer_amd.utils.trace import EventTimer
+
+            self.timer = EventTimer(self.device)
+
+    def reset_stats(self):
+        """Zero counters/timers (e.g. between warmup and the timed run)."""
+        self.stats = StageStats()
+        if self.timer is not None:
+            from defer_amd.utils.trace import EventTimer
+
+            self.timer = EventTimer(self.device)
 
     def _wire_bytes_per_elem(self):
         if self.cfg.compression.startswith("zfp"):
@@ -306,11 +348,19 @@ class DistPipeline:
         r, W = self.rank, self.world
         D = cfg.ring_depth
 
+        tm = self.timer
+
         if W == 1:
             for k in range(steps):
+                if tm:
+                    tm.start()
                 y = self.stage.run(feed(k))
+                if tm:
+                    tm.stop()
                 if collect:
                     collect(k, y)
+            if tm:
+                self.stats.compute_ms = tm.total_ms()
             return
 
         nxt, prv = r + 1, r - 1
@@ -332,7 +382,11 @@ class DistPipeline:
                                         * dtype_bytes(wire.dtype))
                 x = self.in_codec.decode(wire)
             # ---- compute
+            if tm:
+                tm.start()
             y = self.stage.run(x)
+            if tm:
+                tm.stop()
             # ---- repost recv. Safe to reuse the slot: the irecv is posted
             # AFTER the compute consuming it was enqueued, and
             # ProcessGroupNCCL orders the recv after the current stream's
@@ -356,6 +410,8 @@ class DistPipeline:
             self.stats.items += 1
             self.stats.images += self.batch_shape[0]
 
+        if tm:
+            self.stats.compute_ms = tm.total_ms()
         # drain outstanding sends
         for ring in (self.send_ring, self.result_ring):
             if ring is not None:

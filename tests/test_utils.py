@@ -1,0 +1,109 @@
+"""Aux-subsystem tests: partition visualization (plot_model parity,
+reference node.py:39), per-stage event tracing, fail-fast stage errors
+(SURVEY.md §5)."""
+
+import queue
+import threading
+
+import pytest
+import torch
+
+from defer_amd import DEFER, PipelineConfig
+from defer_amd.models import resnet50
+from defer_amd.parallel.partitioner import as_graph_model, partition_model
+from defer_amd.utils.trace import EventTimer
+from defer_amd.utils.visualize import describe, dump_partition, to_dot
+
+
+def test_dot_and_describe():
+    gm = as_graph_model(resnet50())
+    dot = to_dot(gm.graph, name="resnet50", cut_points=["add_8"])
+    assert dot.startswith('digraph "resnet50"')
+    assert '"add_8"' in dot and "peripheries=2" in dot
+    # every node appears, every edge references defined nodes
+    for n in gm.graph.nodes:
+        assert f'"{n.name}"' in dot
+    txt = describe(gm.graph, name="resnet50")
+    assert "layers" in txt and "params" in txt
+    assert "add_8" in txt
+
+
+def test_dump_partition(tmp_path):
+    gm = as_graph_model(resnet50())
+    stages = partition_model(gm, ["add_4", "add_8"])
+    paths = dump_partition(stages, str(tmp_path), ["add_4", "add_8"])
+    assert len(paths) == 2 * 3 + 1
+    overview = (tmp_path / "partition.txt").read_text()
+    assert "stage 0" in overview and "stage 2" in overview
+    assert "add_4" in overview
+    # stage dumps are valid-looking DOT
+    assert (tmp_path / "stage_1.dot").read_text().startswith("digraph")
+
+
+def test_event_timer_cpu():
+    import time
+
+    t = EventTimer(torch.device("cpu"))
+    for _ in range(3):
+        t.start()
+        time.sleep(0.01)
+        t.stop()
+    assert t.count == 3
+    assert 20 < t.total_ms() < 500
+    assert t.mean_ms() > 5
+
+
+def test_defer_stage_failure_fails_fast():
+    class Boom(torch.nn.Module):
+        def forward(self, x):
+            raise ValueError("kaboom")
+
+    from defer_amd.graph import GraphModel, GraphNode, LayerGraph
+
+    g = LayerGraph([GraphNode("ok", torch.nn.Identity(), ["input"]),
+                    GraphNode("boom", Boom(), ["ok"])])
+    eng = DEFER(["cpu", "cpu"],
+                config=PipelineConfig(device="cpu", dtype="fp32"))
+    in_q, out_q = queue.Queue(4), queue.Queue(4)
+    err = []
+
+    def run():
+        try:
+            eng.run_defer(GraphModel(g), ["ok"], in_q, out_q)
+        except RuntimeError as e:
+            err.append(e)
+
+    t = threading.Thread(target=run)
+    t.start()
+    for _ in range(3):
+        in_q.put(torch.randn(2, 4))
+    in_q.put(None)
+    t.join(timeout=60)
+    assert not t.is_alive(), "failed stage hung the chain"
+    assert err and "stage 1" in str(err[0]) and "kaboom" in str(err[0])
+
+
+def test_dist_pipeline_partition_dump(tmp_path):
+    # world=1 path exercises the dump hook without a process group
+    import torch.distributed as dist
+
+    if not dist.is_initialized():
+        import os
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29781")
+        dist.init_process_group("gloo", rank=0, world_size=1)
+    from defer_amd.parallel.pipeline import DistPipeline
+
+    cfg = PipelineConfig(device="cpu", dtype="fp32", batch_size=2,
+                         use_hip_graphs=False, backend="gloo",
+                         log_stage_stats=True,
+                         partition_dump_dir=str(tmp_path / "dump"))
+    pipe = DistPipeline(resnet50(), cfg, (2, 64, 64, 3),
+                        device=torch.device("cpu"))
+    pipe.run(2, feed=lambda k: torch.randn(2, 64, 64, 3),
+             collect=lambda k, y: None)
+    assert (tmp_path / "dump" / "partition.txt").exists()
+    assert pipe.stats.compute_ms > 0
+    pipe.reset_stats()
+    assert pipe.stats.compute_ms == 0
