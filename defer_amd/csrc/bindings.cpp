@@ -6,6 +6,10 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
+#include <functional>
+#include <mutex>
+#include <unordered_map>
+
 #include "kernels.h"
 
 namespace {
@@ -44,6 +48,27 @@ const float* zeros_buf() {
     static Tensor t = at::zeros(
         {8192}, at::TensorOptions().dtype(at::kFloat).device(at::kCUDA));
     return t.data_ptr<float>();
+}
+
+// Prepared-weight cache for the stem paths (K-order repack / channel
+// pad). Weights are constant in inference, so the one-time transform is
+// keyed on (data_ptr, numel, variant) — the reference pays its weight
+// prep once at dispatch too (node.py:34). Unbounded but tiny (one entry
+// per stem conv weight).
+Tensor cached_weight_prep(const Tensor& w, int64_t variant,
+                          const std::function<Tensor()>& make) {
+    static std::unordered_map<uint64_t, std::pair<const void*, Tensor>>
+        cache;
+    static std::mutex mu;
+    uint64_t key = (uint64_t)(uintptr_t)w.data_ptr() * 31 +
+                   (uint64_t)w.numel() * 7 + (uint64_t)variant;
+    std::lock_guard<std::mutex> g(mu);
+    auto it = cache.find(key);
+    if (it != cache.end() && it->second.first == w.data_ptr())
+        return it->second.second;
+    Tensor t = make();
+    cache[key] = {w.data_ptr(), t};
+    return t;
 }
 
 const float* fptr_opt(const c10::optional<Tensor>& t, const char* name,
@@ -105,9 +130,12 @@ Tensor conv2d_bn_act(Tensor x, Tensor w, c10::optional<Tensor> scale,
             auto xp = at::empty({NB, PH, PW, Cin}, x.options());
             defer_hip::launch_pad2d(bptr(x), bptr_mut(xp), NB, H, W, Cin,
                                     PH, PW, (int)pad, (int)pad, s);
-            auto wp = at::empty({Cout, Keff}, w.options());
-            defer_hip::launch_stem_repack_w(bptr(w), bptr_mut(wp), Cout,
-                                            R, S, Cin, TR, s);
+            auto wp = cached_weight_prep(w, TR, [&] {
+                auto t = at::empty({Cout, Keff}, w.options());
+                defer_hip::launch_stem_repack_w(bptr(w), bptr_mut(t),
+                                                Cout, R, S, Cin, TR, s);
+                return t;
+            });
             p.x = bptr(xp);
             p.w = bptr(wp);
             p.K = Keff;
@@ -125,9 +153,12 @@ Tensor conv2d_bn_act(Tensor x, Tensor w, c10::optional<Tensor> scale,
         auto xp = at::empty({NB, H, W, C8}, x.options());
         defer_hip::launch_pad_channels(bptr(x), bptr_mut(xp),
                                        (long)NB * H * W, Cin, C8, s);
-        auto wp = at::empty({Cout, R, S, C8}, w.options());
-        defer_hip::launch_pad_channels(bptr(w), bptr_mut(wp),
-                                       (long)Cout * R * S, Cin, C8, s);
+        auto wp = cached_weight_prep(w, -C8, [&] {
+            auto t = at::empty({Cout, R, S, C8}, w.options());
+            defer_hip::launch_pad_channels(bptr(w), bptr_mut(t),
+                                           (long)Cout * R * S, Cin, C8, s);
+            return t;
+        });
         p.x = bptr(xp);
         p.w = bptr(wp);
         p.K = R * S * C8;

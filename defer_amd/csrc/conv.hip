@@ -37,6 +37,8 @@
 //
 // GEMM mode (R=S=1, H=W=1, Cin=K): out[M,N] = x[M,K] @ w[N,K]^T — used
 // for 1x1/s1 convs, the dense classifier head, and any explicit GEMM.
+#include <cstdlib>
+
 #include "common.h"
 #include "kernels.h"
 
@@ -65,9 +67,21 @@ __device__ __forceinline__ void glds16(const bf16* src, bf16* lds_base) {
         16, 0, 0);
 }
 
+// DA/DB: independent pipeline depths for the A (activation) and B
+// (weight) LDS rings. A streams from HBM (~900 cy latency) and needs
+// DA-1 compute phases of cover; B re-reads L2-resident weights (~250 cy)
+// and DB=2 suffices — the LDS saved goes into a deeper A ring while
+// keeping 2 blocks/CU (160 KB LDS). With DA>DB the per-iteration stage
+// order is B-then-A so the counted s_waitcnt (FIFO vmcnt) can float the
+// newest A stage past the B(it) wait.
+// WPS: minimum waves per SIMD (__launch_bounds__ 2nd arg) = blocks/CU of
+// this 256-thread kernel. The MFMA-bound tile configs run 2 blocks/CU;
+// the pure-bandwidth small-tile GEMM configs (1x1 convs, K<=512) trade
+// registers for 3-4 co-resident blocks so more independent load streams
+// and epilogues overlap.
 template <int ACT, bool HAS_RES, int AMODE, bool B_PERSIST, int BM,
-          int BN, int DEPTH>
-__global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
+          int BN, int DA, int DB, int WPS = 2>
+__global__ __launch_bounds__(NTHREADS, WPS) void conv_igemm_kernel(
     ConvParams p) {
     const bf16* __restrict__ X = (const bf16*)p.x;
     const bf16* __restrict__ Wt = (const bf16*)p.w;
@@ -83,11 +97,12 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
     constexpr int ACH = BM / 32;                // A chunks per thread
     constexpr int BCH = BN / 32;                // B chunks per thread
     constexpr int OPS = ACH + (B_PERSIST ? 0 : BCH);
+    constexpr bool ASYM = !B_PERSIST && (DA != DB);
 
     __shared__ __attribute__((aligned(16)))
-    bf16 lds[(DEPTH * BM + (B_PERSIST ? 1 : DEPTH) * BN) * BK];
+    bf16 lds[(DA * BM + (B_PERSIST ? 1 : DB) * BN) * BK];
     bf16* A0 = lds;
-    bf16* B0 = lds + DEPTH * BM * BK;
+    bf16* B0 = lds + DA * BM * BK;
 
     const int tid = threadIdx.x;
     const int wave = tid / WAVE;
@@ -212,13 +227,12 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
             glds16(src, A + (wave * (ACH * 64) + i * 64) * 8);
         }
     };
-    auto stage_b = [&](int buf) {
+    auto stage_b = [&](int buf, int kt, int cb, int r, int sI) {
         bf16* B = B0 + buf * BN * BK;
         // RSC: same K permutation as A, applied to the OHWI weights
         const long koff = (AMODE == AMODE_RSC)
-                              ? (long)(s_r * p.S + s_s) * p.Cin
-                                    + s_cb * 64
-                              : (long)s_kt * BK;
+                              ? (long)(r * p.S + sI) * p.Cin + cb * 64
+                              : (long)kt * BK;
 #pragma unroll
         for (int i = 0; i < BCH; ++i) {
             int n = n0 + b_row[i];
@@ -352,10 +366,10 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
         }
     };
 
-    // ---- flattened (m-tile, k-tile) pipeline; the stage at the END of
-    // each iteration keeps the cursor DEPTH-1 tiles ahead of compute
-    // (the refilled buffer was last read at compute(it-(DEPTH-1)), which
-    // every wave finished before this iteration's barrier).
+    // ---- flattened (m-tile, k-tile) pipeline. The A stage cursor runs
+    // DA-1 tiles ahead of compute; the B cursor DB-1 (ASYM) or DA-1
+    // (symmetric). Counted vmcnt: tile it's stages must have landed by
+    // wait(it); newer stages stay in flight (the latency cover).
     auto advance = [&]() {
         if (++s_kt == nk) {
             s_kt = 0;
@@ -369,20 +383,64 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
             }
         }
     };
-    auto stage = [&](int buf) {
-        stage_a(buf);
-        if (!B_PERSIST) stage_b(buf);
-        advance();
+    // (ASYM) independent, shallower B cursor: stages tile it+DB-1 while
+    // the A cursor stages it+DA-1. B is L2-resident weight traffic; one
+    // to two compute phases of cover suffice.
+    long b_left = 0, b_tile = 0;
+    int b_kt = 0, b_cb = 0, b_r = 0, b_s = 0;
+    if (ASYM)
+        b_left = (blockIdx.x < mtiles)
+                     ? (long)((mtiles - 1 - blockIdx.x) / gridDim.x + 1)
+                           * nk
+                     : 0;
+    auto b_advance = [&]() {
+        --b_left; ++b_tile;
+        if (++b_kt == nk) {
+            b_kt = 0; b_cb = 0; b_r = 0; b_s = 0;
+        } else if (AMODE == AMODE_RSC) {
+            if (++b_s == p.S) {
+                b_s = 0;
+                if (++b_r == p.R) { b_r = 0; ++b_cb; }
+            }
+        }
     };
 
     if (s_mt < mtiles) a_setup(s_mt);
-    if (B_PERSIST) stage_b(0);      // oldest ops: drained by first wait
+    if (B_PERSIST) stage_b(0, 0, 0, 0, 0);  // oldest ops: first wait drains
     int staged = 0;
+    // `run`: consecutive preceding iterations that issued BOTH stages —
+    // the guard for the ASYM float allowance (tail falls back to full
+    // drains). Prologue interleaves B/A per virtual iteration so the
+    // newest-op window at wait(0) is exactly the steady-state one.
+    int run = 0;
 #pragma unroll
-    for (int d = 0; d < DEPTH - 1; ++d)
-        if (s_mt < mtiles) { stage(d); ++staged; }
+    for (int d = 0; d < DA - 1; ++d) {
+        bool did_b = true;
+        if (ASYM && d < DB - 1) {
+            if (b_left > 0) {
+                stage_b((int)(b_tile % DB), b_kt, b_cb, b_r, b_s);
+                b_advance();
+            } else {
+                did_b = false;
+            }
+        }
+        if (s_mt < mtiles) {
+            stage_a(d);
+            if (!B_PERSIST && !ASYM)
+                stage_b(d, s_kt, s_cb, s_r, s_s);
+            advance();
+            ++staged;
+            if (did_b) ++run;
+        }
+    }
+    if (run < DA - 1) run = 0;          // incomplete prologue: drain fully
 
-    int it = 0;   // computed-iteration counter; tile i -> buffer i%DEPTH
+    // ASYM float allowance: everything newer than B(it) — the A stages
+    // of the last DB-1 iterations plus their B stages beyond the first.
+    constexpr int NFLOAT =
+        ASYM ? ((DB - 2) * OPS + ACH) : ((DA - 2) * OPS);
+
+    int it = 0;   // computed-iteration counter; tile i -> buffer i%DA
     for (int mt = blockIdx.x; mt < mtiles; mt += gridDim.x) {
 #pragma unroll
         for (int mi = 0; mi < 2; ++mi)
@@ -390,28 +448,55 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_igemm_kernel(
             for (int ni = 0; ni < NI; ++ni)
                 acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
         for (int kt = 0; kt < nk; ++kt, ++it) {
-            // tile `it` landed once <= OPS*(tiles staged after it)
-            // remain outstanding
-            if (staged - it - 1 >= 1)
-                asm volatile("s_waitcnt vmcnt(%0)" ::"i"(OPS) : "memory");
-            else
-                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            if (ASYM) {
+                if (run >= DB - 1)
+                    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NFLOAT)
+                                 : "memory");
+                else
+                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            } else {
+                const int avail = staged - it - 1;
+                if (DA >= 4 && avail >= 2)
+                    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NFLOAT)
+                                 : "memory");
+                else if (avail >= 1)
+                    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(OPS)
+                                 : "memory");
+                else
+                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            }
             __builtin_amdgcn_s_barrier();
-            // stage BEFORE compute: buffer (it+DEPTH-1)%DEPTH was last
-            // read by compute(it-1), which every wave finished before
-            // this barrier — issuing the glds here buys them one extra
-            // compute phase of latency cover (DEPTH=2 gets its only
-            // overlap from this)
-            if (s_mt < mtiles) {
-                stage((it + DEPTH - 1) % DEPTH);
+            // stage BEFORE compute: the refilled buffers were last read
+            // by compute(it-1), which every wave finished before this
+            // barrier — issuing the glds here buys one extra compute
+            // phase of latency cover
+            if (ASYM) {
+                bool did_b = false, did_a = false;
+                if (b_left > 0) {
+                    stage_b((int)(b_tile % DB), b_kt, b_cb, b_r, b_s);
+                    b_advance();
+                    did_b = true;
+                }
+                if (s_mt < mtiles) {
+                    stage_a((it + DA - 1) % DA);
+                    advance();
+                    ++staged;
+                    did_a = true;
+                }
+                run = (did_b && did_a) ? run + 1 : 0;
+            } else if (s_mt < mtiles) {
+                stage_a((it + DA - 1) % DA);
+                if (!B_PERSIST)
+                    stage_b((it + DA - 1) % DA, s_kt, s_cb, s_r, s_s);
+                advance();
                 ++staged;
             }
-            compute(it % DEPTH, B_PERSIST ? 0 : it % DEPTH);
+            compute(it % DA,
+                    B_PERSIST ? 0 : (ASYM ? (int)(it % DB) : it % DA));
         }
         // epilogue AFTER the next tiles' stages were issued: their glds
-        // land in buffers (it..it+DEPTH-2)%DEPTH, disjoint from the
-        // scratch buffer (it-1)%DEPTH this writes
-        epilogue(mt, (it - 1) % DEPTH);
+        // land in buffers disjoint from the scratch buffer (it-1)%DA
+        epilogue(mt, (it - 1) % DA);
     }
 }
 
@@ -505,10 +590,34 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
     // tile selection. BN128 halves A re-staging and doubles MFMA per
     // staged byte; memory-bound small-K shapes keep the 3-deep BN64
     // pipeline. BM=64 for small-M shapes (more blocks on the 256 CUs).
+    // DEFER_CONV_VARIANT=legacy reverts to the symmetric-depth policy
+    // (A/B comparison harness).
+    static const bool legacy = [] {
+        const char* e = getenv("DEFER_CONV_VARIANT");
+        return e && e[0] == 'l';
+    }();
     const int mt128 = (p.M + 127) / 128;
     int BMsel, BNsel;
     const bool deepK = p.K >= 1024;
-    if (deepK && p.Cout >= 128) {
+    // 1x1 convs (any stride — stride-2 projections gather via AMODE_RSC
+    // but share the GEMM arithmetic shape) follow the GEMM policy.
+    const bool one1 = gemm_mode || (p.R == 1 && p.S == 1 && p.pad == 0);
+    // 1x1 shapes with K>=512: wide n-tile (halves A re-staging; measured
+    // −10…14% on the L3/L4 1x1s; K=128/256 shapes regress with BN128 —
+    // epilogue-frequency-bound — and keep BN64)
+    const bool wide = (deepK || (!legacy && one1 && !bp &&
+                                 p.K >= 512)) &&
+                      p.Cout >= 128;
+    // pure-bandwidth 1x1s (K<=256): small tiles + 3-4 blocks/CU — these
+    // shapes are HBM/L3-streaming-bound (MFMA <10% busy), so occupancy
+    // and independent load streams beat tile efficiency
+    const bool smallgemm =
+        !legacy && one1 && !wide &&
+        (long)((p.M + 63) / 64) * ((p.Cout + 63) / 64) >= 768;
+    if (smallgemm) {
+        BNsel = 64;
+        BMsel = 64;
+    } else if (wide) {
         BNsel = 128;
         BMsel = (mt128 * (p.Cout / 128) >= 384) ? 128 : 64;
     } else {
@@ -525,7 +634,7 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
     const int mtiles = (p.M + BMsel - 1) / BMsel;
     const int ny = (p.Cout + BNsel - 1) / BNsel;
     int gx = mtiles;
-    const int target = 768;
+    const int target = smallgemm ? (bp ? 1536 : 1152) : 768;
     if ((long)mtiles * ny > target) {
         gx = target / ny > 0 ? target / ny : 1;
         if (gx > mtiles) gx = mtiles;
@@ -533,11 +642,11 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
     dim3 grid(gx, ny);
     dim3 block(NTHREADS);
 
-#define DISPATCH_TILE(A, R, G, BP, BMv, BNv, D)                           \
+#define DISPATCH_TILE(A, R, G, BP, BMv, BNv, D...)                        \
     hipLaunchKernelGGL(                                                   \
         (conv_igemm_kernel<A, R, G, BP, BMv, BNv, D>), grid, block, 0, s, \
         p)
-#define DISPATCH_BOOLS(BMv, BNv, D)                                       \
+#define DISPATCH_BOOLS(BMv, BNv, D...)                                    \
     do {                                                                  \
         if (relu) {                                                       \
             if (has_res) {                                                \
@@ -618,10 +727,27 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
         }                                                                 \
     } while (0)
 
-    if (BMsel == 128 && BNsel == 64) DISPATCH_BOOLS(128, 64, 3);
-    else if (BMsel == 128 && BNsel == 128) DISPATCH_BOOLS(128, 128, 2);
-    else if (BMsel == 64 && BNsel == 128) DISPATCH_BOOLS(64, 128, 3);
-    else DISPATCH_BOOLS(64, 64, 3);
+    // depths (A-ring, B-ring). Measured on MI355X: symmetric depths win
+    // (boundary activations are L3/L2-resident, so one to two compute
+    // phases already cover the load latency and extra LDS per block
+    // costs more than deeper cover buys); the asymmetric deep-A variants
+    // (DEFER_CONV_VARIANT unset ... kept instantiated) lost 6-22% and
+    // remain only for the B_PERSIST single-K-tile shapes where a 4-deep
+    // A ring measured neutral-to-slightly-better.
+    if (BMsel == 128 && BNsel == 64) {
+        if (bp && !legacy) DISPATCH_BOOLS(128, 64, 4, 2);
+        else DISPATCH_BOOLS(128, 64, 3, 3);
+    } else if (BMsel == 128 && BNsel == 128) {
+        DISPATCH_BOOLS(128, 128, 2, 2);
+    } else if (BMsel == 64 && BNsel == 128) {
+        DISPATCH_BOOLS(64, 128, 3, 3);
+    } else if (smallgemm) {
+        if (bp) DISPATCH_BOOLS(64, 64, 4, 2, 4);   // 40 KB LDS, 4 blk/CU
+        else DISPATCH_BOOLS(64, 64, 3, 3, 3);      // 48 KB LDS, 3 blk/CU
+    } else {
+        if (bp && !legacy) DISPATCH_BOOLS(64, 64, 4, 2);
+        else DISPATCH_BOOLS(64, 64, 3, 3);
+    }
 #undef DISPATCH_BOOLS
 #undef DISPATCH_TILE
 }

@@ -359,6 +359,8 @@ class DistPipeline:
                     tm.stop()
                 if collect:
                     collect(k, y)
+                self.stats.items += 1
+                self.stats.images += self.batch_shape[0]
             if tm:
                 self.stats.compute_ms = tm.total_ms()
             return
