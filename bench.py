@@ -51,8 +51,8 @@ def parse_args():
 
 def main():
     args = parse_args()
-    # keep the one-JSON-line stdout contract clean of RCCL banners
-    os.environ.setdefault("NCCL_DEBUG", "WARN")
+    # keep the one-JSON-line stdout contract clean: RCCL prints a version
+    # banner to fd 1 when NCCL_DEBUG is set, so leave it unset here
     from defer_amd.config import PipelineConfig
     from defer_amd.models import DEFER_8STAGE_CUTS, resnet50, vgg19
     from defer_amd.parallel.pipeline import DistPipeline

@@ -592,10 +592,15 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
     // pipeline. BM=64 for small-M shapes (more blocks on the 256 CUs).
     // DEFER_CONV_VARIANT=legacy reverts to the symmetric-depth policy
     // (A/B comparison harness).
-    static const bool legacy = [] {
+    // variants: default = measured policy; "legacy" = symmetric-depth
+    // policy pre-v2; "small" = force the high-occupancy small-tile
+    // config everywhere it is legal (experiment harness)
+    static const char variant = [] {
         const char* e = getenv("DEFER_CONV_VARIANT");
-        return e && e[0] == 'l';
+        return e ? e[0] : '\0';
     }();
+    const bool legacy = variant == 'l';
+    const bool force_small = variant == 's';
     const int mt128 = (p.M + 127) / 128;
     int BMsel, BNsel;
     const bool deepK = p.K >= 1024;
@@ -611,8 +616,14 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
     // pure-bandwidth 1x1s (K<=256): small tiles + 3-4 blocks/CU — these
     // shapes are HBM/L3-streaming-bound (MFMA <10% busy), so occupancy
     // and independent load streams beat tile efficiency
+    // measured small-tile winners besides the 1x1s: the stem paths and
+    // shallow-K 3x3s at moderate M (deep-K 3x3s and the giant-M VGG
+    // b1.c2 regress -2..-20% at BM64 — B-tile re-staging doubles)
+    const bool small_conv =
+        stem_mode || (!one1 && (p.K <= 128 ||
+                                (p.K <= 576 && p.M <= 450000)));
     const bool smallgemm =
-        !legacy && one1 && !wide &&
+        (force_small || (!legacy && ((one1 && !wide) || small_conv))) &&
         (long)((p.M + 63) / 64) * ((p.Cout + 63) / 64) >= 768;
     if (smallgemm) {
         BNsel = 64;
