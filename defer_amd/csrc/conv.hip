@@ -146,10 +146,19 @@ __global__ __launch_bounds__(NTHREADS, WPS) void conv_igemm_kernel(
     }
 
     // ---- per-chunk A state for the STAGED m-tile (refreshed when the
-    // stage cursor crosses into a new m-tile — once per nk k-tiles)
+    // stage cursor crosses into a new m-tile — once per nk k-tiles).
+    // RSC keeps an INCREMENTAL per-chunk pointer (advanced by a scalar
+    // delta per k-tile) plus a (r,s)-validity bitmask precomputed here,
+    // so the hot stage carries no multiplies and no coordinate math —
+    // the k-loop was VALU-issue-bound with per-tile address rebuilds.
     long a_base[ACH];
     int a_ihb[ACH], a_iwb[ACH];
     bool a_mval[ACH];
+    const bf16* a_ptr[ACH];      // RSC: current (cb, r, s) source address
+    u32 a_vmask[ACH];            // RSC: bit (r*S+s) = window in-bounds
+    const int d_s = p.Cin;                                // s+1
+    const int d_r = (p.W - (p.S - 1)) * p.Cin;            // r+1, s=0
+    const int d_cb = 64 - ((p.R - 1) * p.W + (p.S - 1)) * p.Cin;
     auto a_setup = [&](int mt) {
         const int m0 = mt * BM;
 #pragma unroll
@@ -159,6 +168,24 @@ __global__ __launch_bounds__(NTHREADS, WPS) void conv_igemm_kernel(
             if (m >= p.M) m = p.M - 1;
             if (AMODE == AMODE_GEMM) {
                 a_base[i] = (long)m * p.K + a_k8[i] * 8;
+            } else if (AMODE == AMODE_RSC) {
+                u32 t = umagic(m, p.owmul, p.OW);       // m / OW
+                int ow = m - (int)t * p.OW;
+                u32 nb = umagic(t, p.ohmul, p.OH);      // t / OH
+                int oh = (int)t - (int)nb * p.OH;
+                const int ihb = oh * p.stride - p.pad;
+                const int iwb = ow * p.stride - p.pad;
+                a_ptr[i] = X + (long)nb * p.H * p.W * p.Cin
+                             + ((long)ihb * p.W + iwb) * p.Cin
+                             + a_k8[i] * 8;
+                u32 mask = 0;
+                if (a_mval[i])
+                    for (int r = 0; r < p.R; ++r)
+                        for (int sI = 0; sI < p.S; ++sI)
+                            if ((u32)(ihb + r) < (u32)p.H &&
+                                (u32)(iwb + sI) < (u32)p.W)
+                                mask |= 1u << (r * p.S + sI);
+                a_vmask[i] = mask;
             } else if (AMODE == AMODE_STEM) {
                 // spatially pre-padded input (pad absorbed into coords);
                 // p.H/p.W are the PADDED dims, base points at the
@@ -182,9 +209,11 @@ __global__ __launch_bounds__(NTHREADS, WPS) void conv_igemm_kernel(
     };
 
     // stage cursor scalars (RSC walks (cb, r, s) incrementally — the
-    // k-tile order IS (cb*R*S + r*S + s), no division anywhere)
+    // k-tile order IS (cb*R*S + r*S + s), no division anywhere; s_t
+    // mirrors r*S+s for the validity-mask bit index)
     int s_mt = blockIdx.x, s_kt = 0;
-    int s_cb = 0, s_r = 0, s_s = 0;
+    int s_cb = 0, s_r = 0, s_s = 0, s_t = 0;
+    const bool kal = (p.K & 63) == 0;   // k < K by construction
 
     auto stage_a = [&](int buf) {
         bf16* A = A0 + buf * BM * BK;
@@ -192,9 +221,15 @@ __global__ __launch_bounds__(NTHREADS, WPS) void conv_igemm_kernel(
         for (int i = 0; i < ACH; ++i) {
             const bf16* src = Z;
             if (AMODE == AMODE_GEMM) {
-                int k = s_kt * BK + a_k8[i] * 8;
-                if (a_mval[i] && k < p.K)
+                if (kal) {
+                    // K%64==0 (every 1x1/dense in practice): clamped
+                    // rows read valid memory, no per-lane checks
                     src = X + a_base[i] + (long)s_kt * BK;
+                } else {
+                    int k = s_kt * BK + a_k8[i] * 8;
+                    if (a_mval[i] && k < p.K)
+                        src = X + a_base[i] + (long)s_kt * BK;
+                }
             } else if (AMODE == AMODE_CONV) {
                 int k = s_kt * BK + a_k8[i] * 8;
                 u32 rs = umagic(k, p.cmul, p.Cin);      // k / Cin
@@ -217,31 +252,47 @@ __global__ __launch_bounds__(NTHREADS, WPS) void conv_igemm_kernel(
                 if (a_mval[i])
                     src = X + a_base[i]
                               + (long)r * p.W * p.Cin + t;
-            } else {  // AMODE_RSC
-                int ih = a_ihb[i] + s_r;
-                int iw = a_iwb[i] + s_s;
-                if (a_mval[i] && (u32)ih < (u32)p.H && (u32)iw < (u32)p.W)
-                    src = X + a_base[i] + ((long)ih * p.W + iw) * p.Cin
-                              + s_cb * 64 + a_k8[i] * 8;
+            } else {  // AMODE_RSC: incremental pointer + mask-bit select
+                src = a_ptr[i];
+                if (!((a_vmask[i] >> s_t) & 1u)) src = Z;
             }
             glds16(src, A + (wave * (ACH * 64) + i * 64) * 8);
         }
     };
+    // B lane offsets are constant across k-tiles: per-tile addressing is
+    // a SCALAR base (Wt + koff) + 32-bit lane offset, which lowers to
+    // the saddr global_load_lds form — no per-lane 64-bit math in the
+    // loop. Rows past Cout duplicate the last row (their columns are
+    // never stored; the epilogue clamps), so no zero-page redirect.
+    u32 b_voff[BCH];
+#pragma unroll
+    for (int i = 0; i < BCH; ++i) {
+        int n = n0 + b_row[i];
+        if (n >= p.Cout) n = p.Cout - 1;
+        b_voff[i] = (u32)((u64)n * (u64)p.K) + (u32)(b_k8[i] * 8);
+    }
     auto stage_b = [&](int buf, int kt, int cb, int r, int sI) {
         bf16* B = B0 + buf * BN * BK;
         // RSC: same K permutation as A, applied to the OHWI weights
         const long koff = (AMODE == AMODE_RSC)
                               ? (long)(r * p.S + sI) * p.Cin + cb * 64
                               : (long)kt * BK;
+        if (AMODE == AMODE_RSC || kal) {
+            const bf16* bbase = Wt + koff;
 #pragma unroll
-        for (int i = 0; i < BCH; ++i) {
-            int n = n0 + b_row[i];
-            long k = koff + b_k8[i] * 8;
-            const bf16* src =
-                (n < p.Cout && (AMODE == AMODE_RSC || k < p.K))
-                    ? Wt + (long)n * p.K + k
-                    : Z;
-            glds16(src, B + (wave * (BCH * 64) + i * 64) * 8);
+            for (int i = 0; i < BCH; ++i)
+                glds16(bbase + b_voff[i],
+                       B + (wave * (BCH * 64) + i * 64) * 8);
+        } else {
+#pragma unroll
+            for (int i = 0; i < BCH; ++i) {
+                int n = n0 + b_row[i];
+                long k = koff + b_k8[i] * 8;
+                const bf16* src = (n < p.Cout && k < p.K)
+                                      ? Wt + (long)n * p.K + k
+                                      : Z;
+                glds16(src, B + (wave * (BCH * 64) + i * 64) * 8);
+            }
         }
     };
 
@@ -374,13 +425,22 @@ __global__ __launch_bounds__(NTHREADS, WPS) void conv_igemm_kernel(
         if (++s_kt == nk) {
             s_kt = 0;
             s_mt += gridDim.x;
-            s_cb = 0; s_r = 0; s_s = 0;
+            s_cb = 0; s_r = 0; s_s = 0; s_t = 0;
             if (s_mt < mtiles) a_setup(s_mt);
         } else if (AMODE == AMODE_RSC) {
+            int d;
             if (++s_s == p.S) {
                 s_s = 0;
-                if (++s_r == p.R) { s_r = 0; ++s_cb; }
+                if (++s_r == p.R) {
+                    s_r = 0; ++s_cb; s_t = 0; d = d_cb;
+                } else {
+                    ++s_t; d = d_r;
+                }
+            } else {
+                ++s_t; d = d_s;
             }
+#pragma unroll
+            for (int i = 0; i < ACH; ++i) a_ptr[i] += d;
         }
     };
     // (ASYM) independent, shallower B cursor: stages tile it+DB-1 while
@@ -582,7 +642,8 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
     p.smul = magic32((u32)p.S);
     const int nk = (p.K + BK - 1) / BK;
     const bool bp = (nk == 1) && !stem_mode;
-    const bool rsc = !gemm_mode && !stem_mode && !bp && (p.Cin % 64 == 0);
+    const bool rsc = !gemm_mode && !stem_mode && !bp &&
+                     (p.Cin % 64 == 0) && (p.R * p.S <= 32);
     const int amode = stem_mode ? AMODE_STEM
                                 : gemm_mode ? AMODE_GEMM
                                             : (rsc ? AMODE_RSC
