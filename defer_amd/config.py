@@ -19,6 +19,10 @@ class PipelineConfig:
     # num_stages > 1, cuts are chosen by the cost-model auto-partitioner.
     partition_layers: Optional[List[str]] = None
     num_stages: int = 1
+    # Single-item input shape for the auto-partitioner's cost trace
+    # (batch dim 1). Defaults to the reference workload's 224x224x3 NHWC
+    # image (test/test.py:20) when None.
+    input_shape: Optional[tuple] = None
 
     # --- execution --------------------------------------------------------
     device: str = "cuda"          # "cuda" (MI355X) or "cpu" (tests/plumbing)

@@ -193,7 +193,10 @@ Tensor linear(Tensor x, Tensor w, c10::optional<Tensor> bias) {
     TORCH_CHECK(K % 8 == 0, "K must be a multiple of 8");
     auto out = at::empty({M, N}, x.options());
     c10::optional<Tensor> bias_f;
-    if (bias) bias_f = bias->to(at::kFloat).contiguous();
+    if (bias)
+        bias_f = cached_weight_prep(*bias, -1, [&] {
+            return bias->to(at::kFloat).contiguous();
+        });
     ConvParams p{};
     p.x = bptr(x); p.w = bptr(w);
     p.scale = fptr_opt(c10::nullopt, "scale", ones_buf(), N);

@@ -25,9 +25,10 @@ from defer_amd.graph import GraphModel, LayerGraph, from_torch
 XGMI_LINK_GBPS = 153.0
 # Effective sustained compute for weighting conv FLOPs against hop bytes
 # (relative units; ratio is what matters). Calibrated against the measured
-# whole-model ResNet50 bf16 forward on MI355X (~2.9 ms / 64 images at
-# 8.2 GFLOP/image ~= 180 TF effective — tools/convbench.py).
-EFF_TFLOPS = 180.0
+# whole-model ResNet50 bf16 forward on MI355X (~1.83 ms / 64 images at
+# 8.2 GFLOP/image ~= 287 TF effective — bench.py, tile policy v3 +
+# staging rewrite; see profiles/README.md).
+EFF_TFLOPS = 287.0
 
 
 def as_graph_model(model) -> GraphModel:

@@ -142,7 +142,10 @@ class DEFER:
         cfg = self.cfg
         n = len(self.computeNodes)
         if partition_layers is None:
-            partition_layers, stages = auto_partition(model, n)
+            partition_layers, stages = auto_partition(
+                model, n,
+                input_shape=tuple(cfg.input_shape)
+                if cfg.input_shape else (1, 224, 224, 3))
         else:
             stages = partition_model(model, partition_layers)
         if len(stages) != n:

@@ -54,3 +54,36 @@ def test_defer_single_node():
         want = m(xs[0])
     outs, _ = _run_defer(m, [], ["cpu"], xs)
     assert torch.equal(outs[0], want)
+
+
+def test_defer_arbitrary_torch_model():
+    """Capability parity with 'partition any Keras model'
+    (dispatcher.py:107): an ordinary NCHW PyTorch CNN is FX-traced into
+    the LayerGraph IR, auto-partitioned, and streamed through the DEFER
+    orchestrator unchanged."""
+    import torch.nn as nn
+
+    class Net(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.c1 = nn.Conv2d(3, 8, 3, padding=1)
+            self.c2 = nn.Conv2d(8, 8, 3, padding=1)
+            self.c3 = nn.Conv2d(8, 8, 3, padding=1)
+            self.head = nn.Linear(8, 5)
+
+        def forward(self, x):
+            a = torch.relu(self.c1(x))
+            b = torch.relu(self.c2(a))
+            c = torch.relu(self.c3(b) + b)     # skip connection
+            return self.head(c.mean(dim=(2, 3)))
+
+    torch.manual_seed(0)
+    net = Net().eval()
+    xs = [torch.randn(2, 3, 16, 16) for _ in range(3)]
+    with torch.no_grad():
+        want = [net(x) for x in xs]
+    cfg = PipelineConfig(device="cpu", dtype="fp32",
+                         input_shape=(1, 3, 16, 16))
+    outs, _ = _run_defer(net, None, ["cpu", "cpu"], xs, cfg=cfg)
+    for o, w in zip(outs, want):
+        assert torch.allclose(o, w, atol=1e-6)
