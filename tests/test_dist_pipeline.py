@@ -96,3 +96,20 @@ def test_dist_pipeline_with_zfp_lz4_compression():
     the reference's length-prefixed framing). LZ4 is lossless on top of
     ZFP, so the tolerance equals the ZFP-only case."""
     _run(2, ["add_8"], steps=3, compression="zfp+lz4", tol=0.05)
+
+
+def test_node_cli_single_rank(tmp_path):
+    """The node.py-style stage-worker entry runs end to end on CPU
+    (reference node script parity, node.py:126-127)."""
+    import subprocess
+    import sys
+
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1", MASTER_PORT="29779",
+               WORLD_SIZE="1", RANK="0", LOCAL_RANK="0")
+    out = subprocess.run(
+        [sys.executable, "-m", "defer_amd.node", "--device", "cpu",
+         "--items", "2", "--batch", "2", "--report", "1"],
+        capture_output=True, text=True, timeout=300, env=env,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "images/sec" in out.stdout
