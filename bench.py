@@ -123,6 +123,10 @@ def main():
         if args.device == "cuda":
             torch.cuda.synchronize()
 
+    # establish the world communicator collectively BEFORE any p2p op
+    # (lazy per-pair init ordering is then irrelevant on RCCL)
+    barrier_sync()
+
     # ---- warmup (fills pipeline, triggers graph capture)
     pipe.run(args.warmup, feed=feed, collect=collect)
     barrier_sync()

@@ -78,6 +78,8 @@ def main():
     dtype = torch.bfloat16 if args.device == "cuda" else torch.float32
     x = torch.randn(B, 224, 224, 3, device=dev, dtype=dtype)
 
+    dist.barrier()   # collective comm init before any p2p op
+
     state = {"n": 0, "t0": time.perf_counter()}
 
     def collect(k, y):

@@ -113,3 +113,9 @@ def test_node_cli_single_rank(tmp_path):
         cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
     assert out.returncode == 0, out.stderr[-2000:]
     assert "images/sec" in out.stdout
+
+
+def test_eight_stage_auto_cuts_gloo():
+    """The headline 8-stage chain topology (7 relay hops
+    + result return), auto-partitioned, world_size 8 on gloo."""
+    _run(8, None, steps=3)
