@@ -286,7 +286,8 @@ std::tuple<int64_t, int64_t, int64_t> field_shape(const Tensor& t) {
     return {d0, sz[nd - 2], sz[nd - 1]};
 }
 
-Tensor zfp_encode(Tensor x, int64_t rate, c10::optional<Tensor> out) {
+Tensor zfp_encode(Tensor x, int64_t rate, c10::optional<Tensor> out,
+                  int64_t phases = 3) {
     TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x must be GPU contig");
     bool bf16 = x.scalar_type() == at::kBFloat16;
     TORCH_CHECK(bf16 || x.scalar_type() == at::kFloat, "bf16/f32 only");
@@ -304,7 +305,7 @@ Tensor zfp_encode(Tensor x, int64_t rate, c10::optional<Tensor> out) {
     }
     defer_hip::launch_zfp_encode(bptr(x), bptr_mut(o), bf16, (int)d0,
                                  (int)d1, (int)d2, (int)rate,
-                                 cur_stream());
+                                 cur_stream(), (int)phases);
     return o;
 }
 
@@ -364,7 +365,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("maxpool2d", &maxpool2d);
     m.def("global_avg_pool", &global_avg_pool);
     m.def("zfp_encode", &zfp_encode, py::arg("x"), py::arg("rate"),
-          py::arg("out") = py::none());
+          py::arg("out") = py::none(), py::arg("phases") = 3);
     m.def("zfp_decode", &zfp_decode);
     m.def("lz4_compress", &lz4_compress);
     m.def("lz4_decompress", &lz4_decompress);

@@ -138,224 +138,297 @@ struct BitReader {
 };
 
 // ---------------------------------------------------------------------------
-template <bool BF16_IN>
-__global__ void zfp_encode_kernel(const void* __restrict__ xv,
-                                  u32* __restrict__ out, int d0, int d1,
-                                  int d2, int b0, int b1, int b2,
-                                  int rate) {
-    const int lane = threadIdx.x % WAVE;
-    const int wavei = threadIdx.x / WAVE;
+// Two-phase encode (one kernel, LDS hand-off). The bit stream of a block
+// is inherently serial, so phase 1 (wave-per-block: gather, lift, plane
+// ballots -> LDS) runs at wave parallelism and phase 2 gives every
+// THREAD one whole block to serialize from its LDS plane words — 256
+// serializers per workgroup instead of one lane per wave. Bit-exact to
+// the single-lane version: the emission code is identical, only who
+// executes it changed.
+#define CGRP 256                     // blocks per workgroup round
+
+template <bool BF16_IN, int PH = 3>   // PH: bit0=transform, bit1=serialize
+__global__ __launch_bounds__(CGRP, 2) void zfp_encode_kernel(
+    const void* __restrict__ xv, u32* __restrict__ out, int d0, int d1,
+    int d2, int b0, int b1, int b2, int rate) {
+    const int tid = threadIdx.x;
+    const int lane = tid % WAVE;
+    const int wavei = tid / WAVE;
     const int nwaves = blockDim.x / WAVE;
     const long nblocks = (long)b0 * b1 * b2;
     const int li = lane >> 4, lj = (lane >> 2) & 3, lk = lane & 3;
     const int wpb = rate * 2;                   // u32 words per block
 
-    for (long blk = (long)blockIdx.x * nwaves + wavei; blk < nblocks;
-         blk += (long)gridDim.x * nwaves) {
-        int bk = (int)(blk % b2);
-        long t = blk / b2;
-        int bj = (int)(t % b1);
-        int bi = (int)(t / b1);
-        int gi = bi * 4 + li; if (gi > d0 - 1) gi = d0 - 1;
-        int gj = bj * 4 + lj; if (gj > d1 - 1) gj = d1 - 1;
-        int gk = bk * 4 + lk; if (gk > d2 - 1) gk = d2 - 1;
-        long idx = ((long)gi * d1 + gj) * d2 + gk;
-        float v = BF16_IN ? bf2f(((const bf16*)xv)[idx])
-                          : ((const float*)xv)[idx];
+    __shared__ u64 s_planes[CGRP][PLANES + 1];  // +1: LDS bank pad
+    __shared__ u32 s_hdr[CGRP];                 // 0 = zero block
 
-        // max |v| over the wave
-        float av = fabsf(v);
+    // incremental block coordinates for this wave's stride-nwaves walk
+    // (no runtime division in the loop) and a PF-deep gather pipeline so
+    // one block's load latency hides under the previous blocks' work
+    constexpr int PF = 4;
+    for (long base = (long)blockIdx.x * CGRP; base < nblocks;
+         base += (long)gridDim.x * CGRP) {
+        // ---- phase 1: each wave transforms CGRP/nwaves blocks
+        if (PH & 1) {
+            long blk0 = base + wavei;
+            int bk = (int)(blk0 % b2);
+            long t = blk0 / b2;
+            int bj = (int)(t % b1);
+            int bi = (int)(t / b1);
+            // raw-typed prefetch ring: the convert happens at USE so the
+            // compiler leaves the loads in flight (counted waits)
+            bf16 braw[PF];
+            float fraw[PF];
+            auto issue = [&](int d) {           // load block slot d ahead
+                int gi = bi * 4 + li; if (gi > d0 - 1) gi = d0 - 1;
+                int gj = bj * 4 + lj; if (gj > d1 - 1) gj = d1 - 1;
+                int gk = bk * 4 + lk; if (gk > d2 - 1) gk = d2 - 1;
+                long idx = ((long)gi * d1 + gj) * d2 + gk;
+                if (BF16_IN)
+                    braw[d] = ((const bf16*)xv)[idx];
+                else
+                    fraw[d] = ((const float*)xv)[idx];
+                bk += nwaves;                   // advance block coords
+                while (bk >= b2) { bk -= b2; ++bj; }
+                while (bj >= b1) { bj -= b1; ++bi; }
+            };
+            // blocks this wave owns in the round
+            long avail = nblocks - (base + wavei);
+            int total = avail <= 0 ? 0
+                        : (int)((avail < CGRP - wavei ? avail
+                                                      : CGRP - wavei)
+                                + nwaves - 1) / nwaves;
 #pragma unroll
-        for (int off = 32; off > 0; off >>= 1)
-            av = fmaxf(av, __shfl_xor(av, off));
-
-        u32* bout = out + blk * wpb;
-        if (!(av > 0.f) || !isfinite(av)) {     // zero/non-finite block
-            if (lane == 0)
-                for (int w = 0; w < wpb; ++w) bout[w] = 0;
-            continue;
-        }
-        int emax;
-        frexpf(av, &emax);
-        // quantize (f32 product, round-half-even — matches numpy rint)
-        int q = (int)rintf(v * ldexpf(1.0f, QBITS - emax));
-        // transform along k (stride 1), j (4), i (16)
-        q = fwd_axis(q, lane, 1);
-        q = fwd_axis(q, lane, 4);
-        q = fwd_axis(q, lane, 16);
-        // sequency reorder: stream slot `lane` holds coef ZPERM[lane]
-        q = __shfl(q, ZPERM[lane]);
-        // negabinary
-        u32 u = ((u32)q + NBMASK) ^ NBMASK;
-
-        BitWriter wr{bout, 0u, 0, rate * 64};
-        if (lane == 0)
-            wr.put_bits((1u << 15) | ((u32)(emax + 256) & 0x1FFu),
-                        HDR_BITS);
-        int n = 0;
-        for (int p = PLANES - 1; p >= 0 && wr.budget > 0; --p) {
-            u64 x = __ballot((u >> p) & 1);
-            if (lane == 0) {
-                // significant prefix
-                wr.put_bits(x & ((n < 64) ? ((1ull << n) - 1ull)
-                                          : ~0ull), n);
-                x >>= n;
-                // group tests, run-wise: each run is tz zeros then a
-                // one = the word (1 << tz) over tz+1 bits (bit-identical
-                // to per-bit emission; put_bits truncates at the budget
-                // exactly like per-bit puts would)
-                while (n < 64) {
-                    if (wr.budget <= 0) break;
-                    int has = (x != 0);
-                    wr.put1(has);
-                    if (!has) break;
-                    int tz = __builtin_ctzll(x);
-                    int emit = tz + 1;
-                    wr.put_bits(1ull << tz, emit);
-                    x = (emit >= 64) ? 0 : (x >> emit);
-                    n += emit;
+            for (int d = 0; d < PF; ++d)
+                if (d < total) issue(d);
+            for (int ii = 0; ii < total; ++ii) {
+                int s = wavei + ii * nwaves;
+                float v = BF16_IN ? bf2f(braw[ii % PF])
+                                  : fraw[ii % PF];
+                if (ii + PF < total) issue((ii + PF) % PF);
+                float av = fabsf(v);
+#pragma unroll
+                for (int off = 32; off > 0; off >>= 1)
+                    av = fmaxf(av, __shfl_xor(av, off));
+                if (!(av > 0.f) || !isfinite(av)) {
+                    if (lane == 0) s_hdr[s] = 0;
+                    continue;
                 }
+                int emax;
+                frexpf(av, &emax);
+                // quantize (f32 product, round-half-even = numpy rint)
+                int q = (int)rintf(v * ldexpf(1.0f, QBITS - emax));
+                q = fwd_axis(q, lane, 1);
+                q = fwd_axis(q, lane, 4);
+                q = fwd_axis(q, lane, 16);
+                q = __shfl(q, ZPERM[lane]);
+                u32 u = ((u32)q + NBMASK) ^ NBMASK;
+                if (lane == 0)
+                    s_hdr[s] = (1u << 15) | ((u32)(emax + 256) & 0x1FFu);
+                // lane p collects plane p's ballot word; ONE coalesced
+                // 30-lane ds_write replaces 30 single-lane writes (the
+                // single-lane stores serialized the whole phase on the
+                // LDS pipe: measured 725us -> see profiles/README.md)
+                u64 myw = 0;
+#pragma unroll
+                for (int p = 0; p < PLANES; ++p) {
+                    u64 x = __ballot((u >> p) & 1);
+                    if (lane == p) myw = x;
+                }
+                if (lane < PLANES) s_planes[s][lane] = myw;
             }
-            // all lanes need n for the next plane's ballot bookkeeping?
-            // n only lives on lane 0; broadcast not needed (only lane 0
-            // uses it), but the loop break on budget must be uniform:
-            n = __shfl(n, 0);
-            int bud = __shfl(wr.budget, 0);
-            if (lane != 0) wr.budget = bud;
         }
-        if (lane == 0) wr.finish(bout + wpb);
+        __syncthreads();
+        // ---- phase 2: thread tid serializes block base+tid
+        long blk = base + tid;
+        if ((PH & 2) && blk < nblocks) {
+            u32* bout = out + blk * wpb;
+            u32 hdr = s_hdr[tid];
+            if (hdr == 0) {
+                for (int w = 0; w < wpb; ++w) bout[w] = 0;
+            } else {
+                BitWriter wr{bout, 0u, 0, rate * 64};
+                wr.put_bits(hdr, HDR_BITS);
+                int n = 0;
+                for (int p = PLANES - 1; p >= 0 && wr.budget > 0; --p) {
+                    u64 x = s_planes[tid][p];
+                    // significant prefix
+                    wr.put_bits(x & ((n < 64) ? ((1ull << n) - 1ull)
+                                              : ~0ull), n);
+                    x >>= n;
+                    // group tests, run-wise (bit-identical to per-bit
+                    // emission; put_bits truncates at the budget)
+                    while (n < 64) {
+                        if (wr.budget <= 0) break;
+                        int has = (x != 0);
+                        wr.put1(has);
+                        if (!has) break;
+                        int tz = __builtin_ctzll(x);
+                        int emit = tz + 1;
+                        wr.put_bits(1ull << tz, emit);
+                        x = (emit >= 64) ? 0 : (x >> emit);
+                        n += emit;
+                    }
+                }
+                wr.finish(bout + wpb);
+            }
+        }
+        __syncthreads();               // LDS reused next round
     }
 }
 
+// Two-phase decode, mirror of the encoder: phase 1 gives every THREAD
+// one block's bit stream to parse into LDS plane words (256 parallel
+// parsers); phase 2 runs the inverse transform wave-per-block from LDS.
+// Truncation endgame differs from the per-bit reference only in zero
+// bits / internal n, so reconstruction is identical.
 template <bool BF16_OUT>
-__global__ void zfp_decode_kernel(const u32* __restrict__ wire,
-                                  void* __restrict__ yv, int d0, int d1,
-                                  int d2, int b0, int b1, int b2,
-                                  int rate) {
-    const int lane = threadIdx.x % WAVE;
-    const int wavei = threadIdx.x / WAVE;
+__global__ __launch_bounds__(CGRP, 2) void zfp_decode_kernel(
+    const u32* __restrict__ wire, void* __restrict__ yv, int d0, int d1,
+    int d2, int b0, int b1, int b2, int rate) {
+    const int tid = threadIdx.x;
+    const int lane = tid % WAVE;
+    const int wavei = tid / WAVE;
     const int nwaves = blockDim.x / WAVE;
     const long nblocks = (long)b0 * b1 * b2;
     const int li = lane >> 4, lj = (lane >> 2) & 3, lk = lane & 3;
     const int wpb = rate * 2;
 
-    for (long blk = (long)blockIdx.x * nwaves + wavei; blk < nblocks;
-         blk += (long)gridDim.x * nwaves) {
-        int bk = (int)(blk % b2);
-        long t = blk / b2;
-        int bj = (int)(t % b1);
-        int bi = (int)(t / b1);
-        int gi = bi * 4 + li;
-        int gj = bj * 4 + lj;
-        int gk = bk * 4 + lk;
-        bool valid = (gi < d0) && (gj < d1) && (gk < d2);
-        long idx = ((long)gi * d1 + gj) * d2 + gk;
+    __shared__ u64 s_planes[CGRP][PLANES + 1];  // +1: LDS bank pad
+    __shared__ u32 s_hdr[CGRP];
 
-        const u32* bin = wire + blk * wpb;
-        // lane 0 parses word-wise; plane words broadcast to all lanes.
-        // Truncation endgame differs from the per-bit reference only in
-        // zero bits / internal n, so reconstruction is identical.
-        BitReader rd;
-        rd.init(bin, rate * 64);
-        u32 hdr = 0;
-        if (lane == 0) hdr = (u32)rd.take(HDR_BITS);
-        hdr = __shfl(hdr, 0);
-        float outv = 0.f;
-        if (hdr >> 15) {
-            int emax = (int)(hdr & 0x1FFu) - 256;
-            u32 u = 0;
-            int n = 0;
-            int done = 0;
-            for (int p = PLANES - 1; p >= 0; --p) {
-                u64 x = 0;
-                if (lane == 0) {
+    for (long base = (long)blockIdx.x * CGRP; base < nblocks;
+         base += (long)gridDim.x * CGRP) {
+        // ---- phase 1: thread tid parses block base+tid
+        long pblk = base + tid;
+        if (pblk < nblocks) {
+            BitReader rd;
+            rd.init(wire + pblk * wpb, rate * 64);
+            u32 hdr = (u32)rd.take(HDR_BITS);
+            s_hdr[tid] = hdr;
+            if (hdr >> 15) {
+                int n = 0;
+                for (int p = PLANES - 1; p >= 0; --p) {
+                    u64 x = 0;
                     if (rd.empty()) {
-                        done = 1;
-                    } else {
-                        x = rd.take(n);               // significant prefix
-                        while (n < 64) {              // group runs
-                            if (rd.empty()) break;
-                            if (!rd.take(1)) break;   // group test
-                            // scan zeros (may span window refills) until
-                            // the run's 1 bit
-                            bool found = false;
-                            while (n < 64) {
-                                rd.refill();
-                                int lim = 64 - n;
-                                int avail =
-                                    rd.nwin < lim ? rd.nwin : lim;
-                                if (avail == 0) break;  // stream end
-                                u64 w = rd.win &
-                                        ((avail >= 64)
-                                             ? ~0ull
-                                             : ((1ull << avail) - 1ull));
-                                int tz = w ? __builtin_ctzll(w) : avail;
-                                if (tz < avail) {
-                                    rd.take(tz + 1);
-                                    x |= 1ull << (n + tz);
-                                    n += tz + 1;
-                                    found = true;
-                                    break;
-                                }
-                                rd.take(avail);       // all zeros
-                                n += avail;
-                                if (rd.empty()) break;
-                            }
-                            if (!found) break;
-                        }
+                        // remaining planes are zero
+                        for (int pp = p; pp >= 0; --pp)
+                            s_planes[tid][pp] = 0;
+                        break;
                     }
+                    x = rd.take(n);                   // significant prefix
+                    while (n < 64) {                  // group runs
+                        if (rd.empty()) break;
+                        if (!rd.take(1)) break;       // group test
+                        // scan zeros (may span window refills) until the
+                        // run's 1 bit
+                        bool found = false;
+                        while (n < 64) {
+                            rd.refill();
+                            int lim = 64 - n;
+                            int avail = rd.nwin < lim ? rd.nwin : lim;
+                            if (avail == 0) break;    // stream end
+                            u64 w = rd.win &
+                                    ((avail >= 64)
+                                         ? ~0ull
+                                         : ((1ull << avail) - 1ull));
+                            int tz = w ? __builtin_ctzll(w) : avail;
+                            if (tz < avail) {
+                                rd.take(tz + 1);
+                                x |= 1ull << (n + tz);
+                                n += tz + 1;
+                                found = true;
+                                break;
+                            }
+                            rd.take(avail);           // all zeros
+                            n += avail;
+                            if (rd.empty()) break;
+                        }
+                        if (!found) break;
+                    }
+                    s_planes[tid][p] = x;
                 }
-                done = __shfl(done, 0);
-                if (done) break;
-                x = __shfl(x, 0);
-                n = __shfl(n, 0);
-                u |= (u32)((x >> lane) & 1) << p;
             }
-            int q = (int)((u ^ NBMASK) - NBMASK);   // negabinary inverse
-            // inverse sequency: coef position `lane` from stream slot
-            // ZIPERM[lane]
-            q = __shfl(q, ZIPERM[lane]);
-            // inverse transform: axes i, j, k
-            q = inv_axis(q, lane, 16);
-            q = inv_axis(q, lane, 4);
-            q = inv_axis(q, lane, 1);
-            outv = ldexpf((float)q, emax - QBITS);
         }
-        if (valid) {
-            if (BF16_OUT)
-                ((bf16*)yv)[idx] = f2bf(outv);
-            else
-                ((float*)yv)[idx] = outv;
+        __syncthreads();
+        // ---- phase 2: each wave reconstructs CGRP/nwaves blocks
+        for (int s = wavei; s < CGRP; s += nwaves) {
+            long blk = base + s;
+            if (blk >= nblocks) break;
+            int bk = (int)(blk % b2);
+            long t = blk / b2;
+            int bj = (int)(t % b1);
+            int bi = (int)(t / b1);
+            int gi = bi * 4 + li;
+            int gj = bj * 4 + lj;
+            int gk = bk * 4 + lk;
+            bool valid = (gi < d0) && (gj < d1) && (gk < d2);
+            long idx = ((long)gi * d1 + gj) * d2 + gk;
+            u32 hdr = s_hdr[s];
+            float outv = 0.f;
+            if (hdr >> 15) {
+                int emax = (int)(hdr & 0x1FFu) - 256;
+                u32 u = 0;
+#pragma unroll
+                for (int p = 0; p < PLANES; ++p)
+                    u |= (u32)((s_planes[s][p] >> lane) & 1) << p;
+                int q = (int)((u ^ NBMASK) - NBMASK);  // negabinary inv
+                q = __shfl(q, ZIPERM[lane]);
+                q = inv_axis(q, lane, 16);
+                q = inv_axis(q, lane, 4);
+                q = inv_axis(q, lane, 1);
+                outv = ldexpf((float)q, emax - QBITS);
+            }
+            if (valid) {
+                if (BF16_OUT)
+                    ((bf16*)yv)[idx] = f2bf(outv);
+                else
+                    ((float*)yv)[idx] = outv;
+            }
         }
+        __syncthreads();               // LDS reused next round
     }
 }
 
 // ---------------------------------------------------------------------------
 namespace defer_hip {
 
-static int codec_grid(long nblocks, int waves_per_wg) {
-    long wgs = (nblocks + waves_per_wg - 1) / waves_per_wg;
+static int codec_grid(long nblocks, int blocks_per_wg) {
+    long wgs = (nblocks + blocks_per_wg - 1) / blocks_per_wg;
     return (int)(wgs < 2048 ? (wgs > 0 ? wgs : 1) : 2048);
 }
 
 void launch_zfp_encode(const void* x, void* out, bool bf16_in, int d0,
-                       int d1, int d2, int rate, hipStream_t s) {
+                       int d1, int d2, int rate, hipStream_t s,
+                       int phases) {
     int b0 = (d0 + 3) / 4, b1 = (d1 + 3) / 4, b2 = (d2 + 3) / 4;
     long nblocks = (long)b0 * b1 * b2;
-    dim3 grid(codec_grid(nblocks, 4)), block(256);
-    if (bf16_in)
-        hipLaunchKernelGGL((zfp_encode_kernel<true>), grid, block, 0, s,
-                           x, (u32*)out, d0, d1, d2, b0, b1, b2, rate);
-    else
-        hipLaunchKernelGGL((zfp_encode_kernel<false>), grid, block, 0, s,
-                           x, (u32*)out, d0, d1, d2, b0, b1, b2, rate);
+    dim3 grid(codec_grid(nblocks, 256)), block(256);
+#define ENC_DISPATCH(BF)                                                   \
+    do {                                                                   \
+        if (phases == 1)                                                   \
+            hipLaunchKernelGGL((zfp_encode_kernel<BF, 1>), grid, block, 0, \
+                               s, x, (u32*)out, d0, d1, d2, b0, b1, b2,    \
+                               rate);                                      \
+        else if (phases == 2)                                              \
+            hipLaunchKernelGGL((zfp_encode_kernel<BF, 2>), grid, block, 0, \
+                               s, x, (u32*)out, d0, d1, d2, b0, b1, b2,    \
+                               rate);                                      \
+        else                                                               \
+            hipLaunchKernelGGL((zfp_encode_kernel<BF, 3>), grid, block, 0, \
+                               s, x, (u32*)out, d0, d1, d2, b0, b1, b2,    \
+                               rate);                                      \
+    } while (0)
+    if (bf16_in) ENC_DISPATCH(true);
+    else ENC_DISPATCH(false);
+#undef ENC_DISPATCH
 }
 
 void launch_zfp_decode(const void* wire, void* y, bool bf16_out, int d0,
                        int d1, int d2, int rate, hipStream_t s) {
     int b0 = (d0 + 3) / 4, b1 = (d1 + 3) / 4, b2 = (d2 + 3) / 4;
     long nblocks = (long)b0 * b1 * b2;
-    dim3 grid(codec_grid(nblocks, 4)), block(256);
+    dim3 grid(codec_grid(nblocks, 256)), block(256);
     if (bf16_out)
         hipLaunchKernelGGL((zfp_decode_kernel<true>), grid, block, 0, s,
                            (const u32*)wire, y, d0, d1, d2, b0, b1, b2,

@@ -47,8 +47,11 @@ void launch_gap(const void* x, void* y, int NB, int HW, int C,
                 hipStream_t s);
 
 // fixed-rate ZFP-style codec (see csrc/codec.hip / ops/zfp_ref.py)
+// phases: 3 = full encode; 1/2 run only the transform / serialize phase
+// (perf-bisection harness for tools/codecbench.py --phases)
 void launch_zfp_encode(const void* x, void* out, bool bf16_in, int d0,
-                       int d1, int d2, int rate, hipStream_t s);
+                       int d1, int d2, int rate, hipStream_t s,
+                       int phases = 3);
 void launch_zfp_decode(const void* wire, void* y, bool bf16_out, int d0,
                        int d1, int d2, int rate, hipStream_t s);
 

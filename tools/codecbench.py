@@ -4,6 +4,7 @@ import os, sys
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 import torch
 from defer_amd.ops import codec
+import defer_amd._hip_ops as hip
 
 SHAPES = [  # ResNet50 boundary activations at batch 64 (bf16)
     ("add_2 56x56x256", (64, 56, 56, 256)),
@@ -28,11 +29,23 @@ def bench(shape, rate, iters=30):
     torch.cuda.synchronize()
     enc_us = t0.elapsed_time(t1) * 1e3 / iters
     dec_us = t1.elapsed_time(t2) * 1e3 / iters
+    # phase bisection: transform-only / serialize-only encodes
+    ph = {}
+    for phase in (1, 2):
+        torch.cuda.synchronize()
+        a, b = torch.cuda.Event(True), torch.cuda.Event(True)
+        a.record()
+        for _ in range(iters):
+            hip.zfp_encode(x, rate, w, phase)
+        b.record()
+        torch.cuda.synchronize()
+        ph[phase] = a.elapsed_time(b) * 1e3 / iters
     mb = x.numel() * 2 / 1e6
-    return enc_us, dec_us, mb, w.numel() / 1e6
+    return enc_us, dec_us, mb, w.numel() / 1e6, ph
 
 for name, shape in SHAPES:
     for rate in (4, 8):
-        e, d, mb, wmb = bench(shape, rate)
+        e, d, mb, wmb, ph = bench(shape, rate)
         print(f"{name:20s} rate{rate:2d}: enc {e:7.1f} us ({mb/e*1e3:6.0f} GB/s) "
-              f"dec {d:7.1f} us ({mb/d*1e3:6.0f} GB/s)  {mb:.0f}->{wmb:.0f} MB")
+              f"dec {d:7.1f} us ({mb/d*1e3:6.0f} GB/s)  {mb:.0f}->{wmb:.0f} MB"
+              f"  [p1 {ph[1]:.1f}us p2 {ph[2]:.1f}us]")
