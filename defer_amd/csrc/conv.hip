@@ -80,8 +80,8 @@ __device__ __forceinline__ void glds16(const bf16* src, bf16* lds_base) {
 // registers for 3-4 co-resident blocks so more independent load streams
 // and epilogues overlap.
 template <int ACT, bool HAS_RES, int AMODE, bool B_PERSIST, int BM,
-          int BN, int DA, int DB, int WPS = 2>
-__global__ __launch_bounds__(NTHREADS, WPS) void conv_igemm_kernel(
+          int BN, int DA, int DB, int WPS = 2, int TPB = NTHREADS>
+__global__ __launch_bounds__(TPB, WPS) void conv_igemm_kernel(
     ConvParams p) {
     const bf16* __restrict__ X = (const bf16*)p.x;
     const bf16* __restrict__ Wt = (const bf16*)p.w;
@@ -90,12 +90,13 @@ __global__ __launch_bounds__(NTHREADS, WPS) void conv_igemm_kernel(
     bf16* __restrict__ OUT = (bf16*)p.out;
 
     // wave tiling: WMW waves along m (32 rows each), WNW along n
-    constexpr int WMW = BM / 32;                // 2 or 4
-    constexpr int WNW = 4 / WMW;                // 2 or 1
+    constexpr int NW = TPB / WAVE;              // waves per block
+    constexpr int WMW = BM / 32;                // m-slabs of 32 rows
+    constexpr int WNW = NW / WMW;               // waves along n
     constexpr int WN = BN / WNW;                // wave n-width
     constexpr int NI = WN / 16;                 // B frags per wave
-    constexpr int ACH = BM / 32;                // A chunks per thread
-    constexpr int BCH = BN / 32;                // B chunks per thread
+    constexpr int ACH = BM * 8 / TPB;           // A chunks per thread
+    constexpr int BCH = BN * 8 / TPB;           // B chunks per thread
     constexpr int OPS = ACH + (B_PERSIST ? 0 : BCH);
     constexpr bool ASYM = !B_PERSIST && (DA != DB);
 
@@ -328,7 +329,7 @@ __global__ __launch_bounds__(NTHREADS, WPS) void conv_igemm_kernel(
     // consumed; RH output rows per round (RH*BN f32 = one A buffer).
     constexpr int RH = (BM * BK * 2 / 4) / BN;   // 64/32/16 rows per round
     constexpr int ROUNDS = BM / RH;
-    constexpr int CPR = RH * BN / 4 / NTHREADS;  // 16-B chunks per thread
+    constexpr int CPR = RH * BN / 4 / TPB;       // 16-B chunks per thread
     auto epilogue = [&](int mt, int abuf) {
         float* scratch = (float*)(A0 + abuf * BM * BK);
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -359,13 +360,13 @@ __global__ __launch_bounds__(NTHREADS, WPS) void conv_igemm_kernel(
                 // 8-col chunks: 16-B residual loads and 16-B bf16 stores
                 // (8-B stores made the 1x1 layers store-issue-bound, cf.
                 // MI355X_MICROARCH store-tail note)
-                constexpr int CP8 = RH * BN / 8 / NTHREADS;
+                constexpr int CP8 = RH * BN / 8 / TPB;
                 f32x4 v4[CP8][2];
                 bf16x8 rv[CP8];
                 long off[CP8];
 #pragma unroll
                 for (int i = 0; i < CP8; ++i) {
-                    int chunk = tid + i * NTHREADS;
+                    int chunk = tid + i * TPB;
                     int rl = chunk / (BN / 8);
                     int c8 = (chunk % (BN / 8)) * 8;
                     int cs = c8 ^ (((rl >> 2) & 3) << 4);
@@ -393,7 +394,7 @@ __global__ __launch_bounds__(NTHREADS, WPS) void conv_igemm_kernel(
             } else {
 #pragma unroll
                 for (int i = 0; i < CPR; ++i) {
-                    int chunk = tid + i * NTHREADS;
+                    int chunk = tid + i * TPB;
                     int rl = chunk / (BN / 4);
                     int c4 = (chunk % (BN / 4)) * 4;
                     int cs = c4 ^ (((rl >> 2) & 3) << 4);
@@ -686,6 +687,12 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
     const bool smallgemm =
         (force_small || (!legacy && ((one1 && !wide) || small_conv))) &&
         (long)((p.M + 63) / 64) * ((p.Cout + 63) / 64) >= 768;
+    // NOTE: a BM=256 x BN=128 512-thread 1-block/CU config was measured
+    // for the big-M deep-K 3x3s (to halve B re-staging: the BM128 D2
+    // config demands ~117 B/cyc/CU of L2 tile traffic vs ~56 available)
+    // and lost 10-24% — one resident block loses the cross-block
+    // epilogue/prologue overlap that two independent blocks provide.
+    const bool big = false;
     if (smallgemm) {
         BNsel = 64;
         BMsel = 64;
@@ -713,6 +720,7 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
     }
     dim3 grid(gx, ny);
     dim3 block(NTHREADS);
+    (void)big;
 
 #define DISPATCH_TILE(A, R, G, BP, BMv, BNv, D...)                        \
     hipLaunchKernelGGL(                                                   \
