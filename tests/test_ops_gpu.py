@@ -212,3 +212,33 @@ def resnet50_copy_like(m):
 
     _t.manual_seed(0)
     return _r()
+
+
+@pytest.mark.gpu
+def test_stage_stats_event_timer_gpu():
+    """hipEvent per-stage timing (utils/trace.py) measures real device
+    time on a single-stage DistPipeline run."""
+    import os
+
+    import torch.distributed as dist
+
+    from defer_amd.config import PipelineConfig
+    from defer_amd.models import resnet50
+    from defer_amd.parallel.pipeline import DistPipeline
+
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29784")
+        dist.init_process_group("gloo", rank=0, world_size=1)
+    cfg = PipelineConfig(device="cuda", dtype="bf16", batch_size=8,
+                         use_hip_graphs=False, backend="gloo",
+                         log_stage_stats=True)
+    pipe = DistPipeline(resnet50(), cfg, (8, 224, 224, 3),
+                        device=torch.device("cuda", 0))
+    pipe.run(3, feed=lambda k: torch.randn(
+        8, 224, 224, 3, device="cuda", dtype=torch.bfloat16),
+        collect=lambda k, y: None)
+    torch.cuda.synchronize()
+    assert pipe.stats.items == 3
+    ms = pipe.stats.compute_ms
+    assert 0.05 < ms < 5000, ms   # real device time, not zero
