@@ -562,8 +562,342 @@ __global__ __launch_bounds__(TPB, WPS) void conv_igemm_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// Channel pad, NHWC: [rows, C] -> [rows, C8] zero-padded (stem Cin=3 -> 8;
-// also pads OHWI weights viewed as [Cout*R*S, Cin]).
+// Window-reuse 3x3/s1/p1 convolution (AMODE_WIN as its own kernel).
+//
+// The implicit-GEMM kernel re-stages the SAME input bytes once per (r,s)
+// k-tile — 9x for a 3x3 — and its BM128/BN128 configs demand ~117
+// B/cyc/CU of L2 tile traffic vs ~56 available. Here a 2D output tile
+// (TH=8 x TW=16 pixels = BM 128) stages its (TH+2)x(TW+2)x64ch input
+// WINDOW to LDS once per 64-channel block and the nine (r,s) compute
+// phases slide inside LDS: A-staging drops 9x -> ~1.1x (halo).
+//
+// LDS window layout [ihl][c8][iwl] in 16-B chunks, so an MFMA A-read's
+// 16 lanes (cw = lane&15 consecutive) hit consecutive chunks —
+// conflict-free ds_read_b128 — and the (r, s) slide is one scalar chunk
+// offset (r*8*WW + s). The next window (cb+1, crossing m-tiles) is
+// staged in 6 glds slices interleaved with phases 0..5, so the counted
+// vmcnt keeps 3+ phases of load-latency cover without a FIFO conflict
+// with the per-phase B stages (order: B first, then the slice).
+#define WTH 8
+#define WTW 16
+#define WWH (WTH + 2)                  // window rows
+#define WWW (WTW + 2)                  // window cols
+#define WCHUNKS (WWH * 8 * WWW)        // 16-B chunks per window (1440)
+#define WSLOTS 24                      // glds slots (4 waves x 6)
+
+template <int ACT, bool HAS_RES, int BN>
+__global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
+    ConvParams p) {
+    constexpr int BM = WTH * WTW;               // 128
+    constexpr int DB = 2;
+    constexpr int NI = BN / 16;                 // all 4 waves along m
+    constexpr int BCH = BN / 32;
+    const bf16* __restrict__ X = (const bf16*)p.x;
+    const bf16* __restrict__ Wt = (const bf16*)p.w;
+    const bf16* __restrict__ Z = (const bf16*)p.zbuf;
+    const bf16* __restrict__ RES = (const bf16*)p.res;
+    bf16* __restrict__ OUT = (bf16*)p.out;
+
+    __shared__ __attribute__((aligned(16)))
+    bf16 lds[(2 * WCHUNKS * 8) + DB * BN * BK];
+    bf16* W0 = lds;                              // two window buffers
+    bf16* B0 = lds + 2 * WCHUNKS * 8;
+
+    const int tid = threadIdx.x;
+    const int wave = tid / WAVE;
+    const int lane = tid % WAVE;
+    const int wm = wave;                         // WNW = 1
+    const int n0 = blockIdx.y * BN;
+    const int lo16 = lane & 15;
+    const int hi4 = lane >> 4;
+    const int ncb = p.Cin / 64;
+    const int ntw = (p.OW + WTW - 1) / WTW;
+    const int nth = (p.OH + WTH - 1) / WTH;
+    const int mtiles = p.NB * nth * ntw;
+    const long nwin = (long)mtiles * ncb;
+
+    // B lane offsets (scalar-base + 32-bit lane offset form)
+    int b_row[BCH], b_k8[BCH];
+    u32 b_voff[BCH];
+#pragma unroll
+    for (int i = 0; i < BCH; ++i) {
+        int chunk = wave * (BCH * 64) + i * 64 + lane;
+        b_row[i] = chunk / KCH;
+        b_k8[i] = swz(b_row[i], chunk % KCH);
+        int n = n0 + b_row[i];
+        if (n >= p.Cout) n = p.Cout - 1;
+        b_voff[i] = (u32)((u64)n * (u64)p.K) + (u32)(b_k8[i] * 8);
+    }
+    auto stage_b = [&](int buf, int cb, int rs) {
+        bf16* B = B0 + buf * BN * BK;
+        const bf16* bbase = Wt + (long)rs * p.Cin + cb * 64;
+#pragma unroll
+        for (int i = 0; i < BCH; ++i)
+            glds16(bbase + b_voff[i],
+                   B + (wave * (BCH * 64) + i * 64) * 8);
+    };
+
+
+    // ---- window cursor: stages window (s_mt, s_cb). Per-window setup
+    // keeps one SCALAR base (image + channel block) plus per-lane
+    // 32-bit offsets and a 6-bit validity mask — the per-slice address
+    // is rebuilt in stage_win_slice (register pressure: the first cut
+    // of this kernel kept 6 pointers per lane and spilled).
+    int s_mt = blockIdx.x, s_cb = 0;
+    int w_loff[6];                    // per-lane element offsets
+    u32 w_okm = 0;                    // bit j = slice j in bounds
+    const bf16* w_base = X;           // + nb*H*W*Cin + cb*64 (uniform)
+    int dst_buf = 0;                  // LDS buffer this window lands in
+    long w_count = 0;                 // windows staged so far (parity)
+    int c_nb = 0, c_oh0 = 0, c_ow0 = 0;   // compute-tile coords
+    auto win_setup = [&]() {          // for (s_mt, s_cb)
+        int mt = s_mt;
+        int tw = mt % ntw;
+        int t = mt / ntw;
+        int th = t % nth;
+        int nb = t / nth;
+        int oh0 = th * WTH, ow0 = tw * WTW;
+        w_base = X + (long)nb * p.H * p.W * p.Cin + s_cb * 64;
+        w_okm = 0;
+#pragma unroll
+        for (int j = 0; j < 6; ++j) {
+            // slot j covers chunk (j*4 + wave)*64 + lane
+            int chunk = (j * 4 + wave) * 64 + lane;
+            bool in = chunk < WCHUNKS;
+            if (!in) chunk = WCHUNKS - 1;
+            int iwl = chunk % WWW;
+            int q = chunk / WWW;
+            int c8 = q % 8;
+            int ihl = q / 8;
+            int ih = oh0 - 1 + ihl;
+            int iw = ow0 - 1 + iwl;
+            if (in && (u32)ih < (u32)p.H && (u32)iw < (u32)p.W)
+                w_okm |= 1u << j;
+            w_loff[j] = (ih * p.W + iw) * p.Cin + c8 * 8;
+        }
+    };
+    auto stage_win_slice = [&](int j) {
+        // lanes past the last chunk must not WRITE (the clamped source
+        // is harmless but an unguarded glds would land 16B past the
+        // buffer — into the partner window / B ring)
+        int chunk = (j * 4 + wave) * 64 + lane;
+        if (chunk < WCHUNKS) {
+            const bf16* src =
+                (w_okm >> j) & 1 ? w_base + w_loff[j] : Z;
+            glds16(src, W0 + (long)dst_buf * WCHUNKS * 8 + chunk * 8);
+        }
+    };
+    auto win_advance = [&]() {
+        ++w_count;
+        dst_buf ^= 1;
+        if (++s_cb == ncb) {
+            s_cb = 0;
+            s_mt += gridDim.x;
+        }
+        if (w_count < nwin && s_mt < mtiles) win_setup();
+    };
+
+    f32x4 acc[2][NI];
+    auto compute = [&](int wbuf, int rs, int bbuf) {
+        const int r = rs / 3, s = rs % 3;
+        bf16* WB = W0 + (long)wbuf * WCHUNKS * 8;
+        bf16* B = B0 + bbuf * BN * BK;
+        const int poff = r * (8 * WWW) + s;      // (r,s) chunk offset
+#pragma unroll
+        for (int ks = 0; ks < BK / 32; ++ks) {
+            bf16x8 af[2], bfr[NI];
+#pragma unroll
+            for (int mi = 0; mi < 2; ++mi) {
+                int pp = wm * 32 + mi * 16 + lo16;       // tile pixel
+                int base = (pp >> 4) * (8 * WWW)          // r' rows
+                           + (ks * 4 + hi4) * WWW         // c8
+                           + (pp & 15);                   // cw
+                af[mi] = *reinterpret_cast<bf16x8*>(
+                    WB + (base + poff) * 8);
+            }
+#pragma unroll
+            for (int ni = 0; ni < NI; ++ni) {
+                int row = ni * 16 + lo16;
+                bfr[ni] = *reinterpret_cast<bf16x8*>(
+                    B + row * BK + swz(row, ks * 4 + hi4) * 8);
+            }
+#pragma unroll
+            for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+                for (int ni = 0; ni < NI; ++ni)
+                    acc[mi][ni] =
+                        MFMA_BF16_16x16x32(af[mi], bfr[ni], acc[mi][ni]);
+        }
+    };
+
+    // ---- epilogue: bounce through the finished window buffer
+    constexpr int RH = (BN == 64) ? 64 : 32;    // rows per round
+    constexpr int ROUNDS = BM / RH;
+    auto epilogue = [&](int nb, int oh0, int ow0, int wbuf) {
+        float* scratch = (float*)(W0 + (long)wbuf * WCHUNKS * 8);
+        // scale/bias loaded here, once per m-tile, not held across the
+        // k-loop (register budget)
+        float sc[NI], bi[NI];
+#pragma unroll
+        for (int ni = 0; ni < NI; ++ni) {
+            int n = n0 + ni * 16 + lo16;
+            if (n >= p.Cout) n = p.Cout - 1;
+            sc[ni] = p.scale[n];
+            bi[ni] = p.bias[n];
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        const bool interior =
+            (oh0 + WTH <= p.OH) && (ow0 + WTW <= p.OW) &&
+            (n0 + BN <= p.Cout) && (p.Cout % 8 == 0);
+#pragma unroll
+        for (int h = 0; h < ROUNDS; ++h) {
+            const int r0 = h * RH;
+#pragma unroll
+            for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+                for (int ni = 0; ni < NI; ++ni)
+#pragma unroll
+                    for (int e = 0; e < 4; ++e) {
+                        int r = wm * 32 + mi * 16 + hi4 * 4 + e;
+                        if (r < r0 || r >= r0 + RH) continue;
+                        int c = ni * 16 + lo16;
+                        int rl = r - r0;
+                        int cs = c ^ (((rl >> 2) & 3) << 4);
+                        scratch[rl * BN + cs] =
+                            acc[mi][ni][e] * sc[ni] + bi[ni];
+                    }
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+            if (interior) {
+                constexpr int CP8 = RH * BN / 8 / NTHREADS;
+                f32x4 v4[CP8][2];
+                bf16x8 rv[CP8];
+                long off[CP8];
+#pragma unroll
+                for (int i = 0; i < CP8; ++i) {
+                    int chunk = tid + i * NTHREADS;
+                    int rl = chunk / (BN / 8);
+                    int c8 = (chunk % (BN / 8)) * 8;
+                    int cs = c8 ^ (((rl >> 2) & 3) << 4);
+                    int pp = r0 + rl;
+                    long oh = oh0 + (pp >> 4), ow = ow0 + (pp & 15);
+                    off[i] = (((long)nb * p.OH + oh) * p.OW + ow)
+                                 * p.Cout + n0 + c8;
+                    v4[i][0] = *reinterpret_cast<f32x4*>(
+                        scratch + rl * BN + cs);
+                    v4[i][1] = *reinterpret_cast<f32x4*>(
+                        scratch + rl * BN + cs + 4);
+                    if (HAS_RES)
+                        rv[i] = *reinterpret_cast<const bf16x8*>(
+                            RES + off[i]);
+                }
+                __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+                for (int i = 0; i < CP8; ++i) {
+                    bf16x8 o;
+#pragma unroll
+                    for (int j = 0; j < 8; ++j) {
+                        float v = v4[i][j / 4][j % 4];
+                        if (HAS_RES) v += bf2f(rv[i][j]);
+                        o[j] = f2bf(apply_act(v, ACT));
+                    }
+                    *reinterpret_cast<bf16x8*>(OUT + off[i]) = o;
+                }
+            } else {
+                constexpr int CPR = RH * BN / 4 / NTHREADS;
+#pragma unroll
+                for (int i = 0; i < CPR; ++i) {
+                    int chunk = tid + i * NTHREADS;
+                    int rl = chunk / (BN / 4);
+                    int c4 = (chunk % (BN / 4)) * 4;
+                    int cs = c4 ^ (((rl >> 2) & 3) << 4);
+                    int pp = r0 + rl;
+                    int oh = oh0 + (pp >> 4), ow = ow0 + (pp & 15);
+                    int n = n0 + c4;
+                    if (oh >= p.OH || ow >= p.OW || n >= p.Cout)
+                        continue;
+                    f32x4 v4 = *reinterpret_cast<f32x4*>(
+                        scratch + rl * BN + cs);
+                    long off = (((long)nb * p.OH + oh) * p.OW + ow)
+                                   * p.Cout + n;
+#pragma unroll
+                    for (int j = 0; j < 4; ++j) {
+                        if (n + j >= p.Cout) continue;
+                        float v = v4[j];
+                        if (HAS_RES) v += bf2f(RES[off + j]);
+                        OUT[off + j] = f2bf(apply_act(v, ACT));
+                    }
+                }
+            }
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+        }
+    };
+
+    // ---- main loop. Window w = (mt, cb) of THIS block's walk; 9 phases
+    // each. Per phase j: wait -> barrier -> stage B(next tile) -> stage
+    // one window slice (j<6, for window w+1) -> compute. vmcnt float:
+    // the slice issued after B(cur) in the previous phase (1 glds/wave).
+    const long local_wins =
+        (blockIdx.x < mtiles)
+            ? (long)((mtiles - 1 - blockIdx.x) / gridDim.x + 1) * ncb
+            : 0;
+    if (local_wins > 0) {
+        win_setup();                  // window 0
+#pragma unroll
+        for (int j = 0; j < 6; ++j) stage_win_slice(j);
+        stage_b(0, 0, 0);             // B tile 0 = (cb 0, rs 0)
+        win_advance();                // cursor -> window 1
+    }
+    long b_staged = 1;                // B tiles issued
+    int b_cb = 0, b_rs = 1;           // next B tile's (cb, rs)
+    long ph = 0;                      // linear phase = B-tile index
+    long w = 0;                       // computed-window counter
+    bool sliced_prev = false;
+
+    for (int mt = blockIdx.x; mt < mtiles; mt += gridDim.x) {
+        int tw = mt % ntw;
+        int t = mt / ntw;
+        c_oh0 = (t % nth) * WTH;
+        c_nb = t / nth;
+        c_ow0 = tw * WTW;
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+            for (int ni = 0; ni < NI; ++ni)
+                acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+        for (int cb = 0; cb < ncb; ++cb, ++w) {
+            const int cbuf = (int)(w & 1);
+            for (int j = 0; j < 9; ++j, ++ph) {
+                if (sliced_prev && !p.smul)   // p.smul: full-sync debug
+                    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(1)
+                                 : "memory");
+                else
+                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                __builtin_amdgcn_s_barrier();
+                if (b_staged < local_wins * 9) {
+                    stage_b((int)((ph + 1) & 1), b_cb, b_rs);
+                    ++b_staged;
+                    if (++b_rs == 9) {
+                        b_rs = 0;
+                        if (++b_cb == ncb) b_cb = 0;
+                    }
+                }
+                // per-WAVE: a wave whose slice slot is entirely past
+                // WCHUNKS issues no glds and must not float one
+                bool sl = (j < 6) && (w_count < local_wins) &&
+                          (s_mt < mtiles) &&
+                          ((j * 4 + wave) * 64 < WCHUNKS);
+                if (sl) stage_win_slice(j);
+                sliced_prev = sl;
+                compute(cbuf, j, (int)(ph & 1));
+            }
+            if (w + 1 < local_wins) win_advance();
+        }
+        epilogue(c_nb, c_oh0, c_ow0, (int)((w - 1) & 1));
+    }
+}
 __global__ void pad_channels_kernel(const bf16* __restrict__ x,
                                     bf16* __restrict__ y, long rows, int C,
                                     int C8) {
@@ -663,6 +997,47 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
     }();
     const bool legacy = variant == 'l';
     const bool force_small = variant == 's';
+    const bool force_win = variant == 'w' || variant == 'W';
+    // variant 'W': force win AND full vmcnt(0) drains (debug)
+
+    // window-reuse path: exact-fit 3x3/s1/p1 tiles (8x16 output pixels);
+    // stages each input window once per 64-channel block instead of
+    // once per (r,s) k-tile (9x less A traffic into LDS)
+    const bool win =
+        (!legacy && !force_small) && !gemm_mode && !stem_mode &&
+        p.R == 3 && p.S == 3 && p.stride == 1 && p.pad == 1 &&
+        (p.Cin % 64 == 0) && (p.Cout % 64 == 0) &&
+        (p.OH % WTH == 0) && (p.OW % WTW == 0) &&
+        (force_win ||
+         (long)p.NB * (p.OH / WTH) * (p.OW / WTW) *
+                 ((p.Cout + 127) / 128) >= 512);
+    if (win) {
+        // BN64 only: the BN128 instantiation needs >256 VGPRs at 2
+        // blocks/CU and spills
+        const int BNw = 64;
+        const int mt2 = (int)((long)p.NB * (p.OH / WTH) * (p.OW / WTW));
+        const int nyw = p.Cout / BNw;
+        int gxw = mt2;
+        if ((long)mt2 * nyw > 768) {
+            gxw = 768 / nyw > 0 ? 768 / nyw : 1;
+            if (gxw > mt2) gxw = mt2;
+        }
+        p.smul = (variant == 'W') ? 1u : 0u;   // full-sync debug flag
+        dim3 gw(gxw, nyw), bw(NTHREADS);
+#define WIN_TILE(A, RZ, BNv)                                              \
+    hipLaunchKernelGGL((conv_win_kernel<A, RZ, BNv>), gw, bw, 0, s, p)
+#define WIN_BN(A, RZ) WIN_TILE(A, RZ, 64)
+        if (relu) {
+            if (has_res) WIN_BN(ACT_RELU, true);
+            else WIN_BN(ACT_RELU, false);
+        } else {
+            if (has_res) WIN_BN(ACT_NONE, true);
+            else WIN_BN(ACT_NONE, false);
+        }
+#undef WIN_BN
+#undef WIN_TILE
+        return;
+    }
     const int mt128 = (p.M + 127) / 128;
     int BMsel, BNsel;
     const bool deepK = p.K >= 1024;
