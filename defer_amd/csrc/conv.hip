@@ -1003,19 +1003,25 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
     // window-reuse path: exact-fit 3x3/s1/p1 tiles (8x16 output pixels);
     // stages each input window once per 64-channel block instead of
     // once per (r,s) k-tile (9x less A traffic into LDS)
+    const long wth = (p.OH + WTH - 1) / WTH, wtw = (p.OW + WTW - 1) / WTW;
+    // partial tiles are handled (zero-filled windows, guarded stores)
+    // but waste compute; require <= ~18% padding overhead
     const bool win =
         (!legacy && !force_small) && !gemm_mode && !stem_mode &&
         p.R == 3 && p.S == 3 && p.stride == 1 && p.pad == 1 &&
-        (p.Cin % 64 == 0) && (p.Cout % 64 == 0) &&
-        (p.OH % WTH == 0) && (p.OW % WTW == 0) &&
+        (p.Cin % 64 == 0) && (p.Cout % 64 == 0) && p.Cout <= 128 &&
+        // Cout>128 shapes (e.g. 56x56x256 deep-K) lose: the BN64-only
+        // window kernel pays 2-4x the B re-staging of the BN128 RSC
+        // config, which outweighs the 9x A saving there (measured)
+        (wth * WTH * wtw * WTW * 100 <=
+         (long)p.OH * p.OW * 118) &&
         (force_win ||
-         (long)p.NB * (p.OH / WTH) * (p.OW / WTW) *
-                 ((p.Cout + 127) / 128) >= 512);
+         (long)p.NB * wth * wtw * ((p.Cout + 127) / 128) >= 512);
     if (win) {
         // BN64 only: the BN128 instantiation needs >256 VGPRs at 2
         // blocks/CU and spills
         const int BNw = 64;
-        const int mt2 = (int)((long)p.NB * (p.OH / WTH) * (p.OW / WTW));
+        const int mt2 = (int)((long)p.NB * wth * wtw);
         const int nyw = p.Cout / BNw;
         int gxw = mt2;
         if ((long)mt2 * nyw > 768) {

@@ -21,6 +21,9 @@ SHAPES = [  # (B, H, W, Cin, Cout, res)
     (3, 24, 32, 128, 192, False),
     (2, 16, 48, 256, 64, True),
     (1, 56, 112, 64, 64, False),     # W%16=0, H%8=0
+    (2, 56, 56, 64, 64, True),       # 56%16=8: column-masked edge tiles
+    (2, 28, 28, 128, 128, False),    # 28%8=4 and 28%16=12: both masked
+    (2, 14, 14, 128, 64, False),     # tiny, heavy masking
 ]
 
 fails = 0
