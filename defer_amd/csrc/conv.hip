@@ -705,7 +705,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
         const int poff = r * (8 * WWW) + s;      // (r,s) chunk offset
 #pragma unroll
         for (int ks = 0; ks < BK / 32; ++ks) {
-            bf16x8 af[2], bfr[NI];
+            bf16x8 af[2];
 #pragma unroll
             for (int mi = 0; mi < 2; ++mi) {
                 int pp = wm * 32 + mi * 16 + lo16;       // tile pixel
@@ -715,18 +715,25 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
                 af[mi] = *reinterpret_cast<bf16x8*>(
                     WB + (base + poff) * 8);
             }
+            // B frags in halves of <=4: caps live B registers at 16
+            // (the BN128 instantiation spilled with all 8 live)
 #pragma unroll
-            for (int ni = 0; ni < NI; ++ni) {
-                int row = ni * 16 + lo16;
-                bfr[ni] = *reinterpret_cast<bf16x8*>(
-                    B + row * BK + swz(row, ks * 4 + hi4) * 8);
+            for (int hf = 0; hf < (NI + 3) / 4; ++hf) {
+                constexpr int HNI = NI < 4 ? NI : 4;
+                bf16x8 bfr[HNI];
+#pragma unroll
+                for (int q = 0; q < HNI; ++q) {
+                    int row = (hf * 4 + q) * 16 + lo16;
+                    bfr[q] = *reinterpret_cast<bf16x8*>(
+                        B + row * BK + swz(row, ks * 4 + hi4) * 8);
+                }
+#pragma unroll
+                for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+                    for (int q = 0; q < HNI; ++q)
+                        acc[mi][hf * 4 + q] = MFMA_BF16_16x16x32(
+                            af[mi], bfr[q], acc[mi][hf * 4 + q]);
             }
-#pragma unroll
-            for (int mi = 0; mi < 2; ++mi)
-#pragma unroll
-                for (int ni = 0; ni < NI; ++ni)
-                    acc[mi][ni] =
-                        MFMA_BF16_16x16x32(af[mi], bfr[ni], acc[mi][ni]);
         }
     };
 
@@ -1010,16 +1017,15 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
         (!legacy && !force_small) && !gemm_mode && !stem_mode &&
         p.R == 3 && p.S == 3 && p.stride == 1 && p.pad == 1 &&
         (p.Cin % 64 == 0) && (p.Cout % 64 == 0) && p.Cout <= 128 &&
-        // Cout>128 shapes (e.g. 56x56x256 deep-K) lose: the BN64-only
-        // window kernel pays 2-4x the B re-staging of the BN128 RSC
-        // config, which outweighs the 9x A saving there (measured)
+        // Cout>128 shapes (56x56x256-class) stay on the BN128 RSC
+        // config: the BN64 window kernel pays 2-4x the B re-staging
+        // (measured loss) and a BN128 window instantiation spills
+        // ~250 B/lane at the 2-blocks/CU register budget
         (wth * WTH * wtw * WTW * 100 <=
          (long)p.OH * p.OW * 118) &&
         (force_win ||
          (long)p.NB * wth * wtw * ((p.Cout + 127) / 128) >= 512);
     if (win) {
-        // BN64 only: the BN128 instantiation needs >256 VGPRs at 2
-        // blocks/CU and spills
         const int BNw = 64;
         const int mt2 = (int)((long)p.NB * wth * wtw);
         const int nyw = p.Cout / BNw;
