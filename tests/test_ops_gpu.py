@@ -242,3 +242,20 @@ def test_stage_stats_event_timer_gpu():
     assert pipe.stats.items == 3
     ms = pipe.stats.compute_ms
     assert 0.05 < ms < 5000, ms   # real device time, not zero
+
+
+@pytest.mark.gpu
+def test_window_conv_forced_numerics():
+    """The window-reuse 3x3 path vs the fp32 reference, FORCED past its
+    size gate (DEFER_CONV_VARIANT=w is latched at first kernel launch,
+    so this runs in a subprocess)."""
+    import os
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, DEFER_CONV_VARIANT="w")
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, "tools", "wincheck.py")],
+        capture_output=True, text=True, timeout=600, env=env, cwd=root)
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-1000:]
