@@ -69,26 +69,66 @@ def _trace_shapes(graph: LayerGraph, input_shape) -> dict:
 def node_costs(graph: LayerGraph, input_shape=(1, 224, 224, 3),
                bytes_per_elem: float = 2.0) -> Tuple[dict, dict]:
     """Per-node (flops, output_bytes) for one input of `input_shape`."""
+    fl, ob, _ = node_times(graph, input_shape, bytes_per_elem)
+    return fl, ob
+
+
+# Measured kernel-class rates on MI355X (tools/convbench.py, tile policy
+# v3 + window kernel; see profiles/). A node's time estimate is
+# max(flops / EFF_class, streamed bytes / BW_class) — the 1x1 and
+# elementwise classes are streaming-bound, the 3x3s MFMA/L2-bound.
+_EFF_3X3 = 450e12       # window / RSC implicit-GEMM 3x3s
+_EFF_1X1 = 300e12       # GEMM-mode 1x1s
+_EFF_STEM = 150e12      # Cin<64 stem paths
+_BW_STREAM = 4.5e12     # achieved streaming bound of the conv epilogues
+_BW_STEM = 1.3e12       # stem/channel-pad paths (measured vs out bytes)
+_BW_ELTW = 5.0e12       # pooling / elementwise kernels
+
+
+def node_times(graph: LayerGraph, input_shape=(1, 224, 224, 3),
+               bytes_per_elem: float = 2.0) -> Tuple[dict, dict, dict]:
+    """Per-node (flops, output_bytes, est_time_us) for one input.
+
+    The time model folds in each kernel class's measured efficiency so
+    the cut chooser balances real stage times, not raw FLOPs (a 1x1 conv
+    runs ~2-3x slower per FLOP than a 3x3 here — both models' hot
+    kernels are documented in profiles/README.md).
+    """
     from defer_amd.models import layers as L
 
     shapes = _trace_shapes(graph, input_shape)
-    flops, out_bytes = {}, {}
+    flops, out_bytes, time_us = {}, {}, {}
     env_shapes = {LayerGraph.INPUT: tuple(input_shape)}
+
+    def nbytes(shape):
+        return float(torch.tensor(shape).prod()) * bytes_per_elem
+
     for n in graph.nodes:
         out = shapes[n.name]
-        env_shapes[n.name] = out
         f = 0.0
         lay = n.layer
+        in_b = sum(nbytes(env_shapes[p]) for p in n.inputs
+                   if p in env_shapes)
+        out_b = nbytes(out)
+        env_shapes[n.name] = out
         if isinstance(lay, L.ConvBNAct):
-            # out: [N, OH, OW, K]
             f = float(out[0] * out[1] * out[2]) * lay.flops_per_pixel()
+            if lay.cin < 64:
+                t = max(f / _EFF_STEM, out_b / _BW_STEM)
+            elif lay.kernel >= 3:
+                t = max(f / _EFF_3X3, (in_b + out_b) / _BW_STREAM)
+            else:
+                t = max(f / _EFF_1X1, (in_b + out_b) / _BW_STREAM)
         elif isinstance(lay, L.Dense):
             f = 2.0 * out[0] * lay.cin * lay.cout
+            t = max(f / _EFF_1X1, (in_b + out_b) / _BW_STREAM)
         else:
             f = 2.0 * float(torch.tensor(out).prod())
+            t = (in_b + out_b) / _BW_ELTW
         flops[n.name] = f
-        out_bytes[n.name] = float(torch.tensor(out).prod()) * bytes_per_elem
-    return flops, out_bytes
+        out_bytes[n.name] = out_b
+        time_us[n.name] = t * 1e6
+    return flops, out_bytes, time_us
 
 
 def auto_partition(model, num_stages: int, input_shape=(1, 224, 224, 3),
@@ -107,7 +147,8 @@ def auto_partition(model, num_stages: int, input_shape=(1, 224, 224, 3),
     graph = gm.graph
     if num_stages <= 1:
         return [], [gm]
-    flops, out_bytes = node_costs(graph, input_shape, bytes_per_elem)
+    flops, out_bytes, time_us = node_times(graph, input_shape,
+                                           bytes_per_elem)
     cuts_avail = graph.valid_cut_points()
     if len(cuts_avail) < num_stages - 1:
         raise ValueError(
@@ -116,8 +157,10 @@ def auto_partition(model, num_stages: int, input_shape=(1, 224, 224, 3),
     pos = {n.name: i for i, n in enumerate(graph.nodes)}
     cut_pos = sorted(pos[c] for c in cuts_avail)
     names = [n.name for n in graph.nodes]
-    # prefix sums of compute time (us per image) per node
-    t_node = [flops[n] / (eff_tflops * 1e6) for n in names]
+    # prefix sums of compute time (us per image) per node — the
+    # kernel-class-aware estimates; eff_tflops rescales them only if a
+    # caller overrides the default (relative balance is what matters)
+    t_node = [time_us[n] * (EFF_TFLOPS / eff_tflops) for n in names]
     prefix = [0.0]
     for t in t_node:
         prefix.append(prefix[-1] + t)
