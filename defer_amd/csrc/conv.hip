@@ -67,13 +67,11 @@ __device__ __forceinline__ void glds16(const bf16* src, bf16* lds_base) {
         16, 0, 0);
 }
 
-// DA/DB: independent pipeline depths for the A (activation) and B
-// (weight) LDS rings. A streams from HBM (~900 cy latency) and needs
-// DA-1 compute phases of cover; B re-reads L2-resident weights (~250 cy)
-// and DB=2 suffices — the LDS saved goes into a deeper A ring while
-// keeping 2 blocks/CU (160 KB LDS). With DA>DB the per-iteration stage
-// order is B-then-A so the counted s_waitcnt (FIFO vmcnt) can float the
-// newest A stage past the B(it) wait.
+// DA: pipeline depth of the A/B LDS rings (DB must equal DA unless
+// B_PERSIST, where the single B tile is staged once and DB is unused —
+// an asymmetric deep-A/shallow-B variant was implemented and measured
+// 6-22% slower: boundary activations are L3/L2-resident, so extra LDS
+// per block costs more occupancy than deeper latency cover buys).
 // WPS: minimum waves per SIMD (__launch_bounds__ 2nd arg) = blocks/CU of
 // this 256-thread kernel. The MFMA-bound tile configs run 2 blocks/CU;
 // the pure-bandwidth small-tile GEMM configs (1x1 convs, K<=512) trade
@@ -98,10 +96,13 @@ __global__ __launch_bounds__(TPB, WPS) void conv_igemm_kernel(
     constexpr int ACH = BM * 8 / TPB;           // A chunks per thread
     constexpr int BCH = BN * 8 / TPB;           // B chunks per thread
     constexpr int OPS = ACH + (B_PERSIST ? 0 : BCH);
-    constexpr bool ASYM = !B_PERSIST && (DA != DB);
+    // DB is only meaningful for B_PERSIST configs (unused ring slot);
+    // both rings run at depth DA — an asymmetric deep-A/shallow-B
+    // variant was measured 6-22% slower (L3-resident activations) and
+    // removed (git history has the measurements).
 
     __shared__ __attribute__((aligned(16)))
-    bf16 lds[(DA * BM + (B_PERSIST ? 1 : DB) * BN) * BK];
+    bf16 lds[(DA * BM + (B_PERSIST ? 1 : DA) * BN) * BK];
     bf16* A0 = lds;
     bf16* B0 = lds + DA * BM * BK;
 
@@ -418,10 +419,10 @@ __global__ __launch_bounds__(TPB, WPS) void conv_igemm_kernel(
         }
     };
 
-    // ---- flattened (m-tile, k-tile) pipeline. The A stage cursor runs
-    // DA-1 tiles ahead of compute; the B cursor DB-1 (ASYM) or DA-1
-    // (symmetric). Counted vmcnt: tile it's stages must have landed by
-    // wait(it); newer stages stay in flight (the latency cover).
+    // ---- flattened (m-tile, k-tile) pipeline. The stage cursor runs
+    // DA-1 tiles ahead of compute. Counted vmcnt: tile it's stages must
+    // have landed by wait(it); newer stages stay in flight (the latency
+    // cover — up to DA-1 compute phases).
     auto advance = [&]() {
         if (++s_kt == nk) {
             s_kt = 0;
@@ -444,62 +445,21 @@ __global__ __launch_bounds__(TPB, WPS) void conv_igemm_kernel(
             for (int i = 0; i < ACH; ++i) a_ptr[i] += d;
         }
     };
-    // (ASYM) independent, shallower B cursor: stages tile it+DB-1 while
-    // the A cursor stages it+DA-1. B is L2-resident weight traffic; one
-    // to two compute phases of cover suffice.
-    long b_left = 0, b_tile = 0;
-    int b_kt = 0, b_cb = 0, b_r = 0, b_s = 0;
-    if (ASYM)
-        b_left = (blockIdx.x < mtiles)
-                     ? (long)((mtiles - 1 - blockIdx.x) / gridDim.x + 1)
-                           * nk
-                     : 0;
-    auto b_advance = [&]() {
-        --b_left; ++b_tile;
-        if (++b_kt == nk) {
-            b_kt = 0; b_cb = 0; b_r = 0; b_s = 0;
-        } else if (AMODE == AMODE_RSC) {
-            if (++b_s == p.S) {
-                b_s = 0;
-                if (++b_r == p.R) { b_r = 0; ++b_cb; }
-            }
-        }
-    };
-
     if (s_mt < mtiles) a_setup(s_mt);
     if (B_PERSIST) stage_b(0, 0, 0, 0, 0);  // oldest ops: first wait drains
     int staged = 0;
-    // `run`: consecutive preceding iterations that issued BOTH stages —
-    // the guard for the ASYM float allowance (tail falls back to full
-    // drains). Prologue interleaves B/A per virtual iteration so the
-    // newest-op window at wait(0) is exactly the steady-state one.
-    int run = 0;
 #pragma unroll
     for (int d = 0; d < DA - 1; ++d) {
-        bool did_b = true;
-        if (ASYM && d < DB - 1) {
-            if (b_left > 0) {
-                stage_b((int)(b_tile % DB), b_kt, b_cb, b_r, b_s);
-                b_advance();
-            } else {
-                did_b = false;
-            }
-        }
         if (s_mt < mtiles) {
             stage_a(d);
-            if (!B_PERSIST && !ASYM)
-                stage_b(d, s_kt, s_cb, s_r, s_s);
+            if (!B_PERSIST) stage_b(d, s_kt, s_cb, s_r, s_s);
             advance();
             ++staged;
-            if (did_b) ++run;
         }
     }
-    if (run < DA - 1) run = 0;          // incomplete prologue: drain fully
 
-    // ASYM float allowance: everything newer than B(it) — the A stages
-    // of the last DB-1 iterations plus their B stages beyond the first.
-    constexpr int NFLOAT =
-        ASYM ? ((DB - 2) * OPS + ACH) : ((DA - 2) * OPS);
+    // float allowance: up to DA-2 whole stages may stay in flight
+    constexpr int NFLOAT = (DA - 2) * OPS;
 
     int it = 0;   // computed-iteration counter; tile i -> buffer i%DA
     for (int mt = blockIdx.x; mt < mtiles; mt += gridDim.x) {
@@ -509,51 +469,28 @@ __global__ __launch_bounds__(TPB, WPS) void conv_igemm_kernel(
             for (int ni = 0; ni < NI; ++ni)
                 acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
         for (int kt = 0; kt < nk; ++kt, ++it) {
-            if (ASYM) {
-                if (run >= DB - 1)
-                    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NFLOAT)
-                                 : "memory");
-                else
-                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-            } else {
-                const int avail = staged - it - 1;
-                if (DA >= 4 && avail >= 2)
-                    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NFLOAT)
-                                 : "memory");
-                else if (avail >= 1)
-                    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(OPS)
-                                 : "memory");
-                else
-                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-            }
+            const int avail = staged - it - 1;
+            if (DA >= 4 && avail >= 2)
+                asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NFLOAT)
+                             : "memory");
+            else if (avail >= 1)
+                asm volatile("s_waitcnt vmcnt(%0)" ::"i"(OPS)
+                             : "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
             __builtin_amdgcn_s_barrier();
             // stage BEFORE compute: the refilled buffers were last read
             // by compute(it-1), which every wave finished before this
             // barrier — issuing the glds here buys one extra compute
             // phase of latency cover
-            if (ASYM) {
-                bool did_b = false, did_a = false;
-                if (b_left > 0) {
-                    stage_b((int)(b_tile % DB), b_kt, b_cb, b_r, b_s);
-                    b_advance();
-                    did_b = true;
-                }
-                if (s_mt < mtiles) {
-                    stage_a((it + DA - 1) % DA);
-                    advance();
-                    ++staged;
-                    did_a = true;
-                }
-                run = (did_b && did_a) ? run + 1 : 0;
-            } else if (s_mt < mtiles) {
+            if (s_mt < mtiles) {
                 stage_a((it + DA - 1) % DA);
                 if (!B_PERSIST)
                     stage_b((it + DA - 1) % DA, s_kt, s_cb, s_r, s_s);
                 advance();
                 ++staged;
             }
-            compute(it % DA,
-                    B_PERSIST ? 0 : (ASYM ? (int)(it % DB) : it % DA));
+            compute(it % DA, B_PERSIST ? 0 : it % DA);
         }
         // epilogue AFTER the next tiles' stages were issued: their glds
         // land in buffers disjoint from the scratch buffer (it-1)%DA
@@ -1107,7 +1044,7 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
     }
     dim3 grid(gx, ny);
     dim3 block(NTHREADS);
-    (void)big;
+    (void)big;   // see the BM256 note above
 
 #define DISPATCH_TILE(A, R, G, BP, BMv, BNv, D...)                        \
     hipLaunchKernelGGL(                                                   \

@@ -73,8 +73,8 @@ def main():
     dev = (torch.device("cuda", local_rank) if args.device == "cuda"
            else torch.device("cpu"))
     B = args.batch
-    pipe = DistPipeline(args.model == "resnet50" and resnet50() or vgg19(),
-                        cfg, (B, 224, 224, 3), device=dev)
+    model = resnet50() if args.model == "resnet50" else vgg19()
+    pipe = DistPipeline(model, cfg, (B, 224, 224, 3), device=dev)
     dtype = torch.bfloat16 if args.device == "cuda" else torch.float32
     x = torch.randn(B, 224, 224, 3, device=dev, dtype=dtype)
 
