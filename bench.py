@@ -27,7 +27,8 @@ def parse_args():
     ap.add_argument("--steps", type=int, default=64)
     ap.add_argument("--warmup", type=int, default=16)
     ap.add_argument("--model", default="resnet50",
-                    choices=["resnet50", "vgg19"])
+                    choices=["resnet50", "resnet101", "resnet152",
+                             "vgg19"])
     ap.add_argument("--batch", type=int, default=64,
                     help="images per pipeline micro-batch")
     ap.add_argument("--compression", default="none",
@@ -51,10 +52,14 @@ def parse_args():
 
 def main():
     args = parse_args()
-    # keep the one-JSON-line stdout contract clean: RCCL prints a version
-    # banner to fd 1 when NCCL_DEBUG is set, so leave it unset here
+    # ONE-JSON-LINE stdout contract: native libs write banners straight
+    # to fd 1 (gloo's rank banner; RCCL's version line under NCCL_DEBUG),
+    # so park the real stdout and point fd 1 at stderr for the whole
+    # run — the JSON line goes to the saved fd at the end.
+    real_stdout = os.dup(1)
+    os.dup2(2, 1)
     from defer_amd.config import PipelineConfig
-    from defer_amd.models import DEFER_8STAGE_CUTS, resnet50, vgg19
+    from defer_amd.models import DEFER_8STAGE_CUTS, MODELS
     from defer_amd.parallel.pipeline import DistPipeline
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -76,7 +81,7 @@ def main():
             world_size=1)
 
     torch.manual_seed(0)
-    model = resnet50() if args.model == "resnet50" else vgg19()
+    model = MODELS[args.model]()
 
     if args.cuts == "auto":
         cuts = None
@@ -160,9 +165,10 @@ def main():
               file=sys.stderr)
     if pipe.rank == 0:
         out = {
-            "metric": "images/sec (whole node) ResNet50 pipeline"
-                      if args.model == "resnet50"
-                      else "images/sec (whole node) VGG19 pipeline",
+            "metric": "images/sec (whole node) "
+                      + {"resnet50": "ResNet50", "resnet101": "ResNet101",
+                         "resnet152": "ResNet152",
+                         "vgg19": "VGG19"}[args.model] + " pipeline",
             "value": round(ips, 1),
             "unit": "images/sec",
             "n_gpus": world,
@@ -187,7 +193,8 @@ def main():
                 "weights": "random-init",
             },
         }
-        print(json.dumps(out))
+        os.write(real_stdout, (json.dumps(out) + "\n").encode())
+    os.close(real_stdout)
     dist.destroy_process_group()
 
 

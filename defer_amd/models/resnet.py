@@ -1,9 +1,11 @@
-"""ResNet50 built as a defer_amd LayerGraph.
+"""ResNet-50/101/152 built as defer_amd LayerGraphs.
 
 The reference's benchmark model is Keras ResNet50, include_top=True,
 224x224x3 (test/test.py:14). The residual-add layers are named add_1 ..
 add_16 so the reference's partition lists (add_2, add_4, ... add_14 —
-test/test.py:18) name the same cut points here.
+test/test.py:18) name the same cut points here. The 101/152 variants
+(deeper conv4_x/conv3_x stages, same bottleneck blocks) exercise the
+partitioner and pipeline at greater depth.
 """
 
 from typing import List
@@ -13,7 +15,7 @@ from defer_amd.models.layers import (AddAct, ConvBNAct, Dense,
                                      GlobalAvgPool, MaxPool, Softmax)
 
 
-def resnet50(num_classes: int = 1000, include_top: bool = True) -> GraphModel:
+def _resnet(depths, name, num_classes=1000, include_top=True) -> GraphModel:
     nodes: List[GraphNode] = []
     add_idx = [0]
 
@@ -43,8 +45,10 @@ def resnet50(num_classes: int = 1000, include_top: bool = True) -> GraphModel:
         return N(f"add_{add_idx[0]}", AddAct("relu"), [c, sc]), cout
 
     cin = 64
+    widths = (64, 128, 256, 512)
     for stage, (blocks, width, stride) in enumerate(
-            [(3, 64, 1), (4, 128, 2), (6, 256, 2), (3, 512, 2)], start=2):
+            [(depths[i], widths[i], 1 if i == 0 else 2)
+             for i in range(4)], start=2):
         for b in range(blocks):
             x, cin = bottleneck(x, stage, b, cin, width,
                                 stride if b == 0 else 1)
@@ -55,7 +59,19 @@ def resnet50(num_classes: int = 1000, include_top: bool = True) -> GraphModel:
         x = N("fc1000", Dense(cin, num_classes), [x])
         x = N("softmax", Softmax(), [x])
 
-    return GraphModel(LayerGraph(nodes, output=x), name="resnet50")
+    return GraphModel(LayerGraph(nodes, output=x), name=name)
+
+
+def resnet50(num_classes: int = 1000, include_top: bool = True):
+    return _resnet((3, 4, 6, 3), "resnet50", num_classes, include_top)
+
+
+def resnet101(num_classes: int = 1000, include_top: bool = True):
+    return _resnet((3, 4, 23, 3), "resnet101", num_classes, include_top)
+
+
+def resnet152(num_classes: int = 1000, include_top: bool = True):
+    return _resnet((3, 8, 36, 3), "resnet152", num_classes, include_top)
 
 
 # The reference's 8-stage partition (7 cuts at every other residual add,

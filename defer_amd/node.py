@@ -28,7 +28,8 @@ import torch.distributed as dist
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--model", default="resnet50",
-                    choices=["resnet50", "vgg19"])
+                    choices=["resnet50", "resnet101", "resnet152",
+                             "vgg19"])
     ap.add_argument("--batch", type=int, default=64)
     ap.add_argument("--cuts", default="auto")
     ap.add_argument("--items", type=int, default=0,
@@ -41,7 +42,7 @@ def main():
     args = ap.parse_args()
 
     from defer_amd.config import PipelineConfig
-    from defer_amd.models import DEFER_8STAGE_CUTS, resnet50, vgg19
+    from defer_amd.models import DEFER_8STAGE_CUTS, MODELS
     from defer_amd.parallel.pipeline import DistPipeline
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -73,7 +74,7 @@ def main():
     dev = (torch.device("cuda", local_rank) if args.device == "cuda"
            else torch.device("cpu"))
     B = args.batch
-    model = resnet50() if args.model == "resnet50" else vgg19()
+    model = MODELS[args.model]()
     pipe = DistPipeline(model, cfg, (B, 224, 224, 3), device=dev)
     dtype = torch.bfloat16 if args.device == "cuda" else torch.float32
     x = torch.randn(B, 224, 224, 3, device=dev, dtype=dtype)

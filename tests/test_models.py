@@ -104,3 +104,25 @@ def test_stage_executor_applies_fusion_cpu():
     with torch.no_grad():
         got = ex.run(x)
     assert torch.allclose(got, want, atol=1e-5)
+
+
+def test_resnet_deep_variants():
+    """ResNet-101/152: same bottleneck blocks at greater depth; the
+    partitioner handles them with auto cuts."""
+    from defer_amd.models import resnet101, resnet152
+
+    for mk, nadds in ((resnet101, 33), (resnet152, 50)):
+        m = mk()
+        names = m.graph.layer_names()
+        assert f"add_{nadds}" in names and f"add_{nadds+1}" not in names
+        x = torch.randn(1, 64, 64, 3)
+        with torch.no_grad():
+            y = m(x)
+        assert y.shape == (1, 1000)
+        cuts, stages = auto_partition(m, 8, input_shape=(1, 64, 64, 3))
+        assert len(stages) == 8
+        with torch.no_grad():
+            z = x
+            for s in stages:
+                z = s(z)
+        assert torch.allclose(z, y)
