@@ -515,19 +515,21 @@ __global__ __launch_bounds__(TPB, WPS) void conv_igemm_kernel(
 // staged in 6 glds slices interleaved with phases 0..5, so the counted
 // vmcnt keeps 3+ phases of load-latency cover without a FIFO conflict
 // with the per-phase B stages (order: B first, then the slice).
-#define WTH 8
-#define WTW 16
-#define WWH (WTH + 2)                  // window rows
-#define WWW (WTW + 2)                  // window cols
-#define WCHUNKS (WWH * 8 * WWW)        // 16-B chunks per window (1440)
-#define WSLOTS 24                      // glds slots (4 waves x 6)
+#define WTW 16                         // window tile width (fixed)
 
-template <int ACT, bool HAS_RES, int BN>
+template <int ACT, bool HAS_RES, int BN, int TH = 8>
 __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
     ConvParams p) {
-    constexpr int BM = WTH * WTW;               // 128
+    constexpr int TW = WTW;
+    constexpr int BM = TH * TW;                 // 128 (TH8) / 64 (TH4)
+    constexpr int WH = TH + 2, WW = TW + 2;     // window dims
+    constexpr int WCH = WH * 8 * WW;            // 16-B chunks per window
+    constexpr int NSL = (WCH + 255) / 256;      // glds slices per wave
     constexpr int DB = 2;
-    constexpr int NI = BN / 16;                 // all 4 waves along m
+    constexpr int WMW = BM / 32;                // waves along m
+    constexpr int WNW = 4 / WMW;                // waves along n
+    constexpr int WN = BN / WNW;
+    constexpr int NI = WN / 16;
     constexpr int BCH = BN / 32;
     const bf16* __restrict__ X = (const bf16*)p.x;
     const bf16* __restrict__ Wt = (const bf16*)p.w;
@@ -536,20 +538,21 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
     bf16* __restrict__ OUT = (bf16*)p.out;
 
     __shared__ __attribute__((aligned(16)))
-    bf16 lds[(2 * WCHUNKS * 8) + DB * BN * BK];
+    bf16 lds[(2 * WCH * 8) + DB * BN * BK];
     bf16* W0 = lds;                              // two window buffers
-    bf16* B0 = lds + 2 * WCHUNKS * 8;
+    bf16* B0 = lds + 2 * WCH * 8;
 
     const int tid = threadIdx.x;
     const int wave = tid / WAVE;
     const int lane = tid % WAVE;
-    const int wm = wave;                         // WNW = 1
+    const int wm = wave / WNW;
+    const int wn = wave % WNW;
     const int n0 = blockIdx.y * BN;
     const int lo16 = lane & 15;
     const int hi4 = lane >> 4;
     const int ncb = p.Cin / 64;
-    const int ntw = (p.OW + WTW - 1) / WTW;
-    const int nth = (p.OH + WTH - 1) / WTH;
+    const int ntw = (p.OW + TW - 1) / TW;
+    const int nth = (p.OH + TH - 1) / TH;
     const int mtiles = p.NB * nth * ntw;
     const long nwin = (long)mtiles * ncb;
 
@@ -581,7 +584,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
     // is rebuilt in stage_win_slice (register pressure: the first cut
     // of this kernel kept 6 pointers per lane and spilled).
     int s_mt = blockIdx.x, s_cb = 0;
-    int w_loff[6];                    // per-lane element offsets
+    int w_loff[NSL];                  // per-lane element offsets
     u32 w_okm = 0;                    // bit j = slice j in bounds
     const bf16* w_base = X;           // + nb*H*W*Cin + cb*64 (uniform)
     int dst_buf = 0;                  // LDS buffer this window lands in
@@ -593,17 +596,17 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
         int t = mt / ntw;
         int th = t % nth;
         int nb = t / nth;
-        int oh0 = th * WTH, ow0 = tw * WTW;
+        int oh0 = th * TH, ow0 = tw * TW;
         w_base = X + (long)nb * p.H * p.W * p.Cin + s_cb * 64;
         w_okm = 0;
 #pragma unroll
-        for (int j = 0; j < 6; ++j) {
+        for (int j = 0; j < NSL; ++j) {
             // slot j covers chunk (j*4 + wave)*64 + lane
             int chunk = (j * 4 + wave) * 64 + lane;
-            bool in = chunk < WCHUNKS;
-            if (!in) chunk = WCHUNKS - 1;
-            int iwl = chunk % WWW;
-            int q = chunk / WWW;
+            bool in = chunk < WCH;
+            if (!in) chunk = WCH - 1;
+            int iwl = chunk % WW;
+            int q = chunk / WW;
             int c8 = q % 8;
             int ihl = q / 8;
             int ih = oh0 - 1 + ihl;
@@ -618,10 +621,10 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
         // is harmless but an unguarded glds would land 16B past the
         // buffer — into the partner window / B ring)
         int chunk = (j * 4 + wave) * 64 + lane;
-        if (chunk < WCHUNKS) {
+        if (chunk < WCH) {
             const bf16* src =
                 (w_okm >> j) & 1 ? w_base + w_loff[j] : Z;
-            glds16(src, W0 + (long)dst_buf * WCHUNKS * 8 + chunk * 8);
+            glds16(src, W0 + (long)dst_buf * WCH * 8 + chunk * 8);
         }
     };
     auto win_advance = [&]() {
@@ -637,17 +640,17 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
     f32x4 acc[2][NI];
     auto compute = [&](int wbuf, int rs, int bbuf) {
         const int r = rs / 3, s = rs % 3;
-        bf16* WB = W0 + (long)wbuf * WCHUNKS * 8;
+        bf16* WB = W0 + (long)wbuf * WCH * 8;
         bf16* B = B0 + bbuf * BN * BK;
-        const int poff = r * (8 * WWW) + s;      // (r,s) chunk offset
+        const int poff = r * (8 * WW) + s;       // (r,s) chunk offset
 #pragma unroll
         for (int ks = 0; ks < BK / 32; ++ks) {
             bf16x8 af[2];
 #pragma unroll
             for (int mi = 0; mi < 2; ++mi) {
                 int pp = wm * 32 + mi * 16 + lo16;       // tile pixel
-                int base = (pp >> 4) * (8 * WWW)          // r' rows
-                           + (ks * 4 + hi4) * WWW         // c8
+                int base = (pp >> 4) * (8 * WW)           // r' rows
+                           + (ks * 4 + hi4) * WW          // c8
                            + (pp & 15);                   // cw
                 af[mi] = *reinterpret_cast<bf16x8*>(
                     WB + (base + poff) * 8);
@@ -660,7 +663,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
                 bf16x8 bfr[HNI];
 #pragma unroll
                 for (int q = 0; q < HNI; ++q) {
-                    int row = (hf * 4 + q) * 16 + lo16;
+                    int row = wn * WN + (hf * 4 + q) * 16 + lo16;
                     bfr[q] = *reinterpret_cast<bf16x8*>(
                         B + row * BK + swz(row, ks * 4 + hi4) * 8);
                 }
@@ -674,17 +677,20 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
         }
     };
 
-    // ---- epilogue: bounce through the finished window buffer
-    constexpr int RH = (BN == 64) ? 64 : 32;    // rows per round
+    // ---- epilogue: bounce through the finished window buffer.
+    // RH rows per round, sized so RH*BN f32 fits the WCH*4-f32 buffer.
+    constexpr int RH = (WCH * 4 / BN) >= BM
+                           ? BM
+                           : ((WCH * 4 / BN) >= BM / 2 ? BM / 2 : BM / 4);
     constexpr int ROUNDS = BM / RH;
     auto epilogue = [&](int nb, int oh0, int ow0, int wbuf) {
-        float* scratch = (float*)(W0 + (long)wbuf * WCHUNKS * 8);
+        float* scratch = (float*)(W0 + (long)wbuf * WCH * 8);
         // scale/bias loaded here, once per m-tile, not held across the
         // k-loop (register budget)
         float sc[NI], bi[NI];
 #pragma unroll
         for (int ni = 0; ni < NI; ++ni) {
-            int n = n0 + ni * 16 + lo16;
+            int n = n0 + wn * WN + ni * 16 + lo16;
             if (n >= p.Cout) n = p.Cout - 1;
             sc[ni] = p.scale[n];
             bi[ni] = p.bias[n];
@@ -692,7 +698,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();
         const bool interior =
-            (oh0 + WTH <= p.OH) && (ow0 + WTW <= p.OW) &&
+            (oh0 + TH <= p.OH) && (ow0 + TW <= p.OW) &&
             (n0 + BN <= p.Cout) && (p.Cout % 8 == 0);
 #pragma unroll
         for (int h = 0; h < ROUNDS; ++h) {
@@ -705,7 +711,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
                     for (int e = 0; e < 4; ++e) {
                         int r = wm * 32 + mi * 16 + hi4 * 4 + e;
                         if (r < r0 || r >= r0 + RH) continue;
-                        int c = ni * 16 + lo16;
+                        int c = wn * WN + ni * 16 + lo16;
                         int rl = r - r0;
                         int cs = c ^ (((rl >> 2) & 3) << 4);
                         scratch[rl * BN + cs] =
@@ -790,7 +796,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
     if (local_wins > 0) {
         win_setup();                  // window 0
 #pragma unroll
-        for (int j = 0; j < 6; ++j) stage_win_slice(j);
+        for (int j = 0; j < NSL; ++j) stage_win_slice(j);
         stage_b(0, 0, 0);             // B tile 0 = (cb 0, rs 0)
         win_advance();                // cursor -> window 1
     }
@@ -803,9 +809,9 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
     for (int mt = blockIdx.x; mt < mtiles; mt += gridDim.x) {
         int tw = mt % ntw;
         int t = mt / ntw;
-        c_oh0 = (t % nth) * WTH;
+        c_oh0 = (t % nth) * TH;
         c_nb = t / nth;
-        c_ow0 = tw * WTW;
+        c_ow0 = tw * TW;
 #pragma unroll
         for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
@@ -829,10 +835,10 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
                     }
                 }
                 // per-WAVE: a wave whose slice slot is entirely past
-                // WCHUNKS issues no glds and must not float one
-                bool sl = (j < 6) && (w_count < local_wins) &&
+                // WCH issues no glds and must not float one
+                bool sl = (j < NSL) && (w_count < local_wins) &&
                           (s_mt < mtiles) &&
-                          ((j * 4 + wave) * 64 < WCHUNKS);
+                          ((j * 4 + wave) * 64 < WCH);
                 if (sl) stage_win_slice(j);
                 sliced_prev = sl;
                 compute(cbuf, j, (int)(ph & 1));
@@ -944,28 +950,38 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
     const bool force_win = variant == 'w' || variant == 'W';
     // variant 'W': force win AND full vmcnt(0) drains (debug)
 
-    // window-reuse path: exact-fit 3x3/s1/p1 tiles (8x16 output pixels);
-    // stages each input window once per 64-channel block instead of
-    // once per (r,s) k-tile (9x less A traffic into LDS)
-    const long wth = (p.OH + WTH - 1) / WTH, wtw = (p.OW + WTW - 1) / WTW;
-    // partial tiles are handled (zero-filled windows, guarded stores)
-    // but waste compute; require <= ~18% padding overhead
-    const bool win =
+    // window-reuse path: 3x3/s1/p1 with TH x 16 output tiles (TH 8 or
+    // 4); stages each input window once per 64-channel block instead of
+    // once per (r,s) k-tile (9x less A traffic into LDS). Partial tiles
+    // are handled (zero-filled windows, guarded stores) but waste
+    // compute; require <= ~18% padding overhead. Cout>128 shapes
+    // (56x56x256-class) stay on the BN128 RSC config: the BN64 window
+    // kernel pays 2-4x the B re-staging (measured loss) and a BN128
+    // window instantiation spills ~250 B/lane at this register budget.
+    const bool win_ok =
         (!legacy && !force_small) && !gemm_mode && !stem_mode &&
         p.R == 3 && p.S == 3 && p.stride == 1 && p.pad == 1 &&
-        (p.Cin % 64 == 0) && (p.Cout % 64 == 0) && p.Cout <= 128 &&
-        // Cout>128 shapes (56x56x256-class) stay on the BN128 RSC
-        // config: the BN64 window kernel pays 2-4x the B re-staging
-        // (measured loss) and a BN128 window instantiation spills
-        // ~250 B/lane at the 2-blocks/CU register budget
-        (wth * WTH * wtw * WTW * 100 <=
-         (long)p.OH * p.OW * 118) &&
-        (force_win ||
-         (long)p.NB * wth * wtw * ((p.Cout + 127) / 128) >= 512);
-    if (win) {
-        const int BNw = 64;
+        (p.Cin % 64 == 0) && (p.Cout % 64 == 0) && p.Cout <= 128;
+    auto win_fit = [&](int TH) {
+        long th = (p.OH + TH - 1) / TH, tw = (p.OW + WTW - 1) / WTW;
+        if (th * TH * tw * WTW * 100 > (long)p.OH * p.OW * 118)
+            return false;
+        return force_win ||
+               (long)p.NB * th * tw * ((p.Cout + 127) / 128) >= 512;
+    };
+    // TH=4 measured slower than the BN128 RSC config on the 28-px
+    // shapes it would serve (2x B re-staging + 14% tile waste), so the
+    // default policy uses TH=8 only; TH=4 stays exercised by the forced
+    // numerics harness (tools/wincheck.py)
+    const int THsel = !win_ok          ? 0
+                      : win_fit(8)     ? 8
+                      : (force_win && win_fit(4)) ? 4
+                                       : 0;
+    if (THsel) {
+        const long wth = (p.OH + THsel - 1) / THsel;
+        const long wtw = (p.OW + WTW - 1) / WTW;
         const int mt2 = (int)((long)p.NB * wth * wtw);
-        const int nyw = p.Cout / BNw;
+        const int nyw = p.Cout / 64;
         int gxw = mt2;
         if ((long)mt2 * nyw > 768) {
             gxw = 768 / nyw > 0 ? 768 / nyw : 1;
@@ -973,9 +989,13 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
         }
         p.smul = (variant == 'W') ? 1u : 0u;   // full-sync debug flag
         dim3 gw(gxw, nyw), bw(NTHREADS);
-#define WIN_TILE(A, RZ, BNv)                                              \
-    hipLaunchKernelGGL((conv_win_kernel<A, RZ, BNv>), gw, bw, 0, s, p)
-#define WIN_BN(A, RZ) WIN_TILE(A, RZ, 64)
+#define WIN_TILE(A, RZ, TH)                                               \
+    hipLaunchKernelGGL((conv_win_kernel<A, RZ, 64, TH>), gw, bw, 0, s, p)
+#define WIN_BN(A, RZ)                                                     \
+    do {                                                                  \
+        if (THsel == 8) WIN_TILE(A, RZ, 8);                               \
+        else WIN_TILE(A, RZ, 4);                                          \
+    } while (0)
         if (relu) {
             if (has_res) WIN_BN(ACT_RELU, true);
             else WIN_BN(ACT_RELU, false);

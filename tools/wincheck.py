@@ -24,6 +24,8 @@ SHAPES = [  # (B, H, W, Cin, Cout, res)
     (2, 56, 56, 64, 64, True),       # 56%16=8: column-masked edge tiles
     (2, 28, 28, 128, 128, False),    # 28%8=4 and 28%16=12: both masked
     (2, 14, 14, 128, 64, False),     # tiny, heavy masking
+    (2, 12, 16, 64, 64, True),       # TH=4 tile path (OH%8 != 0)
+    (3, 28, 28, 192, 128, False),    # TH=4 with masked cols
 ]
 
 fails = 0
