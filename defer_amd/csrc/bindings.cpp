@@ -53,21 +53,27 @@ const float* zeros_buf() {
 // Prepared-weight cache for the stem paths (K-order repack / channel
 // pad). Weights are constant in inference, so the one-time transform is
 // keyed on (data_ptr, numel, variant) — the reference pays its weight
-// prep once at dispatch too (node.py:34). Unbounded but tiny (one entry
-// per stem conv weight).
+// prep once at dispatch too (node.py:34). The SOURCE tensor is pinned
+// in the entry: the caching allocator would otherwise reuse a freed
+// weight's pointer for a different model's same-shaped weights and the
+// key would collide with stale data (observed as a flaky cross-test
+// mismatch). Unbounded but tiny (one entry per stem conv weight);
+// in-place weight mutation is not detected (inference engine — the
+// reference also ships weights exactly once, dispatcher.py:57).
 Tensor cached_weight_prep(const Tensor& w, int64_t variant,
                           const std::function<Tensor()>& make) {
-    static std::unordered_map<uint64_t, std::pair<const void*, Tensor>>
-        cache;
+    static std::unordered_map<uint64_t, std::pair<Tensor, Tensor>> cache;
     static std::mutex mu;
     uint64_t key = (uint64_t)(uintptr_t)w.data_ptr() * 31 +
                    (uint64_t)w.numel() * 7 + (uint64_t)variant;
     std::lock_guard<std::mutex> g(mu);
     auto it = cache.find(key);
-    if (it != cache.end() && it->second.first == w.data_ptr())
+    if (it != cache.end() &&
+        it->second.first.data_ptr() == w.data_ptr() &&
+        it->second.first.numel() == w.numel())
         return it->second.second;
     Tensor t = make();
-    cache[key] = {w.data_ptr(), t};
+    cache[key] = {w, t};     // pinning w keeps its pointer unique
     return t;
 }
 
@@ -110,6 +116,35 @@ Tensor conv2d_bn_act(Tensor x, Tensor w, c10::optional<Tensor> scale,
     p.Cout = Cout;
 
     if (Cin % 8 != 0) {
+        // small-Cin window path (VGG-style 3x3 stems): channel-pad to
+        // 8, repack weights j-major (cached), stage each 8x16 output
+        // tile's input window to LDS once — kills the ~9x input
+        // re-reads of the channel-pad gather path (b1.c1 143 -> 91 us).
+        // The 7x7/s2 ResNet stem stays on the spatial-prepad run-packed
+        // path: its K is 168 vs the window kernel's zero-padded 448
+        // (Cin 3 -> 8), and the 2.7x extra MFMA work measured slower.
+        if (!res && R == S && R == 3 &&
+            (stride == 1 || stride == 2) && Cout == 64) {
+            int Kpad = (R * R * 8 + 63) / 64 * 64;
+            auto xp = at::empty({NB, H, W, 8}, x.options());
+            defer_hip::launch_pad_channels(bptr(x), bptr_mut(xp),
+                                           (long)NB * H * W, Cin, 8, s);
+            auto wp = cached_weight_prep(w, 1000 + Kpad, [&] {
+                auto t = at::empty({Cout, Kpad}, w.options());
+                defer_hip::launch_swin_repack_w(bptr(w), bptr_mut(t),
+                                               Cout, R, Cin, Kpad, s);
+                return t;
+            });
+            ConvParams q = p;
+            q.x = bptr(xp);
+            q.w = bptr(wp);
+            q.NB = NB; q.H = H; q.W = W; q.Cin = 8;
+            q.OH = OH; q.OW = OW; q.R = R; q.S = S;
+            q.stride = (int)stride; q.pad = (int)pad;
+            q.K = R * S * 8;
+            if (defer_hip::launch_conv_swin(q, relu, R, (int)stride, s))
+                return out;
+        }
         if ((stride * Cin) % 2 == 0) {
             // STEM path (ResNet 7x7/s2, Cin=3): spatially pre-pad the
             // raw input and gather K as (r, run)-major where a run is

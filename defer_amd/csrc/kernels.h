@@ -30,6 +30,13 @@ void launch_pad_channels(const void* x, void* y, long rows, int C, int C8,
 // STEM path helpers: spatial zero-pad + weight repack to (r, run) K-order
 void launch_pad2d(const void* x, void* y, int NB, int H, int W, int C,
                   int PH, int PW, int ph0, int pw0, hipStream_t s);
+// small-Cin window stem path (R in {3,7}, stride in {1,2}, Cout 64,
+// Cin pre-padded to 8; weights pre-repacked j-major zero-padded).
+// Returns false if the shape does not qualify (caller falls back).
+bool launch_conv_swin(const ConvParams& p, bool relu, int R, int stride,
+                      hipStream_t s);
+void launch_swin_repack_w(const void* w, void* wp, int Cout, int R,
+                          int C, int Kpad, hipStream_t s);
 void launch_stem_repack_w(const void* w, void* wp, int Cout, int R, int S,
                           int C, int TR, hipStream_t s);
 

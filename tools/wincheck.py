@@ -28,7 +28,33 @@ SHAPES = [  # (B, H, W, Cin, Cout, res)
     (3, 28, 28, 192, 128, False),    # TH=4 with masked cols
 ]
 
+# small-Cin window (stem) shapes: (B, H, W, Cin, Cout, R, stride, pad)
+SWIN = [
+    (2, 16, 32, 3, 64, 3, 1, 1),     # VGG-stem class
+    (2, 30, 28, 3, 64, 3, 1, 1),     # masked edge tiles
+    (2, 24, 32, 4, 64, 3, 2, 1),     # Cin=4, stride 2
+    (1, 224, 224, 3, 64, 3, 1, 1),   # full VGG b1.c1 shape
+]
+
 fails = 0
+for (B, H, W, Cin, Cout, R, stride, pad) in SWIN:
+    torch.manual_seed(1)
+    x = torch.randn(B, H, W, Cin, dtype=torch.bfloat16)
+    w = (torch.randn(Cout, R, R, Cin, dtype=torch.bfloat16) * 0.05)
+    sc = torch.rand(Cout) + 0.5
+    bi = torch.randn(Cout) * 0.1
+    want = ref.conv2d_bn_act(x.float(), w.float(), sc, bi, stride, pad,
+                             "relu", None)
+    y = ops.conv2d_bn_act(x.cuda(), w.cuda(), sc.cuda(), bi.cuda(),
+                          stride=stride, padding=pad, act="relu")
+    got = y.float().cpu()
+    rel = ((got - want).abs().max().item()
+           / max(want.abs().max().item(), 1e-6))
+    ok = rel < 0.05 and torch.isfinite(got).all()
+    print(f"swin {(B,H,W,Cin,Cout,R,stride)}: maxrel {rel:.4f} "
+          f"{'OK' if ok else 'FAIL'}")
+    fails += not ok
+
 for (B, H, W, Cin, Cout, res) in SHAPES:
     torch.manual_seed(0)
     x = torch.randn(B, H, W, Cin, dtype=torch.bfloat16)
