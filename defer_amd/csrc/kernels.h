@@ -19,7 +19,10 @@ struct ConvParams {
     int NB, H, W, Cin;
     int OH, OW, R, S, stride, pad;
     // magic-multiply reciprocals (filled by launch_conv_igemm):
-    // floor(n/d) = umulhi(n, ceil(2^32/d)) for n*d < 2^32
+    // floor(n/d) = umulhi(n, ceil(2^32/d)) for n*d < 2^32.
+    // The window kernels do not divide; they reuse `smul` as a
+    // full-sync debug flag (DEFER_CONV_VARIANT=W forces vmcnt(0)
+    // drains every phase).
     unsigned int owmul, ohmul, cmul, smul;
 };
 
