@@ -517,20 +517,22 @@ __global__ __launch_bounds__(TPB, WPS) void conv_igemm_kernel(
 // with the per-phase B stages (order: B first, then the slice).
 #define WTW 16                         // window tile width (fixed)
 
-template <int ACT, bool HAS_RES, int BN, int TH = 8>
-__global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
+template <int ACT, bool HAS_RES, int BN, int TH = 8, int TPB = NTHREADS,
+          int WPS = 2>
+__global__ __launch_bounds__(TPB, WPS) void conv_win_kernel(
     ConvParams p) {
     constexpr int TW = WTW;
     constexpr int BM = TH * TW;                 // 128 (TH8) / 64 (TH4)
     constexpr int WH = TH + 2, WW = TW + 2;     // window dims
     constexpr int WCH = WH * 8 * WW;            // 16-B chunks per window
-    constexpr int NSL = (WCH + 255) / 256;      // glds slices per wave
+    constexpr int NW = TPB / WAVE;              // waves per block
+    constexpr int NSL = (WCH + NW * 64 - 1) / (NW * 64);  // slices/wave
     constexpr int DB = 2;
     constexpr int WMW = BM / 32;                // waves along m
-    constexpr int WNW = 4 / WMW;                // waves along n
+    constexpr int WNW = NW / WMW;               // waves along n
     constexpr int WN = BN / WNW;
     constexpr int NI = WN / 16;
-    constexpr int BCH = BN / 32;
+    constexpr int BCH = BN * 8 / TPB;
     const bf16* __restrict__ X = (const bf16*)p.x;
     const bf16* __restrict__ Wt = (const bf16*)p.w;
     const bf16* __restrict__ Z = (const bf16*)p.zbuf;
@@ -601,8 +603,8 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
         w_okm = 0;
 #pragma unroll
         for (int j = 0; j < NSL; ++j) {
-            // slot j covers chunk (j*4 + wave)*64 + lane
-            int chunk = (j * 4 + wave) * 64 + lane;
+            // slot j covers chunk (j*NW + wave)*64 + lane
+            int chunk = (j * NW + wave) * 64 + lane;
             bool in = chunk < WCH;
             if (!in) chunk = WCH - 1;
             int iwl = chunk % WW;
@@ -620,7 +622,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
         // lanes past the last chunk must not WRITE (the clamped source
         // is harmless but an unguarded glds would land 16B past the
         // buffer — into the partner window / B ring)
-        int chunk = (j * 4 + wave) * 64 + lane;
+        int chunk = (j * NW + wave) * 64 + lane;
         if (chunk < WCH) {
             const bf16* src =
                 (w_okm >> j) & 1 ? w_base + w_loff[j] : Z;
@@ -678,10 +680,13 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
     };
 
     // ---- epilogue: bounce through the finished window buffer.
-    // RH rows per round, sized so RH*BN f32 fits the WCH*4-f32 buffer.
-    constexpr int RH = (WCH * 4 / BN) >= BM
-                           ? BM
-                           : ((WCH * 4 / BN) >= BM / 2 ? BM / 2 : BM / 4);
+    // RH rows per round, sized so RH*BN f32 fits the WCH*4-f32 buffer
+    // and at least one 16-B chunk per thread per round.
+    constexpr int RH0 = (WCH * 4 / BN) >= BM
+                            ? BM
+                            : ((WCH * 4 / BN) >= BM / 2 ? BM / 2
+                                                        : BM / 4);
+    constexpr int RH = (RH0 * BN >= TPB * 8) ? RH0 : BM;
     constexpr int ROUNDS = BM / RH;
     auto epilogue = [&](int nb, int oh0, int ow0, int wbuf) {
         float* scratch = (float*)(W0 + (long)wbuf * WCH * 8);
@@ -720,13 +725,13 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
             asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
             __builtin_amdgcn_s_barrier();
             if (interior) {
-                constexpr int CP8 = RH * BN / 8 / NTHREADS;
+                constexpr int CP8 = RH * BN / 8 / TPB;
                 f32x4 v4[CP8][2];
                 bf16x8 rv[CP8];
                 long off[CP8];
 #pragma unroll
                 for (int i = 0; i < CP8; ++i) {
-                    int chunk = tid + i * NTHREADS;
+                    int chunk = tid + i * TPB;
                     int rl = chunk / (BN / 8);
                     int c8 = (chunk % (BN / 8)) * 8;
                     int cs = c8 ^ (((rl >> 2) & 3) << 4);
@@ -755,10 +760,10 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
                     *reinterpret_cast<bf16x8*>(OUT + off[i]) = o;
                 }
             } else {
-                constexpr int CPR = RH * BN / 4 / NTHREADS;
+                constexpr int CPR = RH * BN / 4 / TPB;
 #pragma unroll
                 for (int i = 0; i < CPR; ++i) {
-                    int chunk = tid + i * NTHREADS;
+                    int chunk = tid + i * TPB;
                     int rl = chunk / (BN / 4);
                     int c4 = (chunk % (BN / 4)) * 4;
                     int cs = c4 ^ (((rl >> 2) & 3) << 4);
@@ -838,7 +843,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_win_kernel(
                 // WCH issues no glds and must not float one
                 bool sl = (j < NSL) && (w_count < local_wins) &&
                           (s_mt < mtiles) &&
-                          ((j * 4 + wave) * 64 < WCH);
+                          ((j * NW + wave) * 64 < WCH);
                 if (sl) stage_win_slice(j);
                 sliced_prev = sl;
                 compute(cbuf, j, (int)(ph & 1));
@@ -949,6 +954,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_swin_kernel(
     constexpr int WH = (TH - 1) * STRIDE + R;   // window rows
     constexpr int WW = (TW - 1) * STRIDE + R;   // window cols
     constexpr int WCH = WH * WW;                // 16-B chunks per window
+    constexpr int NW = NTHREADS / WAVE;         // waves per block (4)
     constexpr int NSL = (WCH + 255) / 256;      // glds slices per wave
     constexpr int KP = (R * R * 8 + 63) / 64 * 64;   // padded K
     constexpr int NK = KP / 64;                 // k-tiles (phases)
@@ -1025,7 +1031,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void conv_swin_kernel(
         }
     };
     auto stage_win_slice = [&](int j) {
-        int chunk = (j * 4 + wave) * 64 + lane;
+        int chunk = (j * NW + wave) * 64 + lane;
         if (chunk < WCH) {
             const bf16* src =
                 (w_okm >> j) & 1 ? w_base + w_loff[j] : Z;
@@ -1271,6 +1277,12 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
     // (56x56x256-class) stay on the BN128 RSC config: the BN64 window
     // kernel pays 2-4x the B re-staging (measured loss) and a BN128
     // window instantiation spills ~250 B/lane at this register budget.
+    // Cout <= 128 only. Wider variants were all built and measured
+    // slower than the BN128 RSC config they would replace: BN128 x 4
+    // waves spills ~250 B/lane; BN128 x 512 threads at 4 waves/SIMD
+    // spills ~200 B/lane; at 2 waves/SIMD it runs ONE 8-wave block per
+    // CU and loses the cross-block epilogue overlap (b3 179 -> 211 us,
+    // same failure mode as the BM256 implicit-GEMM tile).
     const bool win_ok =
         (!legacy && !force_small) && !gemm_mode && !stem_mode &&
         p.R == 3 && p.S == 3 && p.stride == 1 && p.pad == 1 &&
@@ -1291,10 +1303,11 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
                       : (force_win && win_fit(4)) ? 4
                                        : 0;
     if (THsel) {
+        const int BNw = 64;
         const long wth = (p.OH + THsel - 1) / THsel;
         const long wtw = (p.OW + WTW - 1) / WTW;
         const int mt2 = (int)((long)p.NB * wth * wtw);
-        const int nyw = p.Cout / 64;
+        const int nyw = p.Cout / BNw;
         int gxw = mt2;
         if ((long)mt2 * nyw > 768) {
             gxw = 768 / nyw > 0 ? 768 / nyw : 1;
@@ -1302,12 +1315,14 @@ void launch_conv_igemm(const ConvParams& p0, bool relu, bool has_res,
         }
         p.smul = (variant == 'W') ? 1u : 0u;   // full-sync debug flag
         dim3 gw(gxw, nyw), bw(NTHREADS);
-#define WIN_TILE(A, RZ, TH)                                               \
-    hipLaunchKernelGGL((conv_win_kernel<A, RZ, 64, TH>), gw, bw, 0, s, p)
+#define WIN_TILE(A, RZ, BNv, TPBv, WPSv)                                  \
+    hipLaunchKernelGGL((conv_win_kernel<A, RZ, BNv, 8, TPBv, WPSv>), gw,  \
+                       bw, 0, s, p)
 #define WIN_BN(A, RZ)                                                     \
     do {                                                                  \
-        if (THsel == 8) WIN_TILE(A, RZ, 8);                               \
-        else WIN_TILE(A, RZ, 4);                                          \
+        if (THsel == 8) WIN_TILE(A, RZ, 64, 256, 2);                      \
+        else hipLaunchKernelGGL((conv_win_kernel<A, RZ, 64, 4>), gw, bw,  \
+                                0, s, p);                                 \
     } while (0)
         if (relu) {
             if (has_res) WIN_BN(ACT_RELU, true);
