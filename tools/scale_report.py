@@ -1,0 +1,49 @@
+#!/usr/bin/env python3
+"""Scaling-curve report: feed it bench.py JSON lines (one per GPU count)
+and it prints the 1/2/4/8-stage curve with speedup and efficiency — the
+final report SURVEY.md §5 calls for, matching the reference's published
+metric shape (throughput vs node count, README.md:12: +53% at 8 nodes).
+
+    python tools/scale_report.py BENCH_N1.json BENCH_N2.json ...
+    cat results.jsonl | python tools/scale_report.py
+"""
+import json
+import sys
+
+
+def main(argv):
+    lines = []
+    if len(argv) > 1:
+        for p in argv[1:]:
+            with open(p) as f:
+                lines += [ln for ln in f if ln.strip().startswith("{")]
+    else:
+        lines = [ln for ln in sys.stdin if ln.strip().startswith("{")]
+    runs = sorted((json.loads(ln) for ln in lines),
+                  key=lambda d: d["n_gpus"])
+    if not runs:
+        print("no bench JSON lines found", file=sys.stderr)
+        return 1
+    base = runs[0]
+    print(f"# {base['metric']}  ({base['config']['model']}, "
+          f"batch {base['config']['global_batch']}, {base['dtype']}, "
+          f"{base['data']})")
+    print(f"{'gpus':>4} {'images/sec':>12} {'ms/item':>9} "
+          f"{'speedup':>8} {'efficiency':>10}  parallelism")
+    for r in runs:
+        n = r["n_gpus"]
+        sp = r["value"] / base["value"] * base["n_gpus"]
+        eff = sp / n * 100
+        print(f"{n:>4} {r['value']:>12.1f} {r['ms_per_step']:>9.3f} "
+              f"{sp:>7.2f}x {eff:>9.1f}%  "
+              f"{r['config'].get('parallelism', '?')}")
+    ref = "+53% at 8 nodes (reference, README.md:12)"
+    if runs[-1]["n_gpus"] > 1:
+        gain = (runs[-1]["value"] / base["value"] - 1) * 100
+        print(f"\nvs single device: {gain:+.0f}% at {runs[-1]['n_gpus']} "
+              f"GPUs — reference baseline: {ref}")
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main(sys.argv))
