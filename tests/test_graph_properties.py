@@ -1,0 +1,110 @@
+"""Property-based tests of the LayerGraph partitioner core (the dag_util
+equivalent): on random skip-connected DAGs,
+
+- `valid_cut_points` must match a brute-force articulation check, and
+- splitting at ANY subset of valid cuts must reproduce the whole-graph
+  forward bitwise (the reference's part1..partN contract,
+  dispatcher.py:27-42).
+"""
+
+import torch
+import torch.nn as nn
+from hypothesis import given, settings, strategies as st
+
+from defer_amd.graph import GraphModel, GraphNode, LayerGraph
+
+
+def _build(n_nodes: int, skips: list) -> LayerGraph:
+    """Chain of Linear(4,4) nodes; skips[i] = (src, dst) adds an extra
+    edge via an add node after dst."""
+    torch.manual_seed(n_nodes * 31 + len(skips))
+    nodes = [GraphNode("n0", nn.Linear(4, 4), ["input"])]
+    for i in range(1, n_nodes):
+        nodes.append(GraphNode(f"n{i}", nn.Linear(4, 4), [f"n{i-1}"]))
+    g_nodes = list(nodes)
+    # splice add nodes (skip connections) after their dst position
+    for j, (src, dst) in enumerate(skips):
+        name = f"add{j}"
+        # insert after dst's current position
+        pos = max(i for i, n in enumerate(g_nodes)
+                  if n.name == f"n{dst}")
+        prev = g_nodes[pos].name
+        g_nodes.insert(pos + 1,
+                       GraphNode(name, lambda a, b: a + b,
+                                 [prev, f"n{src}"]))
+        # rewire the following node to consume the add
+        if pos + 2 < len(g_nodes):
+            nxt = g_nodes[pos + 2]
+            g_nodes[pos + 2] = GraphNode(
+                nxt.name, nxt.layer,
+                [name if p == prev else p for p in nxt.inputs],
+                nxt.kwargs)
+    return LayerGraph(g_nodes)
+
+
+def _brute_force_cuts(g: LayerGraph):
+    """A cut after node i is valid iff no earlier node's output (or the
+    input) is consumed after i."""
+    pos = {n.name: i for i, n in enumerate(g.nodes)}
+    cuts = []
+    for i in range(len(g.nodes) - 1):
+        ok = True
+        for n in g.nodes:
+            if pos[n.name] <= i:
+                continue
+            for p in n.inputs:
+                if p == LayerGraph.INPUT or pos[p] < i:
+                    ok = False
+        if ok:
+            cuts.append(g.nodes[i].name)
+    return cuts
+
+
+@st.composite
+def graphs(draw):
+    n = draw(st.integers(min_value=2, max_value=10))
+    n_skips = draw(st.integers(min_value=0, max_value=3))
+    skips = []
+    for _ in range(n_skips):
+        src = draw(st.integers(min_value=0, max_value=n - 2))
+        dst = draw(st.integers(min_value=src + 1, max_value=n - 1))
+        skips.append((src, dst))
+    return _build(n, skips)
+
+
+@settings(max_examples=60, deadline=None)
+@given(graphs(), st.data())
+def test_valid_cuts_match_brute_force_and_split_is_exact(g, data):
+    assert g.valid_cut_points() == _brute_force_cuts(g)
+
+    valid = g.valid_cut_points()
+    if valid:
+        k = data.draw(st.integers(min_value=1, max_value=len(valid)))
+        idx = sorted(data.draw(
+            st.lists(st.integers(0, len(valid) - 1), min_size=k,
+                     max_size=k, unique=True)))
+        cuts = [valid[i] for i in idx]
+        stages = g.split(cuts)
+        assert len(stages) == len(cuts) + 1
+        x = torch.randn(3, 4)
+        with torch.no_grad():
+            want = g.forward(x)
+            z = x
+            for s in stages:
+                z = s.forward(z)
+        assert torch.equal(z, want)
+
+
+@settings(max_examples=30, deadline=None)
+@given(graphs())
+def test_graphmodel_parameters_partition_cleanly(g):
+    """Every parameter of the whole model appears in exactly one stage
+    (weights ship once, dispatcher.py:57)."""
+    gm = GraphModel(g)
+    valid = g.valid_cut_points()
+    if not valid:
+        return
+    stages = [GraphModel(s) for s in g.split([valid[len(valid) // 2]])]
+    total = sum(p.numel() for p in gm.parameters())
+    split_total = sum(p.numel() for s in stages for p in s.parameters())
+    assert split_total == total

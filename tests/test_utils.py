@@ -107,3 +107,25 @@ def test_dist_pipeline_partition_dump(tmp_path):
     assert pipe.stats.compute_ms > 0
     pipe.reset_stats()
     assert pipe.stats.compute_ms == 0
+
+
+def test_scale_report_tool(tmp_path):
+    import json
+    import subprocess
+    import sys
+
+    rows = [dict(metric="images/sec (whole node) ResNet50 pipeline",
+                 value=v, unit="images/sec", n_gpus=n, ms_per_step=1.0,
+                 dtype="bf16", data="synthetic",
+                 config=dict(model="resnet50", global_batch=64,
+                             parallelism=f"pp{n}"))
+            for n, v in [(1, 1000.0), (2, 1900.0), (4, 3500.0)]]
+    p = tmp_path / "runs.jsonl"
+    p.write_text("".join(json.dumps(r) + "\n" for r in rows))
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, "tools", "scale_report.py"),
+         str(p)], capture_output=True, text=True, timeout=60)
+    assert out.returncode == 0, out.stderr
+    assert "1.90x" in out.stdout and "95.0%" in out.stdout
+    assert "+250%" in out.stdout
