@@ -111,6 +111,7 @@ def test_dist_pipeline_partition_dump(tmp_path):
 
 def test_scale_report_tool(tmp_path):
     import json
+    import os
     import subprocess
     import sys
 
