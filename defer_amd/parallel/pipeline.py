@@ -417,10 +417,17 @@ class DistPipeline:
 
         if tm:
             self.stats.compute_ms = tm.total_ms()
-        # drain outstanding sends
+        # drain outstanding sends, CLEARING the slots: a consumed Work
+        # must never be waited again (gloo's Work.wait() is not
+        # idempotent — re-waiting one deadlocks, which hung the second
+        # run() call of a warmup+timed sequence at world_size > 1)
         for ring in (self.send_ring, self.result_ring):
             if ring is not None:
-                for w in list(ring.works) + list(
-                        getattr(ring, "size_works", [])):
+                for i, w in enumerate(ring.works):
                     if w is not None:
                         w.wait()
+                        ring.works[i] = None
+                for i, w in enumerate(getattr(ring, "size_works", [])):
+                    if w is not None:
+                        w.wait()
+                        ring.size_works[i] = None

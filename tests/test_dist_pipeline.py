@@ -119,3 +119,27 @@ def test_eight_stage_auto_cuts_gloo():
     """The headline 8-stage chain topology (7 relay hops
     + result return), auto-partitioned, world_size 8 on gloo."""
     _run(8, None, steps=3)
+
+
+def test_bench_torchrun_two_ranks_cpu():
+    """The driver's exact launch form (torch.distributed.run, nnodes=1,
+    loopback rendezvous) runs bench.py end to end on CPU with 2 ranks
+    and emits exactly one JSON line on stdout."""
+    import json
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29791", os.path.join(root, "bench.py"),
+         "--gpus", "2", "--device", "cpu", "--batch", "1",
+         "--steps", "4", "--warmup", "1"],
+        capture_output=True, text=True, timeout=600, cwd=root)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [ln for ln in out.stdout.splitlines() if ln.strip()]
+    assert len(lines) == 1, out.stdout
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 2 and d["config"]["parallelism"] == "pp2"
+    assert d["value"] > 0
