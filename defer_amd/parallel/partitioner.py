@@ -13,7 +13,7 @@ because on MI355X the steady-state pipeline throughput is
 max-over-stages(compute, xGMI relay).
 """
 
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import torch
 import torch.nn as nn

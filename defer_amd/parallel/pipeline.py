@@ -16,14 +16,14 @@ Reference behavior being rebuilt:
 import queue
 import threading
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Callable, List, Optional
 
 import torch
 import torch.distributed as dist
 
 from defer_amd.config import PipelineConfig
-from defer_amd.graph import GraphModel, LayerGraph
+from defer_amd.graph import GraphModel
 from defer_amd.parallel.comm import Codec, P2PRing, dtype_bytes, make_ring
 from defer_amd.parallel.partitioner import (as_graph_model, auto_partition,
                                             partition_model)

@@ -74,7 +74,6 @@ def test_node_costs_total_flops():
 def test_fusion_preserves_semantics():
     from defer_amd.graph import GraphModel
     from defer_amd.parallel.fusion import fuse_residual_adds
-    from defer_amd.models.layers import ConvBNAct, AddAct
 
     m = resnet50()
     x = torch.randn(1, 64, 64, 3)

@@ -5,7 +5,6 @@ reference node.py:39), per-stage event tracing, fail-fast stage errors
 import queue
 import threading
 
-import pytest
 import torch
 
 from defer_amd import DEFER, PipelineConfig
