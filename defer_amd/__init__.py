@@ -21,5 +21,5 @@ dispatcher.py:21,107):
 __version__ = "0.1.0"
 
 from defer_amd.config import PipelineConfig  # noqa: F401
-from defer_amd.parallel.pipeline import DEFER  # noqa: F401
+from defer_amd.parallel.pipeline import DEFER, DistPipeline  # noqa: F401
 from defer_amd.parallel.partitioner import partition_model, auto_partition  # noqa: F401
