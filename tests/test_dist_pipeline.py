@@ -143,3 +143,49 @@ def test_bench_torchrun_two_ranks_cpu():
     d = json.loads(lines[0])
     assert d["n_gpus"] == 2 and d["config"]["parallelism"] == "pp2"
     assert d["value"] > 0
+
+
+def _worker_two_runs(rank, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        torch.manual_seed(0)
+        from defer_amd.parallel.pipeline import DistPipeline
+
+        cfg = PipelineConfig(device="cpu", dtype="fp32", ring_depth=4,
+                             backend="gloo")
+        pipe = DistPipeline(resnet50(), cfg, (1, 64, 64, 3))
+        torch.manual_seed(7)
+        xs = [torch.randn(1, 64, 64, 3) for _ in range(6)]
+        got = {}
+        # warmup-then-timed shape: MULTIPLE run() calls on one pipeline
+        # (regression: drained send Works were re-waited and deadlocked)
+        pipe.run(2, feed=lambda k: xs[k],
+                 collect=lambda k, y: got.__setitem__(k, y.clone()))
+        dist.barrier()
+        pipe.run(4, feed=lambda k: xs[2 + k],
+                 collect=lambda k, y: got.__setitem__(2 + k,
+                                                      y.clone()))
+        if rank == 0:
+            assert sorted(got) == list(range(6))
+            q.put(("done", rank, None))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_two_sequential_runs_world2():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_worker_two_runs, args=(r, 2, 29641, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=420)
+    alive = [p.is_alive() for p in procs]
+    for p in procs:
+        if p.is_alive():
+            p.terminate()
+    assert not any(alive), "second run() deadlocked"
+    assert all(p.exitcode == 0 for p in procs)
