@@ -29,8 +29,11 @@ def parse_args():
     ap.add_argument("--model", default="resnet50",
                     choices=["resnet50", "resnet101", "resnet152",
                              "vgg19"])
-    ap.add_argument("--batch", type=int, default=64,
-                    help="images per pipeline micro-batch")
+    ap.add_argument("--batch", type=int, default=256,
+                    help="images per pipeline micro-batch (throughput "
+                         "saturates ~128-256; see profiles/README.md "
+                         "batch table; the reference streams batch-1 "
+                         "items, test/test.py:20-23)")
     ap.add_argument("--compression", default="none",
                     choices=["none", "zfp", "zfp+lz4"])
     ap.add_argument("--zfp-bits", type=int, default=8)
