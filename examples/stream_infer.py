@@ -46,9 +46,21 @@ def main():
                                          else "cpu",
                                          dtype="bf16" if on_gpu
                                          else "fp32"))
+    if cuts is not None and len(cuts) + 1 != args.nodes:
+        raise SystemExit(f"{len(cuts)} cuts make {len(cuts) + 1} stages "
+                         f"but --nodes is {args.nodes}")
+
     inputs, outputs = queue.Queue(10), queue.Queue(10)
-    t = threading.Thread(target=engine.run_defer,
-                         args=(model, cuts, inputs, outputs))
+    err = []
+
+    def serve():
+        try:
+            engine.run_defer(model, cuts, inputs, outputs)
+        except Exception as e:              # surface engine errors to the
+            err.append(e)                   # feeder instead of hanging it
+            outputs.put(None)
+
+    t = threading.Thread(target=serve)
     t.start()
 
     dtype = torch.bfloat16 if on_gpu else torch.float32
@@ -58,7 +70,12 @@ def main():
     for _ in range(args.items):
         inputs.put(x)
     inputs.put(None)                      # clean shutdown
-    got = [outputs.get(timeout=600) for _ in range(args.items)]
+    got = []
+    for _ in range(args.items):
+        y = outputs.get(timeout=600)
+        if y is None:
+            raise SystemExit(f"pipeline failed: {err[0]!r}")
+        got.append(y)
     t.join(timeout=600)
     dt = time.perf_counter() - t0
     print(f"{args.items} items x batch {args.batch} through "
