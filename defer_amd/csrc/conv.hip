@@ -1,5 +1,16 @@
-// Implicit-GEMM convolution for gfx950 (CDNA4), NHWC bf16, MFMA
-// 16x16x32 with fp32 accumulation.
+// Convolution kernels for gfx950 (CDNA4), NHWC bf16, MFMA 16x16x32
+// with fp32 accumulation. Three kernel families, selected by a measured
+// per-shape policy in launch_conv_igemm (docs/OPTIMIZATION_LOG.md has
+// the A/B numbers behind every choice):
+//
+//   conv_igemm_kernel — implicit GEMM, the general path (1x1s, deep-K
+//     3x3s, strided convs, the dense head, the 7x7 stem's run-packed
+//     gather);
+//   conv_win_kernel   — window-reuse 3x3/s1: an 8x16 output tile stages
+//     its input window to LDS once per 64-channel block and slides the
+//     nine (r,s) offsets in LDS (9x less A staging);
+//   conv_swin_kernel  — the small-Cin (<8, padded to 8) window variant
+//     for VGG-style 3x3 stems.
 //
 // GEMM view: C[M, N] = A[M, K] x B[K, N]
 //   M = NB*OH*OW (output pixels), N = Cout, K = R*S*Cin,
@@ -7,13 +18,13 @@
 //   B = weights, stored OHWI [Cout][R][S][Cin] so B^T rows are contiguous.
 //
 // Structure (cdna_hip_programming.md §5, T3+T4 counted-vmcnt pipeline):
-// template-parameterized BM x BN output tile, BK=64, 4 waves,
-// DEPTH-buffered LDS staged by global_load_lds width 16 (lane-linear
+// template-parameterized BM x BN output tile, BK=64, TPB/64 waves,
+// DA-deep LDS rings staged by global_load_lds width 16 (lane-linear
 // dest; XOR swizzle applied to the *source* chunk index and the read
-// address — rule 21). The stage cursor runs DEPTH-1 K-tiles ahead; each
-// iteration waits a COUNTED s_waitcnt vmcnt(OPS) — never 0 mid-loop —
-// and a raw s_barrier, so glds stay in flight across barriers and HBM
-// latency hides under MFMA. Each block walks multiple m-tiles
+// address — rule 21). The stage cursor runs DA-1 K-tiles ahead; each
+// iteration waits a COUNTED s_waitcnt vmcnt — never 0 mid-loop —
+// and a raw s_barrier, so glds stay in flight across barriers and
+// load latency hides under MFMA. Each block walks multiple m-tiles
 // (grid-stride) so the pipeline never drains between tiles; when the
 // whole K fits one tile, the weight tile is staged once (B_PERSIST).
 //
