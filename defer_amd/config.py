@@ -28,7 +28,13 @@ class PipelineConfig:
     device: str = "cuda"          # "cuda" (MI355X) or "cpu" (tests/plumbing)
     dtype: str = "bf16"           # compute dtype on GPU; "fp32" on CPU
     batch_size: int = 64          # images per pipeline item (micro-batch)
-    use_hip_graphs: bool = True   # capture each stage's forward in a hipGraph
+    # Capture each stage's forward in a hipGraph. Off by default:
+    # measured no gain at bench batch sizes (launch overhead is already
+    # negligible, profiles/README.md), and concurrent captures from the
+    # threaded orchestrator's workers race on the shared legacy stream
+    # (observed intermittent first-item corruption). DistPipeline (one
+    # process per GPU) can enable it safely via bench --graphs.
+    use_hip_graphs: bool = False
 
     # --- inter-stage relay (the data plane) -------------------------------
     # Codec for boundary activations. The reference compresses with

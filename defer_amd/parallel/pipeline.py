@@ -50,8 +50,12 @@ class StageExecutor:
     Weights are pre-cast: conv/dense weights to the compute dtype, folded
     BN scale/bias kept fp32 for the HIP kernel epilogues. Optionally
     captures the stage forward into a hipGraph (torch.cuda.CUDAGraph) to
-    amortize launch overhead for small batches.
+    amortize launch overhead for small batches; captures are serialized
+    process-wide (hipGraph capture is not safe concurrently with other
+    threads' work on the shared legacy stream).
     """
+
+    _capture_lock = threading.Lock()
 
     def __init__(self, stage: GraphModel, device, dtype: torch.dtype,
                  use_graph: bool = False, fuse: bool = True):
@@ -95,7 +99,8 @@ class StageExecutor:
                 return self.model(x)
         if self._graph is None:
             try:
-                self._capture(x)
+                with StageExecutor._capture_lock:
+                    self._capture(x)
             except Exception:
                 # capture not supported for this stage: eager fallback
                 self.use_graph = False
