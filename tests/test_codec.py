@@ -84,3 +84,32 @@ def test_gpu_roundtrip_bf16():
     rel = (x.float() - y.float()).abs().max() / x.float().abs().max()
     assert rel < 0.15
     assert w.numel() * 1.0 / (x.numel() * 2) < 0.51  # >=2x vs bf16
+
+
+def test_spec_properties_random_shapes():
+    """Property sweep of the numpy spec: for random field shapes and
+    scales, the fixed-rate stream has the exact declared size, decode
+    never produces non-finite values, and the error bound tightens with
+    rate (hypothesis-driven shapes)."""
+    from hypothesis import given, settings, strategies as st
+
+    @settings(max_examples=40, deadline=None)
+    @given(st.integers(1, 9), st.integers(1, 9), st.integers(1, 17),
+           st.floats(1e-3, 1e3), st.integers(0, 2**31 - 1))
+    def check(d0, d1, d2, scale, seed):
+        rng = np.random.default_rng(seed)
+        a = (rng.standard_normal((d0, d1, d2)) * scale).astype(np.float32)
+        prev = None
+        for rate in (6, 12):
+            w = zfp_ref.encode(a, rate)
+            assert w.nbytes == zfp_ref.wire_bytes(a.shape, rate)
+            b = zfp_ref.decode(w, a.shape, rate)
+            assert np.isfinite(b).all()
+            err = float(np.abs(a - b).max())
+            if prev is not None:
+                assert err <= prev + 1e-6
+            prev = err
+        # rate 12 keeps max rel error small on smooth-scaled data
+        assert prev <= max(1e-6, 0.05 * float(np.abs(a).max()))
+
+    check()
