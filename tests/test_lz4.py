@@ -87,3 +87,36 @@ def test_gpu_roundtrip_on_zfp_payload():
     rel = ((y.float() - x.float()).abs().max()
            / x.float().abs().max()).item()
     assert rel < 0.15, rel
+
+
+def test_spec_roundtrip_property_sweep():
+    """Hypothesis sweep: compress∘decompress is the identity for
+    arbitrary byte strings (mixed runs/random segments), and the wire
+    never exceeds the declared worst-case bound."""
+    from hypothesis import given, settings, strategies as st
+
+    @settings(max_examples=60, deadline=None)
+    @given(st.lists(
+        st.tuples(st.sampled_from(["run", "rand", "cycle"]),
+                  st.integers(1, 3000), st.integers(0, 255)),
+        min_size=1, max_size=6), st.integers(0, 2**31 - 1))
+    def check(segments, seed):
+        rng = np.random.default_rng(seed)
+        parts = []
+        for kind, n, b in segments:
+            if kind == "run":
+                parts.append(np.full(n, b, np.uint8))
+            elif kind == "cycle":
+                parts.append(np.resize(
+                    np.arange(b % 7 + 2, dtype=np.uint8), n))
+            else:
+                parts.append(rng.integers(0, 256, n).astype(np.uint8))
+        data = np.concatenate(parts)
+        comp = lz4_ref.compress(data)
+        # worst-case wire bound (comm.py's VarP2PRing buffer sizing)
+        nb = (data.nbytes + lz4_ref.BLK - 1) // lz4_ref.BLK
+        assert comp.nbytes <= 4 * (2 + nb + 1) + nb * 4352
+        out = lz4_ref.decompress(comp)
+        assert np.array_equal(out, data)
+
+    check()
