@@ -108,3 +108,36 @@ def test_graphmodel_parameters_partition_cleanly(g):
     total = sum(p.numel() for p in gm.parameters())
     split_total = sum(p.numel() for s in stages for p in s.parameters())
     assert split_total == total
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.integers(1, 4), st.integers(0, 2**31 - 1), st.booleans())
+def test_fusion_preserves_forward_and_names(nblocks, seed, act_none):
+    """fuse_residual_adds: conv(act=none)+AddAct pairs collapse into the
+    fused module without changing the forward result or losing the
+    add_N names (the partition cut points)."""
+    from defer_amd.models.layers import AddAct, ConvBNAct
+    from defer_amd.parallel.fusion import fuse_residual_adds
+
+    torch.manual_seed(seed % (2**31))
+    nodes = [GraphNode("c0", ConvBNAct(8, 16, 3, 1, 1, "relu"),
+                       ["input"])]
+    x_name, cin = "c0", 16
+    for b in range(nblocks):
+        conv = ConvBNAct(cin, cin, 3, 1, 1,
+                         "none" if not act_none else "relu")
+        nodes.append(GraphNode(f"b{b}_conv", conv, [x_name]))
+        nodes.append(GraphNode(f"add_{b}",
+                               AddAct("relu"),
+                               [f"b{b}_conv", x_name]))
+        x_name = f"add_{b}"
+    g = LayerGraph(nodes)
+    fused = fuse_residual_adds(g)
+    names = [n.name for n in fused.nodes]
+    for b in range(nblocks):
+        assert f"add_{b}" in names          # cut names preserved
+    x = torch.randn(2, 12, 12, 8)
+    with torch.no_grad():
+        a = g.forward(x)
+        b = fused.forward(x)
+    assert torch.allclose(a, b, atol=1e-5), float((a - b).abs().max())
