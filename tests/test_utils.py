@@ -129,3 +129,18 @@ def test_scale_report_tool(tmp_path):
     assert out.returncode == 0, out.stderr
     assert "1.90x" in out.stdout and "95.0%" in out.stdout
     assert "+250%" in out.stdout
+
+
+def test_visualize_function_layers():
+    """DOT/describe handle non-module (lambda / function) layers — the
+    FX frontend produces them for `x + y`, relu, etc."""
+    from defer_amd.graph import GraphNode, LayerGraph
+
+    g = LayerGraph([
+        GraphNode("lin", torch.nn.Linear(4, 4), ["input"]),
+        GraphNode("add", lambda a, b: a + b, ["lin", "input"]),
+    ])
+    dot = to_dot(g, name="fn")
+    assert '"add"' in dot and "digraph" in dot
+    txt = describe(g, name="fn")
+    assert "add" in txt and "<lambda>" in txt or "fn" in txt
