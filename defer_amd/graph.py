@@ -203,15 +203,44 @@ def from_torch(model: nn.Module) -> LayerGraph:
 
                 def fn(x, *a, _m=meth, **k):
                     return getattr(x, _m)(*a, **k)
-            ins = [name_map[a.name] for a in node.args
-                   if isinstance(a, fx.Node)]
-            consts = [a for a in node.args if not isinstance(a, fx.Node)]
+
+            # Tensor inputs can appear anywhere in the argument tree
+            # (e.g. torch.cat([a, b], dim=-1)): replace every fx.Node
+            # with a positional placeholder and rebuild at call time.
+            class _Ref:
+                __slots__ = ("i",)
+
+                def __init__(self, i):
+                    self.i = i
+
+            refs: List[str] = []
+
+            def scan(a):
+                if isinstance(a, fx.Node):
+                    refs.append(a.name)
+                    return _Ref(len(refs) - 1)
+                if isinstance(a, (list, tuple)):
+                    return type(a)(scan(x) for x in a)
+                return a
+
+            sargs = tuple(scan(a) for a in node.args)
+            skwargs = {k: scan(v) for k, v in node.kwargs.items()}
+            ins = [name_map[r] for r in refs]
             nm = node.name
 
-            def wrapped(*tensors, _fn=fn, _consts=tuple(consts), **kw):
-                return _fn(*tensors, *_consts, **kw)
+            def wrapped(*tensors, _fn=fn, _sa=sargs, _sk=skwargs,
+                        _Ref=_Ref):
+                def subst(a):
+                    if isinstance(a, _Ref):
+                        return tensors[a.i]
+                    if isinstance(a, (list, tuple)):
+                        return type(a)(subst(x) for x in a)
+                    return a
 
-            nodes.append(GraphNode(nm, wrapped, ins, dict(node.kwargs)))
+                return _fn(*(subst(a) for a in _sa),
+                           **{k: subst(v) for k, v in _sk.items()})
+
+            nodes.append(GraphNode(nm, wrapped, ins))
             name_map[node.name] = nm
         elif node.op == "output":
             out_arg = node.args[0]

@@ -115,3 +115,22 @@ def test_graph_model_stage_models():
     # parameters are owned (state_dict round-trips)
     sd = parts[0].state_dict()
     assert any("layers.a" in k for k in sd)
+
+
+def test_from_torch_nested_tensor_args():
+    """Tensor inputs inside list/tuple arguments (torch.cat) and keyword
+    constants resolve correctly through the FX frontend."""
+    class Net(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.l = nn.Linear(4, 4)
+
+        def forward(self, x):
+            y = self.l(x)
+            z = torch.cat([y, x], dim=-1)
+            return torch.clamp(z, min=-1.0, max=1.0).sum(dim=-1)
+
+    net = Net()
+    g = from_torch(net)
+    x = torch.randn(3, 4)
+    assert torch.allclose(g.forward(x), net(x))
