@@ -87,3 +87,36 @@ def test_defer_arbitrary_torch_model():
     outs, _ = _run_defer(net, None, ["cpu", "cpu"], xs, cfg=cfg)
     for o, w in zip(outs, want):
         assert torch.allclose(o, w, atol=1e-6)
+
+
+def test_defer_concat_dag_model():
+    """DenseNet-style torch.cat DAG through from_torch + auto-partition +
+    the DEFER orchestrator: tensor inputs nested inside the cat list must
+    survive FX conversion, and valid cuts must respect the long-range
+    concat edges (every block consumes the original input)."""
+    import torch.nn as nn
+
+    class Dense(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.c1 = nn.Conv2d(3, 8, 3, padding=1)
+            self.c2 = nn.Conv2d(11, 8, 3, padding=1)   # cat(x, f1)
+            self.c3 = nn.Conv2d(19, 8, 3, padding=1)   # cat(x, f1, f2)
+            self.head = nn.Linear(8, 5)
+
+        def forward(self, x):
+            f1 = torch.relu(self.c1(x))
+            f2 = torch.relu(self.c2(torch.cat([x, f1], dim=1)))
+            f3 = torch.relu(self.c3(torch.cat([x, f1, f2], dim=1)))
+            return self.head(f3.mean(dim=(2, 3)))
+
+    torch.manual_seed(0)
+    net = Dense().eval()
+    xs = [torch.randn(2, 3, 16, 16) for _ in range(2)]
+    with torch.no_grad():
+        want = [net(x) for x in xs]
+    cfg = PipelineConfig(device="cpu", dtype="fp32",
+                         input_shape=(1, 3, 16, 16))
+    outs, _ = _run_defer(net, None, ["cpu", "cpu"], xs, cfg=cfg)
+    for o, w in zip(outs, want):
+        assert torch.allclose(o, w, atol=1e-6)
