@@ -144,3 +144,58 @@ def test_visualize_function_layers():
     assert '"add"' in dot and "digraph" in dot
     txt = describe(g, name="fn")
     assert "add" in txt and "<lambda>" in txt or "fn" in txt
+
+
+def test_powerbench_tool_wraps_json_line(tmp_path):
+    """tools/powerbench.py relays the wrapped command's JSON line and
+    augments it with power fields (null watts where rocm-smi has no GPU),
+    and propagates a non-zero exit code."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    tool = os.path.join(root, "tools", "powerbench.py")
+    inner = ("import json; print('banner noise'); "
+             "print(json.dumps({'metric': 'm', 'value': 10.0, "
+             "'unit': 'images/sec'}))")
+    out = subprocess.run(
+        [sys.executable, tool, "--", sys.executable, "-c", inner],
+        capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr
+    row = json.loads(out.stdout.strip().splitlines()[-1])
+    assert row["metric"] == "m" and row["value"] == 10.0
+    assert "avg_watts" in row and "power_samples" in row
+
+    fail = subprocess.run(
+        [sys.executable, tool, "--", sys.executable, "-c",
+         "import sys; sys.exit(3)"],
+        capture_output=True, text=True, timeout=120)
+    assert fail.returncode == 3
+
+
+def test_powerbench_parses_rocm_smi_json():
+    """read_power_w sums per-card socket power from rocm-smi --json
+    output (skipping caps/max fields)."""
+    import importlib.util
+    import json
+    import os
+    import sys
+    from unittest import mock
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    spec = importlib.util.spec_from_file_location(
+        "powerbench", os.path.join(root, "tools", "powerbench.py"))
+    pb = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(pb)
+
+    fake = json.dumps({
+        "card0": {"Average Graphics Package Power (W)": "312.0",
+                  "Max Graphics Package Power (W)": "1400.0"},
+        "card1": {"Current Socket Graphics Package Power (W)": "88.5"},
+    })
+    done = mock.Mock()
+    done.stdout = fake
+    with mock.patch.object(pb.subprocess, "run", return_value=done):
+        assert abs(pb.read_power_w() - 400.5) < 1e-9
