@@ -43,6 +43,11 @@ def parse_args():
                     help="capture stage forwards in hipGraphs")
     ap.add_argument("--device", default="cuda", choices=["cuda", "cpu"])
     ap.add_argument("--ring-depth", type=int, default=4)
+    ap.add_argument("--dual-rail", action="store_true",
+                    help="split each fixed-size hop across two xGMI "
+                         "rails (direct + via an idle third GPU); "
+                         "doubles effective hop bandwidth for "
+                         "boundaries above one link's 153 GB/s")
     ap.add_argument("--no-return-results", action="store_true")
     ap.add_argument("--stats", action="store_true",
                     help="print per-stage stats to stderr (hipEvent "
@@ -98,7 +103,7 @@ def main():
         dtype="bf16" if args.device == "cuda" else "fp32",
         batch_size=args.batch, use_hip_graphs=args.graphs,
         compression=args.compression, zfp_rate_bits=args.zfp_bits,
-        ring_depth=args.ring_depth,
+        ring_depth=args.ring_depth, dual_rail=args.dual_rail,
         backend="nccl" if args.device == "cuda" else "gloo",
         return_results=not args.no_return_results,
         log_stage_stats=args.stats,
@@ -193,6 +198,7 @@ def main():
                 + (f"@{args.zfp_bits}b" if args.compression != "none"
                    else ""),
                 "hip_graphs": bool(args.graphs),
+                "dual_rail": bool(args.dual_rail),
                 "weights": "random-init",
             },
         }

@@ -48,6 +48,17 @@ class PipelineConfig:
     # in micro-batches; 288 GB HBM per GPU means depth is cheap).
     ring_depth: int = 4
 
+    # Dual-rail relay: split each fixed-size hop across two xGMI rails —
+    # half direct src->dst, half via an idle third GPU (src->via->dst,
+    # item-granularity store-and-forward). In a pipeline chain every
+    # non-adjacent directed link is idle, so this doubles effective hop
+    # bandwidth for the early fat boundaries that exceed one 153 GB/s
+    # link (ResNet50 layer1: 1.6 MB/img = 10.5 us/hop, the 8-stage
+    # bottleneck — profiles/README.md "Predicted pipeline scaling").
+    # Off by default pending multi-GPU validation; requires world > 2
+    # and a fixed-size wire (compression "none" or "zfp").
+    dual_rail: bool = False
+
     # --- distributed ------------------------------------------------------
     backend: str = "nccl"         # "nccl" (RCCL over xGMI) or "gloo" (CPU)
     # Whether the last stage sends results back to rank 0 (the reference's

@@ -188,3 +188,205 @@ class VarP2PRing(P2PRing):
 def make_ring(codec: Codec, depth: int) -> P2PRing:
     return VarP2PRing(codec, depth) if codec.variable \
         else P2PRing(codec, depth)
+
+
+# --------------------------------------------------------------------------
+# dual-rail relay: one hop over two xGMI rails
+# --------------------------------------------------------------------------
+
+def split_point(numel: int) -> int:
+    """Element index splitting a wire buffer into the direct half and the
+    forwarded half (64-element aligned when the buffer is big enough for
+    alignment to matter)."""
+    half = numel // 2
+    if numel >= 256:
+        half -= half % 64
+    return max(half, 1)
+
+
+def hop_via(hop: int, world: int):
+    """Forwarder rank for pipeline hop `hop` (ranks hop -> hop+1), or
+    None if the hop stays single-rail.
+
+    In a chain on the fully connected xGMI topology, route the second
+    half via hop+2 (links hop->hop+2 and hop+2->hop+1 are idle: the
+    first is a distance-2 pair, the second is the reverse direction of
+    the next hop). The last hop wraps to rank 0 — its links
+    (W-2 -> 0, 0 -> W-1) are also idle for world > 3; at world == 3 the
+    wrap would share the directed link 0->2 with hop 0's rail, so the
+    last hop stays single-rail there. This assignment uses every
+    directed link at most once across all hops (incl. the W-1 -> 0
+    result return), so each rail runs at full link rate."""
+    if hop + 2 < world:
+        return hop + 2
+    return 0 if world > 3 else None
+
+
+class DualRailRing(P2PRing):
+    """Fixed-size hop whose wire is split across two rails: bytes
+    [0, split) go direct src->dst, bytes [split, end) go src->via->dst
+    through a HopForwarder on the via rank. Store-and-forward is at item
+    granularity: it adds one pipeline slot of latency but halves the
+    steady-state traffic per link, doubling effective hop bandwidth —
+    the lever for boundaries over one xGMI link's 153 GB/s (ResNet50
+    layer1, profiles/README.md "Predicted pipeline scaling")."""
+
+    def __init__(self, codec: Codec, depth: int, via: int):
+        super().__init__(codec, depth)
+        self.via = via
+        self.split = split_point(codec.wire_numel)
+        self.works2: List[Optional[dist.Work]] = [None] * depth
+
+    def _halves(self, k: int):
+        b = self.slot(k)
+        return b[:self.split], b[self.split:]
+
+    def _wait2(self, k: int):
+        w = self.works2[k % self.depth]
+        if w is not None:
+            w.wait()
+            self.works2[k % self.depth] = None
+
+    # -- receiver side
+    def prime(self, steps: int, src: int):
+        for k in range(min(self.depth, steps)):
+            a, b = self._halves(k)
+            self.set_work(k, dist.irecv(a, src=src))
+            self.works2[k % self.depth] = dist.irecv(b, src=self.via)
+
+    def wait_recv(self, k: int, steps: int, src: int) -> torch.Tensor:
+        self.wait(k)
+        self._wait2(k)
+        return self.slot(k)
+
+    def repost(self, k: int, steps: int, src: int):
+        kk = k + self.depth
+        if kk < steps:
+            a, b = self._halves(kk)
+            self.set_work(kk, dist.irecv(a, src=src))
+            self.works2[kk % self.depth] = dist.irecv(b, src=self.via)
+
+    # -- sender side
+    def send_encoded(self, k: int, y: torch.Tensor, dst: int) -> int:
+        self.wait(k)
+        self._wait2(k)
+        wire = self.codec.encode(y, out=self.slot(k))
+        a, b = wire[:self.split], wire[self.split:]
+        self.set_work(k, dist.isend(a, dst=dst))
+        self.works2[k % self.depth] = dist.isend(b, dst=self.via)
+        return wire.numel() * dtype_bytes(wire.dtype)
+
+
+class HopForwarder:
+    """The via rank's relay for one dual-rail hop: receive the second
+    half of each item from `src`, re-send it to `dst`, in item order,
+    through a `depth`-deep buffer ring.
+
+    Two drive modes, matching the two backends' wait semantics:
+    - CUDA/RCCL: Work.wait() is a stream-level wait that never blocks
+      the host, so pump() is called inline from the run loop and simply
+      enqueues recv -> send chains, ordered by the stream dependency.
+    - gloo: Work.wait() blocks the host (and is_completed() never turns
+      true for p2p works without it), so the relay runs on its own
+      daemon thread doing plain blocking waits. The thread only touches
+      the (src -> via) and (via -> dst) pairs, which no other flow uses,
+      so it never contends with the owning rank's own hops."""
+
+    def __init__(self, half_numel: int, wire_dtype, device, depth: int,
+                 src: int, dst: int):
+        self.bufs = [torch.empty(half_numel, dtype=wire_dtype,
+                                 device=device)
+                     for _ in range(depth)]
+        self.depth = depth
+        self.src = src
+        self.dst = dst
+        self.nbytes = half_numel * dtype_bytes(wire_dtype)
+        self.inline = (torch.device(device).type == "cuda")
+        self.recv_works: List[Optional[dist.Work]] = [None] * depth
+        self.send_works: List[Optional[dist.Work]] = [None] * depth
+        self.steps = 0
+        self.posted = 0
+        self.relayed = 0
+        self._thread = None
+        self._error = None
+
+    def begin(self, steps: int):
+        assert self.relayed == self.steps, "previous run not drained"
+        self.steps = steps
+        self.posted = 0
+        self.relayed = 0
+        if self.inline:
+            self.pump()
+        else:
+            import threading
+
+            self._thread = threading.Thread(target=self._relay_all,
+                                            daemon=True)
+            self._thread.start()
+
+    # ---- gloo: blocking relay loop on the side thread
+    def _relay_all(self):
+        try:
+            for k in range(min(self.depth, self.steps)):
+                self.recv_works[k % self.depth] = dist.irecv(
+                    self.bufs[k % self.depth], src=self.src)
+                self.posted += 1
+            for k in range(self.steps):
+                i = k % self.depth
+                self.recv_works[i].wait()
+                self.recv_works[i] = None
+                self.send_works[i] = dist.isend(self.bufs[i],
+                                                dst=self.dst)
+                self.relayed += 1
+                if k + self.depth < self.steps:
+                    # slot reuse: item k must be delivered before its
+                    # buffer takes item k+depth
+                    self.send_works[i].wait()
+                    self.send_works[i] = None
+                    self.recv_works[i] = dist.irecv(self.bufs[i],
+                                                    src=self.src)
+                    self.posted += 1
+            for i, sw in enumerate(self.send_works):
+                if sw is not None:
+                    sw.wait()
+                    self.send_works[i] = None
+        except Exception as e:   # surfaced by drain()
+            self._error = e
+
+    # ---- CUDA: stream-ordered inline pump
+    def pump(self):
+        if not self.inline:
+            return
+        # enqueue relays for every item; wait() only orders streams
+        while self.relayed < self.posted:
+            i = self.relayed % self.depth
+            self.recv_works[i].wait()
+            self.recv_works[i] = None
+            self.send_works[i] = dist.isend(self.bufs[i], dst=self.dst)
+            self.relayed += 1
+        while (self.posted < self.steps
+               and self.posted - self.relayed < self.depth):
+            i = self.posted % self.depth
+            sw = self.send_works[i]
+            if sw is not None:
+                sw.wait()
+                self.send_works[i] = None
+            self.recv_works[i] = dist.irecv(self.bufs[i], src=self.src)
+            self.posted += 1
+
+    def drain(self):
+        if self.inline:
+            while self.relayed < self.steps:
+                self.pump()
+            for i, sw in enumerate(self.send_works):
+                if sw is not None:
+                    sw.wait()
+                    self.send_works[i] = None
+        else:
+            self._thread.join()
+            self._thread = None
+            if self._error is not None:
+                e, self._error = self._error, None
+                raise RuntimeError(
+                    f"dual-rail forwarder (hop {self.src}->{self.dst}) "
+                    f"failed") from e

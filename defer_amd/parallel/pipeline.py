@@ -24,7 +24,9 @@ import torch.distributed as dist
 
 from defer_amd.config import PipelineConfig
 from defer_amd.graph import GraphModel
-from defer_amd.parallel.comm import Codec, P2PRing, dtype_bytes, make_ring
+from defer_amd.parallel.comm import (Codec, DualRailRing, HopForwarder,
+                                     P2PRing, dtype_bytes, hop_via,
+                                     make_ring, split_point)
 from defer_amd.parallel.partitioner import (as_graph_model, auto_partition,
                                             partition_model)
 
@@ -41,6 +43,7 @@ class StageStats:
     compute_ms: float = 0.0   # device time, hipEvent pairs (DistPipeline)
     bytes_in: int = 0
     bytes_out: int = 0
+    bytes_fwd: int = 0        # dual-rail: bytes relayed for another hop
 
 
 class StageExecutor:
@@ -260,10 +263,18 @@ class DistPipeline:
             cuts = list(cfg.partition_layers)
             stages = partition_model(gm, cuts)
         else:
+            from defer_amd.parallel.partitioner import XGMI_LINK_GBPS
+
+            # dual-rail doubles effective hop bandwidth on every hop for
+            # world >= 4 (world == 3 leaves the last hop single-rail, so
+            # stay conservative there)
+            dual = (cfg.dual_rail and self.world >= 4
+                    and cfg.compression != "zfp+lz4")
             cuts, stages = auto_partition(
                 gm, self.world,
                 input_shape=(1,) + self.batch_shape[1:],
-                bytes_per_elem=self._wire_bytes_per_elem())
+                bytes_per_elem=self._wire_bytes_per_elem(),
+                link_gbps=XGMI_LINK_GBPS * (2.0 if dual else 1.0))
         if len(stages) != self.world:
             raise ValueError(f"{len(stages)} stages != world {self.world}")
         self.cuts = cuts
@@ -280,15 +291,41 @@ class DistPipeline:
         self.recv_ring = None
         self.send_ring = None
         self.result_ring = None
+        self.fwd = None
+        # dual-rail applies to fixed-size data hops only (zfp+lz4 is
+        # variable-size) and needs a third rank to route through
+        dual = (cfg.dual_rail and self.world > 2
+                and cfg.compression != "zfp+lz4")
         if self.world > 1:
             if self.rank > 0:
                 self.in_codec = Codec(cfg, self.in_shape, self.dtype,
                                       self.device)
-                self.recv_ring = make_ring(self.in_codec, cfg.ring_depth)
+                via = hop_via(self.rank - 1, self.world) if dual else None
+                self.recv_ring = (
+                    DualRailRing(self.in_codec, cfg.ring_depth, via)
+                    if via is not None
+                    else make_ring(self.in_codec, cfg.ring_depth))
             if self.rank < self.world - 1:
                 self.out_codec = Codec(cfg, self.out_shape, self.dtype,
                                        self.device)
-                self.send_ring = make_ring(self.out_codec, cfg.ring_depth)
+                via = hop_via(self.rank, self.world) if dual else None
+                self.send_ring = (
+                    DualRailRing(self.out_codec, cfg.ring_depth, via)
+                    if via is not None
+                    else make_ring(self.out_codec, cfg.ring_depth))
+            if dual:
+                # this rank's forwarding duty (at most one hop routes
+                # through any given rank — see comm.hop_via)
+                fwd_hop = next((i for i in range(self.world - 1)
+                                if hop_via(i, self.world) == self.rank),
+                               None)
+                if fwd_hop is not None:
+                    c = Codec(cfg, self._chain_shapes[fwd_hop],
+                              self.dtype, self.device)
+                    self.fwd = HopForwarder(
+                        c.wire_numel - split_point(c.wire_numel),
+                        c.wire_dtype, self.device, cfg.ring_depth,
+                        src=fwd_hop, dst=fwd_hop + 1)
             if cfg.return_results:
                 # logits hop last -> 0, never compressed (tiny)
                 res_cfg = PipelineConfig(compression="none")
@@ -303,6 +340,8 @@ class DistPipeline:
                                            self.dtype, self.device)
                     self.result_ring = P2PRing(self.res_codec,
                                                cfg.ring_depth)
+        if self.world > 1:
+            self._warmup_p2p(dual)
         self.stats = StageStats()
         self.timer = None
         if cfg.log_stage_stats:
@@ -317,6 +356,38 @@ class DistPipeline:
             from defer_amd.utils.trace import EventTimer
 
             self.timer = EventTimer(self.device)
+
+    def _warmup_p2p(self, dual: bool):
+        """Establish every p2p pair this pipeline uses, in one canonical
+        global order, before the overlapped run loop issues anything.
+
+        ProcessGroupNCCL creates a dedicated communicator per p2p pair
+        lazily, and the init is host-blocking and collective over the
+        pair — if ranks first-touch their pairs in inconsistent orders
+        (rank 0's first op is the result-ring irecv from the last rank
+        while rank 1 blocks on rank 0's data hop), init can cycle and
+        hang. A blocking 1-element exchange per directed pair, walked by
+        all ranks in the same sorted order, is cycle-free by induction
+        and makes every later isend/irecv hit an existing communicator.
+        (The reference's analogue is its per-channel connect handshake,
+        dispatcher.py:48,60; on gloo this is just a cheap hello.)"""
+        W = self.world
+        pairs = [(i, i + 1) for i in range(W - 1)]
+        if self.cfg.return_results:
+            pairs.append((W - 1, 0))
+        if dual:
+            for i in range(W - 1):
+                via = hop_via(i, W)
+                if via is not None:
+                    pairs.append((i, via))
+                    pairs.append((via, i + 1))
+        t = torch.zeros(1, device=self.device)
+        for src, dst in sorted(set(pairs)):
+            if self.rank == src:
+                dist.send(t, dst=dst)
+            elif self.rank == dst:
+                dist.recv(t, src=src)
+        dist.barrier()
 
     def _wire_bytes_per_elem(self):
         if self.cfg.compression.startswith("zfp"):
@@ -381,8 +452,12 @@ class DistPipeline:
             self.recv_ring.prime(steps, prv)
         if r == 0 and cfg.return_results:
             self.result_ring.prime(steps, last)
+        if self.fwd is not None:
+            self.fwd.begin(steps)
 
         for k in range(steps):
+            if self.fwd is not None:
+                self.fwd.pump()
             # ---- obtain input
             if r == 0:
                 x = feed(k)
@@ -422,6 +497,11 @@ class DistPipeline:
 
         if tm:
             self.stats.compute_ms = tm.total_ms()
+        # finish forwarding duty first: downstream ranks still wait on
+        # relayed halves of the tail items
+        if self.fwd is not None:
+            self.fwd.drain()
+            self.stats.bytes_fwd += self.fwd.relayed * self.fwd.nbytes
         # drain outstanding sends, CLEARING the slots: a consumed Work
         # must never be waited again (gloo's Work.wait() is not
         # idempotent — re-waiting one deadlocks, which hung the second
@@ -432,6 +512,10 @@ class DistPipeline:
                     if w is not None:
                         w.wait()
                         ring.works[i] = None
+                for i, w in enumerate(getattr(ring, "works2", [])):
+                    if w is not None:
+                        w.wait()
+                        ring.works2[i] = None
                 for i, w in enumerate(getattr(ring, "size_works", [])):
                     if w is not None:
                         w.wait()

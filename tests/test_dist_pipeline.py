@@ -13,7 +13,8 @@ from defer_amd.config import PipelineConfig
 from defer_amd.models import resnet50
 
 
-def _worker(rank, world, port, q, cuts, steps, compression):
+def _worker(rank, world, port, q, cuts, steps, compression,
+            dual_rail=False):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
@@ -25,7 +26,7 @@ def _worker(rank, world, port, q, cuts, steps, compression):
         cfg = PipelineConfig(device="cpu", dtype="fp32",
                              partition_layers=cuts, ring_depth=2,
                              compression=compression, zfp_rate_bits=14,
-                             backend="gloo")
+                             dual_rail=dual_rail, backend="gloo")
         B = 1
         pipe = DistPipeline(model, cfg, (B, 64, 64, 3))
 
@@ -48,13 +49,15 @@ def _worker(rank, world, port, q, cuts, steps, compression):
         dist.destroy_process_group()
 
 
-def _run(world, cuts, steps=4, compression="none", tol=0.0):
+def _run(world, cuts, steps=4, compression="none", tol=0.0,
+         dual_rail=False):
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    port = (29501 + world
+    port = (29501 + world + (17 if dual_rail else 0)
             + {"none": 0, "zfp": 7, "zfp+lz4": 11}[compression])
     procs = [ctx.Process(target=_worker,
-                         args=(r, world, port, q, cuts, steps, compression))
+                         args=(r, world, port, q, cuts, steps, compression,
+                               dual_rail))
              for r in range(world)]
     for p in procs:
         p.start()
@@ -189,3 +192,100 @@ def test_two_sequential_runs_world2():
             p.terminate()
     assert not any(alive), "second run() deadlocked"
     assert all(p.exitcode == 0 for p in procs)
+
+
+def test_dual_rail_four_stage():
+    """Dual-rail relay at world 4: every data hop splits across a direct
+    rail and a forwarder rank (hop i via rank i+2, last hop via rank 0);
+    outputs must be bitwise-identical to the single-rail pipeline."""
+    _run(4, ["add_4", "add_8", "add_12"], steps=6, dual_rail=True)
+
+
+def test_dual_rail_world3_mixed():
+    """World 3: hop 0 is dual-rail (via rank 2), hop 1 stays single-rail
+    (the wrap would collide on the 0->2 link) — mixed-mode chain."""
+    _run(3, ["add_4", "add_12"], steps=5, dual_rail=True)
+
+
+def test_dual_rail_eight_stage_auto():
+    """The headline topology with dual-rail on: 8 stages, auto cuts,
+    7 dual hops + result return."""
+    _run(8, None, steps=4, dual_rail=True)
+
+
+def test_dual_rail_with_zfp():
+    """Dual-rail over fixed-size ZFP wire (both halves of the compressed
+    buffer ride different rails)."""
+    _run(4, ["add_4", "add_8", "add_12"], steps=3, compression="zfp",
+         tol=0.05, dual_rail=True)
+
+
+def _worker_two_runs_dual(rank, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        torch.manual_seed(0)
+        from defer_amd.parallel.pipeline import DistPipeline
+
+        cfg = PipelineConfig(device="cpu", dtype="fp32", ring_depth=2,
+                             partition_layers=["add_4", "add_8",
+                                               "add_12"],
+                             dual_rail=True, backend="gloo")
+        pipe = DistPipeline(resnet50(), cfg, (1, 64, 64, 3))
+        torch.manual_seed(7)
+        xs = [torch.randn(1, 64, 64, 3) for _ in range(6)]
+        got = {}
+        # warmup-then-timed shape: forwarder must begin/drain cleanly
+        # across run() calls
+        pipe.run(2, feed=lambda k: xs[k],
+                 collect=lambda k, y: got.__setitem__(k, y.clone()))
+        dist.barrier()
+        pipe.run(4, feed=lambda k: xs[2 + k],
+                 collect=lambda k, y: got.__setitem__(2 + k, y.clone()))
+        if rank == 0:
+            assert sorted(got) == list(range(6))
+            q.put(("done", rank, None))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_two_sequential_runs_dual_rail_world4():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_worker_two_runs_dual,
+                         args=(r, 4, 29653, q)) for r in range(4)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=420)
+    alive = [p.is_alive() for p in procs]
+    for p in procs:
+        if p.is_alive():
+            p.terminate()
+    assert not any(alive), "dual-rail second run() hung"
+    assert all(p.exitcode == 0 for p in procs)
+
+
+def test_bench_torchrun_dual_rail_cpu():
+    """The round-2 GPU validation form on CPU: driver-style torchrun
+    launch of bench.py --dual-rail with 4 ranks."""
+    import json
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", "29793", os.path.join(root, "bench.py"),
+         "--gpus", "4", "--device", "cpu", "--batch", "1",
+         "--steps", "3", "--warmup", "1", "--dual-rail",
+         "--cuts", "add_4,add_8,add_12"],
+        capture_output=True, text=True, timeout=600, cwd=root)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [ln for ln in out.stdout.splitlines() if ln.strip()]
+    assert len(lines) == 1, out.stdout
+    d = json.loads(lines[0])
+    assert d["config"]["dual_rail"] is True
+    assert d["value"] > 0
