@@ -18,6 +18,10 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--model", default="resnet50")
     ap.add_argument("--stages", default="2,4,8")
+    ap.add_argument("--dual-rail", action="store_true",
+                    help="assume dual-rail hops (2x link bandwidth, "
+                         "world >= 4; matches DistPipeline's cut "
+                         "choice with cfg.dual_rail)")
     args = ap.parse_args()
 
     from defer_amd.models import MODELS
@@ -34,13 +38,15 @@ def main():
     print(f"{args.model}: total predicted compute {total:.1f} us/img "
           f"({1e6 / total:.0f} img/s/GPU equivalent)")
     for ns in (int(s) for s in args.stages.split(",")):
-        cuts, _ = auto_partition(g, ns)
+        mult = 2.0 if args.dual_rail and ns >= 4 else 1.0
+        cuts, _ = auto_partition(g, ns,
+                                 link_gbps=XGMI_LINK_GBPS * mult)
         bounds = sorted(pos[c] for c in cuts) + [len(names) - 1]
         st, start = [], 0
         for e in bounds:
             st.append(sum(t[start:e + 1]))
             start = e + 1
-        hops = [ob[c] / (XGMI_LINK_GBPS * 1e3) for c in cuts]
+        hops = [ob[c] / (XGMI_LINK_GBPS * mult * 1e3) for c in cuts]
         bot = max(st + hops)
         kind = "compute" if bot in st else "xGMI hop"
         print(f"pp{ns}: cuts={cuts}")
