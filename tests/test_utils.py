@@ -199,3 +199,38 @@ def test_powerbench_parses_rocm_smi_json():
     done.stdout = fake
     with mock.patch.object(pb.subprocess, "run", return_value=done):
         assert abs(pb.read_power_w() - 400.5) < 1e-9
+
+
+def test_dual_rail_link_assignment_is_collision_free():
+    """comm.hop_via's claim, checked exhaustively for world 3..16: across
+    all hops' rails (direct link + the two forwarder links) plus the
+    result-return link, every DIRECTED xGMI link is used at most once —
+    each rail therefore runs at full link rate."""
+    from defer_amd.parallel.comm import hop_via
+
+    for world in range(3, 17):
+        links = []
+        for i in range(world - 1):
+            links.append((i, i + 1))          # direct rail
+            via = hop_via(i, world)
+            if via is not None:
+                assert via not in (i, i + 1)
+                links.append((i, via))        # src -> forwarder
+                links.append((via, i + 1))    # forwarder -> dst
+        links.append((world - 1, 0))          # result return
+        assert len(links) == len(set(links)), (world, links)
+        # every hop is dual-rail except the last one at world == 3
+        n_single = sum(1 for i in range(world - 1)
+                       if hop_via(i, world) is None)
+        assert n_single == (1 if world == 3 else 0)
+
+
+def test_dual_rail_split_point():
+    from defer_amd.parallel.comm import split_point
+
+    assert split_point(2) == 1
+    assert split_point(100) == 50
+    assert split_point(1 << 20) == (1 << 19)
+    n = 1000003
+    s = split_point(n)
+    assert 0 < s < n and s % 64 == 0
