@@ -39,6 +39,7 @@ def main():
     ap.add_argument("--compression", default="none",
                     choices=["none", "zfp", "zfp+lz4"])
     ap.add_argument("--device", default="cuda", choices=["cuda", "cpu"])
+    ap.add_argument("--dual-rail", action="store_true")
     args = ap.parse_args()
 
     from defer_amd.config import PipelineConfig
@@ -69,7 +70,7 @@ def main():
         partition_layers=cuts, num_stages=world, device=args.device,
         dtype="bf16" if args.device == "cuda" else "fp32",
         batch_size=args.batch, use_hip_graphs=False,
-        compression=args.compression,
+        compression=args.compression, dual_rail=args.dual_rail,
         backend=backend, return_results=True)
     dev = (torch.device("cuda", local_rank) if args.device == "cuda"
            else torch.device("cpu"))
