@@ -234,3 +234,48 @@ def test_dual_rail_split_point():
     n = 1000003
     s = split_point(n)
     assert 0 < s < n and s % 64 == 0
+
+
+def test_http_serving_example():
+    """examples/serve_http.py: REST front-end over the DEFER pipeline
+    (2 CPU stages) answers concurrent requests with the same outputs as
+    the bare model."""
+    import importlib.util
+    import os
+    import sys
+    from concurrent.futures import ThreadPoolExecutor
+
+    from fastapi.testclient import TestClient
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    spec = importlib.util.spec_from_file_location(
+        "serve_http", os.path.join(root, "examples", "serve_http.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+
+    torch.manual_seed(0)
+    app = mod.build_app(["cpu", "cpu"], input_hw=64)
+    from defer_amd.models import resnet50
+    torch.manual_seed(0)
+    ref = resnet50()
+
+    client = TestClient(app)
+    torch.manual_seed(11)
+    xs = [torch.randn(1, 64, 64, 3) for _ in range(4)]
+
+    def post(x):
+        r = client.post("/infer", json={"data": x.flatten().tolist(),
+                                        "shape": list(x.shape)})
+        assert r.status_code == 200, r.text
+        return torch.tensor(r.json()["probs"])
+
+    with ThreadPoolExecutor(4) as pool:
+        outs = list(pool.map(post, xs))
+    with torch.no_grad():
+        for x, y in zip(xs, outs):
+            want = ref(x)
+            assert torch.allclose(y, want, atol=1e-4), \
+                (y - want).abs().max()
+
+    bad = client.post("/infer", json={"data": [1.0], "shape": [1]})
+    assert bad.status_code == 400
