@@ -60,6 +60,10 @@ def build_app(devices, model_name="resnet50", input_hw=224, cuts=None):
 
     app = FastAPI(title="defer_amd")
 
+    @app.on_event("shutdown")
+    def _shutdown():
+        in_q.put(None)          # drains the pipeline threads cleanly
+
     @app.post("/infer")
     def infer(req: Request):
         try:
