@@ -168,7 +168,7 @@ def main():
         print(f"[stage {pipe.rank}] items={st.items} images={st.images} "
               f"compute={cms:.3f}ms/item (dev busy ~{busy:.0f}%) "
               f"wire_in={st.bytes_in/1e6:.1f}MB wire_out="
-              f"{st.bytes_out/1e6:.1f}MB "
+              f"{st.bytes_out/1e6:.1f}MB fwd={st.bytes_fwd/1e6:.1f}MB "
               f"({st.bytes_out/max(st.items,1)/1e6:.2f} MB/item)",
               file=sys.stderr)
     if pipe.rank == 0:
