@@ -35,7 +35,7 @@ def parse_args():
                          "batch table; the reference streams batch-1 "
                          "items, test/test.py:20-23)")
     ap.add_argument("--compression", default="none",
-                    choices=["none", "zfp", "zfp+lz4"])
+                    choices=["none", "fp8", "zfp", "zfp+lz4"])
     ap.add_argument("--zfp-bits", type=int, default=8)
     ap.add_argument("--cuts", default="auto",
                     help='"auto", "defer8", or comma-separated layer names')

@@ -39,8 +39,9 @@ class PipelineConfig:
     # --- inter-stage relay (the data plane) -------------------------------
     # Codec for boundary activations. The reference compresses with
     # lz4(zfp(x)) (dispatcher.py:81-84). "none" ships raw bf16 over xGMI;
-    # "zfp" ships fixed-rate ZFP blocks; "zfp+lz4" adds the LZ4 stage.
-    compression: str = "none"     # "none" | "zfp" | "zfp+lz4"
+    # "fp8" ships a cast-only e4m3 wire (1 B/value, ~free); "zfp" ships
+    # fixed-rate ZFP blocks; "zfp+lz4" adds the LZ4 stage.
+    compression: str = "none"     # "none" | "fp8" | "zfp" | "zfp+lz4"
     zfp_rate_bits: int = 8        # fixed-rate bits per value (ZFP)
 
     # Depth of the per-stage device-resident activation ring buffers (the

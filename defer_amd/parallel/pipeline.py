@@ -392,6 +392,8 @@ class DistPipeline:
     def _wire_bytes_per_elem(self):
         if self.cfg.compression.startswith("zfp"):
             return self.cfg.zfp_rate_bits / 8.0
+        if self.cfg.compression == "fp8":
+            return 1.0
         return 2.0 if self.cfg.dtype == "bf16" else 4.0
 
     def _boundary_shapes(self, stages):

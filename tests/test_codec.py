@@ -113,3 +113,41 @@ def test_spec_properties_random_shapes():
         assert prev <= max(1e-6, 0.05 * float(np.abs(a).max()))
 
     check()
+
+
+def test_fp8_wire_codec_roundtrip():
+    """comm.Codec "fp8": fixed-size uint8 wire, decode(encode(x)) within
+    e4m3 quantization error for activation-range data."""
+    import torch
+
+    from defer_amd.config import PipelineConfig
+    from defer_amd.parallel.comm import Codec
+
+    shape = (2, 7, 7, 32)
+    cfg = PipelineConfig(compression="fp8")
+    c = Codec(cfg, shape, torch.float32, "cpu")
+    assert c.wire_numel == 2 * 7 * 7 * 32 + 4   # +4: per-tensor scale
+    assert c.wire_dtype == torch.uint8 and not c.variable
+    torch.manual_seed(3)
+    x = torch.randn(*shape).clamp(-8, 8)
+    out = c.alloc_wire()
+    wire = c.encode(x, out=out)
+    assert wire.dtype == torch.uint8
+    y = c.decode(wire)
+    assert y.shape == x.shape
+    mask = x.abs() > 0.05          # skip the subnormal tail
+    rel = ((y - x).abs() / x.abs().clamp_min(1e-6))[mask]
+    assert rel.max() <= 0.0667, rel.max()   # e4m3: 3 mantissa bits
+
+    # amax scaling: values past e4m3's 448 max must survive (the
+    # unscaled cast overflows them to NaN)
+    big = torch.randn(*shape) * 500
+    yb = c.decode(c.encode(big, out=c.alloc_wire()))
+    assert torch.isfinite(yb).all()
+    mask = big.abs() > 1.0
+    rel = ((yb - big).abs() / big.abs())[mask]
+    assert rel.max() <= 0.0667, rel.max()
+
+    # all-zero input round-trips to zeros
+    z = c.decode(c.encode(torch.zeros(*shape), out=c.alloc_wire()))
+    assert torch.equal(z, torch.zeros(*shape))
