@@ -151,3 +151,56 @@ def test_fp8_wire_codec_roundtrip():
     # all-zero input round-trips to zeros
     z = c.decode(c.encode(torch.zeros(*shape), out=c.alloc_wire()))
     assert torch.equal(z, torch.zeros(*shape))
+
+
+def test_codec_sizing_on_all_real_boundary_shapes():
+    """Every codec mode must size a wire for every boundary shape the
+    four models actually produce under auto/defer8 cuts (the zfp block
+    math and the lz4 worst-case bound run on real NHWC shapes, incl.
+    odd spatial sizes) — a sizing error here would mismatch RCCL
+    send/recv sizes and hang the round-end multi-GPU runs."""
+    import torch
+
+    from defer_amd.config import PipelineConfig
+    from defer_amd.models import DEFER_8STAGE_CUTS, MODELS
+    from defer_amd.parallel.comm import Codec
+    from defer_amd.parallel.partitioner import (auto_partition,
+                                                partition_model)
+
+    B = 3
+    for name, ctor in MODELS.items():
+        model = ctor()
+        cut_sets = [auto_partition(model, 4)[0]]
+        if name == "resnet50":
+            cut_sets.append(DEFER_8STAGE_CUTS)
+        for cuts in cut_sets:
+            stages = partition_model(model, cuts)
+            # chain shapes via batch-1 trace
+            shape = (1, 224, 224, 3)
+            outs = []
+            for s in stages:
+                x = torch.zeros(*shape)
+                with torch.no_grad():
+                    y = s.graph.forward(x)
+                outs.append((B,) + tuple(y.shape[1:]))
+                shape = tuple(y.shape)
+            for mode in ("none", "fp8", "zfp", "zfp+lz4"):
+                cfg = PipelineConfig(compression=mode)
+                for i in range(len(stages) - 1):
+                    c = Codec(cfg, outs[i], torch.bfloat16, "cpu")
+                    numel = 1
+                    for d in outs[i]:
+                        numel *= d
+                    assert c.wire_numel > 0
+                    if mode == "none":
+                        assert c.wire_numel == numel
+                        assert c.wire_dtype == torch.bfloat16
+                    elif mode == "fp8":
+                        assert c.wire_numel == numel + 4
+                    elif mode == "zfp":
+                        # fixed-rate: rate bits per value, padded to
+                        # whole 4x4x4 blocks
+                        assert c.wire_numel >= numel * cfg.zfp_rate_bits // 8
+                    else:
+                        assert c.variable
+                        assert c.wire_numel > numel * cfg.zfp_rate_bits // 8
