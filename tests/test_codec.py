@@ -204,3 +204,37 @@ def test_codec_sizing_on_all_real_boundary_shapes():
                     else:
                         assert c.variable
                         assert c.wire_numel > numel * cfg.zfp_rate_bits // 8
+
+
+def test_fp8_wire_codec_property_sweep():
+    """fp8 codec across random shapes and dynamic ranges: finite
+    output, e4m3 relative error bound above the scaled-subnormal floor,
+    exact zeros."""
+    import torch
+    from hypothesis import given, settings, strategies as st
+
+    from defer_amd.config import PipelineConfig
+    from defer_amd.parallel.comm import Codec
+
+    @settings(max_examples=40, deadline=None)
+    @given(st.integers(1, 4), st.integers(1, 9), st.integers(1, 9),
+           st.integers(1, 19),
+           st.floats(min_value=-3.0, max_value=3.0),
+           st.integers(0, 2**31 - 1))
+    def check(b, h, w, c, log_scale, seed):
+        shape = (b, h, w, c)
+        torch.manual_seed(seed)
+        x = torch.randn(*shape) * (10.0 ** log_scale)
+        codec = Codec(PipelineConfig(compression="fp8"), shape,
+                      torch.float32, "cpu")
+        y = codec.decode(codec.encode(x, out=codec.alloc_wire()))
+        assert y.shape == x.shape
+        assert torch.isfinite(y).all()
+        amax = x.abs().max()
+        mask = x.abs() > amax * 2 ** -6   # above the scaled denormal floor
+        if mask.any():
+            rel = ((y - x).abs() / x.abs())[mask]
+            assert rel.max() <= 0.0667, float(rel.max())
+        assert torch.equal(y == 0, x == 0) or (x == 0).sum() == 0
+
+    check()
