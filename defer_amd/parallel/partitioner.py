@@ -119,8 +119,24 @@ def node_times(graph: LayerGraph, input_shape=(1, 224, 224, 3),
                 t = max(f / _EFF_3X3, (in_b + out_b) / _BW_STREAM)
             else:
                 t = max(f / _EFF_1X1, (in_b + out_b) / _BW_STREAM)
+        elif isinstance(lay, nn.Conv2d):
+            # FX-imported models carry plain torch convs (NCHW):
+            # out = (B, C, H, W)
+            kh, kw = lay.kernel_size
+            f = (2.0 * float(out[0] * out[2] * out[3])
+                 * kh * kw * (lay.in_channels // lay.groups)
+                 * lay.out_channels)
+            if lay.in_channels < 64:
+                t = max(f / _EFF_STEM, out_b / _BW_STEM)
+            elif max(kh, kw) >= 3:
+                t = max(f / _EFF_3X3, (in_b + out_b) / _BW_STREAM)
+            else:
+                t = max(f / _EFF_1X1, (in_b + out_b) / _BW_STREAM)
         elif isinstance(lay, L.Dense):
             f = 2.0 * out[0] * lay.cin * lay.cout
+            t = max(f / _EFF_1X1, (in_b + out_b) / _BW_STREAM)
+        elif isinstance(lay, nn.Linear):
+            f = 2.0 * float(out[0]) * lay.in_features * lay.out_features
             t = max(f / _EFF_1X1, (in_b + out_b) / _BW_STREAM)
         else:
             f = 2.0 * float(torch.tensor(out).prod())
