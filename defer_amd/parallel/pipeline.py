@@ -296,6 +296,13 @@ class DistPipeline:
         # variable-size) and needs a third rank to route through
         dual = (cfg.dual_rail and self.world > 2
                 and cfg.compression != "zfp+lz4")
+        if cfg.dual_rail and not dual and self.rank == 0:
+            import sys
+
+            print("defer_amd: dual_rail ignored "
+                  + ("(variable-size zfp+lz4 wire)"
+                     if cfg.compression == "zfp+lz4"
+                     else "(needs world > 2)"), file=sys.stderr)
         if self.world > 1:
             if self.rank > 0:
                 self.in_codec = Codec(cfg, self.in_shape, self.dtype,
