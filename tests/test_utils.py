@@ -279,3 +279,20 @@ def test_http_serving_example():
 
     bad = client.post("/infer", json={"data": [1.0], "shape": [1]})
     assert bad.status_code == 400
+
+
+def test_stream_infer_example_runs():
+    """examples/stream_infer.py (the reference's test/test.py usage
+    shape) runs end to end on CPU."""
+    import os
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, "examples",
+                                      "stream_infer.py"),
+         "--device", "cpu", "--items", "3", "--batch", "1"],
+        capture_output=True, text=True, timeout=300, cwd=root)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "images/sec" in out.stdout
