@@ -60,6 +60,15 @@ class PipelineConfig:
     # and a fixed-size wire (compression "none" or "zfp").
     dual_rail: bool = False
 
+    # Directory of per-stage checkpoints written by
+    # checkpoint.save_stages (part{i}.pt + manifest.json). Each rank
+    # loads ONLY its own stage's weights — the reference's per-node
+    # weight shipping (dispatcher.py:57, node.py:53-75) without moving
+    # the full model to every GPU. When set and partition_layers is
+    # None, the manifest's cut list is used. None -> random-init (the
+    # BASELINE.json benchmark path).
+    weights_dir: Optional[str] = None
+
     # --- distributed ------------------------------------------------------
     backend: str = "nccl"         # "nccl" (RCCL over xGMI) or "gloo" (CPU)
     # Whether the last stage sends results back to rank 0 (the reference's

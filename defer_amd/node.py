@@ -40,6 +40,9 @@ def main():
                     choices=["none", "fp8", "zfp", "zfp+lz4"])
     ap.add_argument("--device", default="cuda", choices=["cuda", "cpu"])
     ap.add_argument("--dual-rail", action="store_true")
+    ap.add_argument("--weights-dir", default=None,
+                    help="per-stage checkpoint dir (checkpoint."
+                         "save_stages); cuts come from its manifest")
     args = ap.parse_args()
 
     from defer_amd.config import PipelineConfig
@@ -71,6 +74,7 @@ def main():
         dtype="bf16" if args.device == "cuda" else "fp32",
         batch_size=args.batch, use_hip_graphs=False,
         compression=args.compression, dual_rail=args.dual_rail,
+        weights_dir=args.weights_dir,
         backend=backend, return_results=True)
     dev = (torch.device("cuda", local_rank) if args.device == "cuda"
            else torch.device("cpu"))

@@ -257,6 +257,17 @@ class DistPipeline:
         self.batch_shape = tuple(batch_shape)
 
         gm = as_graph_model(model)
+        if cfg.weights_dir and cfg.partition_layers is None:
+            # partition exactly as the checkpoint was partitioned
+            from defer_amd import checkpoint
+
+            mani = checkpoint.load_manifest(cfg.weights_dir)
+            if mani["num_stages"] != self.world:
+                raise ValueError(
+                    f"checkpoint has {mani['num_stages']} stages, "
+                    f"world is {self.world}")
+            if self.world > 1:
+                cfg.partition_layers = list(mani["cut_points"])
         if self.world == 1:
             cuts, stages = [], [gm]
         elif cfg.partition_layers is not None:
@@ -285,6 +296,13 @@ class DistPipeline:
                            cut_points=cuts)
         # shape-trace on CPU BEFORE weights move to the GPU
         self.in_shape, self.out_shape = self._boundary_shapes(stages)
+        if cfg.weights_dir:
+            # per-rank stage weights only (reference: per-node weight
+            # shipping, dispatcher.py:57) — before fusion/pre-cast
+            from defer_amd import checkpoint
+
+            checkpoint.load_stage(stages[self.rank], cfg.weights_dir,
+                                  self.rank)
         self.stage = StageExecutor(stages[self.rank], self.device,
                                    self.dtype, cfg.use_hip_graphs)
 
