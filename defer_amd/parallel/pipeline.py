@@ -164,6 +164,11 @@ class DEFER:
 
             dump_partition(stages, cfg.partition_dump_dir,
                            cut_points=partition_layers)
+        if cfg.weights_dir:
+            from defer_amd import checkpoint
+
+            for i, s in enumerate(stages):
+                checkpoint.load_stage(s, cfg.weights_dir, i)
         dtype = cfg.torch_dtype()
         execs = [StageExecutor(s, dev, dtype, cfg.use_hip_graphs)
                  for s, dev in zip(stages, self.computeNodes)]

@@ -157,3 +157,36 @@ def test_weights_dir_world2_manifest_cuts(tmp_path):
     assert all(p.exitcode == 0 for p in procs)
     tag, err = q.get()
     assert tag == "err" and err <= 1e-5, err
+
+
+def test_defer_orchestrator_loads_checkpoints(tmp_path):
+    """The threaded DEFER orchestrator honors weights_dir too: a fresh
+    model on two CPU nodes reproduces the checkpointed model."""
+    import queue
+    import threading
+
+    from defer_amd import DEFER, PipelineConfig, checkpoint
+    from defer_amd.models import resnet50
+
+    torch.manual_seed(0)
+    src = resnet50()
+    cuts = ["add_8"]
+    checkpoint.save_stages(src, cuts, str(tmp_path))
+
+    torch.manual_seed(77)
+    fresh = resnet50()
+    eng = DEFER(["cpu", "cpu"], config=PipelineConfig(
+        device="cpu", dtype="fp32", weights_dir=str(tmp_path)))
+    in_q, out_q = queue.Queue(4), queue.Queue(4)
+    t = threading.Thread(target=eng.run_defer,
+                         args=(fresh, cuts, in_q, out_q))
+    t.start()
+    torch.manual_seed(5)
+    x = torch.randn(1, 64, 64, 3)
+    in_q.put(x)
+    in_q.put(None)
+    got = out_q.get(timeout=120)
+    t.join(timeout=120)
+    with torch.no_grad():
+        want = src(x)
+    assert torch.allclose(got, want, atol=1e-5)
