@@ -24,7 +24,8 @@ import torch
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 
-def build_app(devices, model_name="resnet50", input_hw=224, cuts=None):
+def build_app(devices, model_name="resnet50", input_hw=224, cuts=None,
+              weights_dir=None):
     from fastapi import FastAPI, HTTPException
     from pydantic import BaseModel
 
@@ -35,7 +36,8 @@ def build_app(devices, model_name="resnet50", input_hw=224, cuts=None):
     cfg = PipelineConfig(
         device="cuda" if dev0.startswith("cuda") else "cpu",
         dtype="bf16" if dev0.startswith("cuda") else "fp32",
-        input_shape=(1, input_hw, input_hw, 3))
+        input_shape=(1, input_hw, input_hw, 3),
+        weights_dir=weights_dir)
     engine = DEFER(devices, config=cfg)
     in_q, out_q = queue.Queue(64), queue.Queue(64)
     model = MODELS[model_name]()
@@ -92,8 +94,12 @@ def main():
     ap.add_argument("--model", default="resnet50")
     ap.add_argument("--port", type=int, default=8731)
     ap.add_argument("--input-hw", type=int, default=224)
+    ap.add_argument("--weights-dir", default=None,
+                    help="per-stage checkpoint dir (random-init if "
+                         "unset)")
     args = ap.parse_args()
-    app = build_app(args.devices.split(","), args.model, args.input_hw)
+    app = build_app(args.devices.split(","), args.model, args.input_hw,
+                    weights_dir=args.weights_dir)
     uvicorn.run(app, host="127.0.0.1", port=args.port)
 
 
