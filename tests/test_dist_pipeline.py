@@ -174,6 +174,18 @@ def test_bench_torchrun_two_ranks_cpu():
     d = json.loads(lines[0])
     assert d["n_gpus"] == 2 and d["config"]["parallelism"] == "pp2"
     assert d["value"] > 0
+    # full driver-contract schema: every required field, right types
+    schema = {"metric": str, "value": float, "unit": str, "n_gpus": int,
+              "steps": int, "warmup": int, "ms_per_step": float,
+              "higher_is_better": bool, "scaling": str,
+              "dtype": str, "data": str, "config": dict}
+    for key, typ in schema.items():
+        assert key in d and isinstance(d[key], typ), (key, d.get(key))
+    assert "vs_baseline" in d            # null allowed
+    assert d["scaling"] in ("weak", "strong")
+    assert d["data"] == "synthetic" and d["higher_is_better"] is True
+    for key in ("model", "global_batch", "parallelism"):
+        assert key in d["config"], key
 
 
 def _worker_two_runs(rank, world, port, q):
