@@ -83,6 +83,19 @@ class PipelineConfig:
 
     extra: dict = field(default_factory=dict)
 
+    def __post_init__(self):
+        if self.compression not in ("none", "fp8", "zfp", "zfp+lz4"):
+            raise ValueError(
+                f"unknown compression {self.compression!r} "
+                "(none|fp8|zfp|zfp+lz4)")
+        if self.dtype not in ("bf16", "fp16", "fp32"):
+            raise ValueError(f"unknown dtype {self.dtype!r} "
+                             "(bf16|fp16|fp32)")
+        if self.ring_depth < 1:
+            raise ValueError("ring_depth must be >= 1")
+        if not 1 <= self.zfp_rate_bits <= 30:
+            raise ValueError("zfp_rate_bits must be in [1, 30]")
+
     def torch_dtype(self):
         import torch
 
