@@ -296,3 +296,19 @@ def test_stream_infer_example_runs():
         capture_output=True, text=True, timeout=300, cwd=root)
     assert out.returncode == 0, out.stderr[-2000:]
     assert "images/sec" in out.stdout
+
+
+def test_pipeline_config_validation():
+    import pytest
+
+    from defer_amd.config import PipelineConfig
+
+    PipelineConfig()                      # defaults valid
+    with pytest.raises(ValueError, match="compression"):
+        PipelineConfig(compression="gzip")
+    with pytest.raises(ValueError, match="dtype"):
+        PipelineConfig(dtype="fp64")
+    with pytest.raises(ValueError, match="ring_depth"):
+        PipelineConfig(ring_depth=0)
+    with pytest.raises(ValueError, match="zfp_rate_bits"):
+        PipelineConfig(zfp_rate_bits=40)
