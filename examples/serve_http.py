@@ -60,11 +60,14 @@ def build_app(devices, model_name="resnet50", input_hw=224, cuts=None,
         data: list
         shape: list
 
-    app = FastAPI(title="defer_amd")
+    from contextlib import asynccontextmanager
 
-    @app.on_event("shutdown")
-    def _shutdown():
+    @asynccontextmanager
+    async def lifespan(_app):
+        yield
         in_q.put(None)          # drains the pipeline threads cleanly
+
+    app = FastAPI(title="defer_amd", lifespan=lifespan)
 
     @app.post("/infer")
     def infer(req: Request):
