@@ -21,7 +21,7 @@ def zfp_encode(x: torch.Tensor, rate: int, out=None) -> torch.Tensor:
             raise RuntimeError("HIP extension missing for zfp_encode")
         return m.zfp_encode(x.contiguous(), rate, out)
     w = zfp_ref.encode(x.float().numpy(), rate)
-    t = torch.from_numpy(np.ascontiguousarray(w))
+    t = torch.from_numpy(np.ascontiguousarray(w).copy())
     if out is not None:
         out.copy_(t)
         return out
