@@ -122,12 +122,18 @@ def main():
               for _ in range(n_inputs)] if pipe.rank == 0 else None
 
     def feed(k):
+        feed_t[k] = time.perf_counter()
         return inputs[k % n_inputs]
 
     sink = {}
+    feed_t = {}
+    lat_ms = []
 
     def collect(k, y):
         sink["last"] = (k, y.shape)
+        t = feed_t.pop(k, None)
+        if t is not None:
+            lat_ms.append((time.perf_counter() - t) * 1e3)
 
     import torch.distributed as dist
 
@@ -162,6 +168,16 @@ def main():
 
     if args.stats:
         import sys
+        if pipe.rank == 0 and lat_ms:
+            srt = sorted(lat_ms[-args.steps:])
+
+            def pct(p):
+                return srt[min(len(srt) - 1, int(p * len(srt)))]
+
+            print(f"[latency] item p50={pct(0.5):.2f}ms "
+                  f"p90={pct(0.9):.2f}ms p99={pct(0.99):.2f}ms "
+                  f"(feed->collect, includes {world}-deep pipeline "
+                  f"occupancy)", file=sys.stderr)
         st = pipe.stats
         cms = st.compute_ms / max(st.items, 1)
         busy = st.compute_ms / 1e3 / el * 100 if el > 0 else 0.0
