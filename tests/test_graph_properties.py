@@ -234,3 +234,14 @@ def test_from_torch_random_dags_roundtrip_and_split(model_spec):
             for s in g.split([mid]):
                 z = s.forward(z)
         assert torch.equal(z, want)
+
+        # auto_partition must also handle the FX graph (cost model over
+        # plain convs + cat/add/relu function layers)
+        from defer_amd.parallel.partitioner import auto_partition
+
+        _, stages = auto_partition(g, 2, input_shape=(1, 3, 6, 6))
+        with torch.no_grad():
+            z = x
+            for s in stages:
+                z = s(z)
+        assert torch.equal(z, want)
