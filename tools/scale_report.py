@@ -11,16 +11,35 @@ import json
 import sys
 
 
+def _load(text):
+    """Accept a whole-file JSON value (object, array of run objects, or
+    an object wrapping a list of them) or JSONL bench lines."""
+    try:
+        v = json.loads(text)
+    except ValueError:
+        return [json.loads(ln) for ln in text.splitlines()
+                if ln.strip().startswith("{")]
+    if isinstance(v, list):
+        return [r for r in v if isinstance(r, dict) and "n_gpus" in r]
+    if isinstance(v, dict):
+        if "n_gpus" in v:
+            return [v]
+        for val in v.values():          # e.g. {"runs": [...]}
+            if (isinstance(val, list) and val
+                    and isinstance(val[0], dict) and "n_gpus" in val[0]):
+                return val
+    return []
+
+
 def main(argv):
-    lines = []
+    runs = []
     if len(argv) > 1:
         for p in argv[1:]:
             with open(p) as f:
-                lines += [ln for ln in f if ln.strip().startswith("{")]
+                runs += _load(f.read())
     else:
-        lines = [ln for ln in sys.stdin if ln.strip().startswith("{")]
-    runs = sorted((json.loads(ln) for ln in lines),
-                  key=lambda d: d["n_gpus"])
+        runs = _load(sys.stdin.read())
+    runs = sorted(runs, key=lambda d: d["n_gpus"])
     if not runs:
         print("no bench JSON lines found", file=sys.stderr)
         return 1
