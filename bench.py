@@ -4,13 +4,22 @@
 (/root/reference/test/test.py:25-37: results counted over a window) on
 MI355X at 1/2/4/8 stages.
 
-Single GPU:   python bench.py --gpus 1 --steps 50 --warmup 10
-Multi GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N
+Single GPU:   python bench.py --gpus 1 --steps 200 --warmup 20
+Multi GPU:    python bench.py --gpus N ...          (self-spawns N ranks)
+         or:  python -m torch.distributed.run --nnodes=1 --nproc-per-node N
                 --master-addr 127.0.0.1 bench.py --gpus N ...
 
 Rank r runs pipeline stage r; boundary activations relay over RCCL/xGMI
 (optionally ZFP-compressed). Synthetic 224x224x3 data, random-init
 weights (no network access for datasets), bf16 compute.
+
+A "step" processes one global batch (--batch images). At world > 1 the
+global batch is split into --micro-batch-sized pipeline items so the
+chain stays full (the reference streams single images through its node
+chain, test/test.py:20-23,47-49; micro-batches are the throughput-viable
+version of that operating point). After the timed region a short
+serialized pass measures true per-item end-to-end latency (feed at rank0
+-> result back at rank0, host-synchronized), reported as latency_ms.
 """
 
 import argparse
@@ -21,19 +30,25 @@ import time
 import torch
 
 
-def parse_args():
+def parse_args(argv=None):
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=64)
-    ap.add_argument("--warmup", type=int, default=16)
+    ap.add_argument("--steps", type=int, default=400,
+                    help="timed steps; one step = one --batch global batch "
+                         "(default sized so the timed region is ~2.5 s at "
+                         "1 GPU and longer at N>1: rocm-smi sampling can "
+                         "see the GPU busy)")
+    ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--model", default="resnet50",
                     choices=["resnet50", "resnet101", "resnet152",
-                             "vgg19"])
+                             "vgg19", "vgg19_gap"])
     ap.add_argument("--batch", type=int, default=256,
-                    help="images per pipeline micro-batch (throughput "
-                         "saturates ~128-256; see profiles/README.md "
-                         "batch table; the reference streams batch-1 "
-                         "items, test/test.py:20-23)")
+                    help="images per step (global batch; throughput "
+                         "saturates ~128-256, profiles/README.md)")
+    ap.add_argument("--micro-batch", type=int, default=0,
+                    help="images per pipeline item; 0 = auto (=batch at "
+                         "1 stage; 64 at >1 stage so 20 driver steps are "
+                         ">=80 items and pipeline fill/drain stays <10%%)")
     ap.add_argument("--compression", default="none",
                     choices=["none", "fp8", "zfp", "zfp+lz4"])
     ap.add_argument("--zfp-bits", type=int, default=8)
@@ -49,17 +64,23 @@ def parse_args():
                          "doubles effective hop bandwidth for "
                          "boundaries above one link's 153 GB/s")
     ap.add_argument("--no-return-results", action="store_true")
+    ap.add_argument("--latency-items", type=int, default=32,
+                    help="serialized items for the post-run latency "
+                         "measurement (0 = skip)")
+    ap.add_argument("--calibration", default=None, metavar="JSON",
+                    help="per-layer measured-cost file for auto cuts "
+                         "(tools/calibrate.py); default: use the "
+                         "in-tree profile for this model if present")
     ap.add_argument("--stats", action="store_true",
                     help="print per-stage stats to stderr (hipEvent "
                          "compute time, wire bytes)")
     ap.add_argument("--dump-partition", default=None, metavar="DIR",
                     help="write per-stage DOT/text partition dumps "
                          "(plot_model parity, reference node.py:39)")
-    return ap.parse_args()
+    return ap.parse_args(argv)
 
 
-def main():
-    args = parse_args()
+def worker(args):
     # ONE-JSON-LINE stdout contract: native libs write banners straight
     # to fd 1 (gloo's rank banner; RCCL's version line under NCCL_DEBUG),
     # so park the real stdout and point fd 1 at stderr for the whole
@@ -73,15 +94,19 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if world != args.gpus:
+        raise SystemExit(
+            f"bench.py: WORLD_SIZE={world} but --gpus {args.gpus}; launch "
+            f"with matching -nproc-per-node or let bench.py self-spawn")
     dist_mode = world > 1
+    import torch.distributed as dist
     if dist_mode:
-        import torch.distributed as dist
         backend = "nccl" if args.device == "cuda" else "gloo"
         if args.device == "cuda":
-            torch.cuda.set_device(local_rank)
+            ndev = torch.cuda.device_count()
+            torch.cuda.set_device(local_rank % ndev)
         dist.init_process_group(backend)
     else:
-        import torch.distributed as dist
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29771")
         dist.init_process_group(
@@ -98,44 +123,45 @@ def main():
     else:
         cuts = [c for c in args.cuts.split(",") if c]
 
+    B = args.batch
+    M = args.micro_batch
+    if M <= 0:
+        M = B if world == 1 else min(64, B)
+    if B % M:
+        raise SystemExit(f"--batch {B} not divisible by --micro-batch {M}")
+    per_step = B // M
+
     cfg = PipelineConfig(
         partition_layers=cuts, num_stages=world, device=args.device,
         dtype="bf16" if args.device == "cuda" else "fp32",
-        batch_size=args.batch, use_hip_graphs=args.graphs,
+        batch_size=M, use_hip_graphs=args.graphs,
         compression=args.compression, zfp_rate_bits=args.zfp_bits,
         ring_depth=args.ring_depth, dual_rail=args.dual_rail,
         backend="nccl" if args.device == "cuda" else "gloo",
         return_results=not args.no_return_results,
         log_stage_stats=args.stats,
+        calibration_file=args.calibration,
         partition_dump_dir=args.dump_partition)
 
-    dev = (torch.device("cuda", local_rank) if args.device == "cuda"
-           else torch.device("cpu"))
-    B = args.batch
-    pipe = DistPipeline(model, cfg, (B, 224, 224, 3), device=dev)
+    dev = (torch.device("cuda",
+                        local_rank % max(torch.cuda.device_count(), 1))
+           if args.device == "cuda" else torch.device("cpu"))
+    pipe = DistPipeline(model, cfg, (M, 224, 224, 3), device=dev)
 
     dtype = torch.bfloat16 if args.device == "cuda" else torch.float32
     # static synthetic input batches, pre-generated on device (the
     # reference feeds one preprocessed image repeatedly, test/test.py:20-23)
     n_inputs = 4
-    inputs = [torch.randn(B, 224, 224, 3, device=dev, dtype=dtype)
+    inputs = [torch.randn(M, 224, 224, 3, device=dev, dtype=dtype)
               for _ in range(n_inputs)] if pipe.rank == 0 else None
 
     def feed(k):
-        feed_t[k] = time.perf_counter()
         return inputs[k % n_inputs]
 
     sink = {}
-    feed_t = {}
-    lat_ms = []
 
     def collect(k, y):
         sink["last"] = (k, y.shape)
-        t = feed_t.pop(k, None)
-        if t is not None:
-            lat_ms.append((time.perf_counter() - t) * 1e3)
-
-    import torch.distributed as dist
 
     def barrier_sync():
         dist.barrier()
@@ -147,13 +173,13 @@ def main():
     barrier_sync()
 
     # ---- warmup (fills pipeline, triggers graph capture)
-    pipe.run(args.warmup, feed=feed, collect=collect)
+    pipe.run(max(args.warmup, 2) * per_step, feed=feed, collect=collect)
     barrier_sync()
     pipe.reset_stats()
 
-    # ---- timed region: exactly --steps items
+    # ---- timed region: exactly --steps global batches
     t0 = time.perf_counter()
-    pipe.run(args.steps, feed=feed, collect=collect)
+    pipe.run(args.steps * per_step, feed=feed, collect=collect)
     barrier_sync()
     t1 = time.perf_counter()
 
@@ -166,18 +192,40 @@ def main():
     images = args.steps * B
     ips = images / el
 
-    if args.stats:
-        import sys
+    # ---- untimed: serialized per-item end-to-end latency (rank0 feed ->
+    # rank0 result, host-synced per item; the reference's batch-1
+    # streaming protocol operating point, test/test.py:47-49)
+    lat = None
+    L = args.latency_items
+    if L > 0 and (world == 1 or not args.no_return_results):
+        lat_ms = []
+        t_feed = {}
+
+        def lfeed(k):
+            if args.device == "cuda":
+                torch.cuda.synchronize()
+            t_feed[k] = time.perf_counter()
+            return inputs[k % n_inputs]
+
+        def lcollect(k, y):
+            if args.device == "cuda":
+                torch.cuda.synchronize()
+            lat_ms.append((time.perf_counter() - t_feed.pop(k)) * 1e3)
+
+        pipe.run(L, feed=lfeed if pipe.rank == 0 else feed,
+                 collect=lcollect if pipe.rank == 0 else collect)
+        barrier_sync()
         if pipe.rank == 0 and lat_ms:
-            srt = sorted(lat_ms[-args.steps:])
+            srt = sorted(lat_ms)
 
             def pct(p):
-                return srt[min(len(srt) - 1, int(p * len(srt)))]
+                return round(srt[min(len(srt) - 1, int(p * len(srt)))], 3)
 
-            print(f"[latency] item p50={pct(0.5):.2f}ms "
-                  f"p90={pct(0.9):.2f}ms p99={pct(0.99):.2f}ms "
-                  f"(feed->collect, includes {world}-deep pipeline "
-                  f"occupancy)", file=sys.stderr)
+            lat = {"p50": pct(0.5), "p90": pct(0.9), "p99": pct(0.99),
+                   "items": L, "micro_batch": M}
+
+    if args.stats:
+        import sys
         st = pipe.stats
         cms = st.compute_ms / max(st.items, 1)
         busy = st.compute_ms / 1e3 / el * 100 if el > 0 else 0.0
@@ -191,8 +239,9 @@ def main():
         out = {
             "metric": "images/sec (whole node) "
                       + {"resnet50": "ResNet50", "resnet101": "ResNet101",
-                         "resnet152": "ResNet152",
-                         "vgg19": "VGG19"}[args.model] + " pipeline",
+                         "resnet152": "ResNet152", "vgg19": "VGG19",
+                         "vgg19_gap": "VGG19-GAP"}[args.model]
+                      + " pipeline",
             "value": round(ips, 1),
             "unit": "images/sec",
             "n_gpus": world,
@@ -204,9 +253,11 @@ def main():
             "vs_baseline": None,
             "dtype": cfg.dtype,
             "data": "synthetic",
+            "latency_ms": lat,
             "config": {
                 "model": args.model,
                 "global_batch": B,
+                "micro_batch": M,
                 "input": "224x224x3 NHWC",
                 "parallelism": f"pp{world}",
                 "cuts": pipe.cuts,
@@ -221,6 +272,29 @@ def main():
         os.write(real_stdout, (json.dumps(out) + "\n").encode())
     os.close(real_stdout)
     dist.destroy_process_group()
+
+
+def _spawned(rank, args, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(args.gpus)
+    worker(args)
+
+
+def main():
+    args = parse_args()
+    if "WORLD_SIZE" in os.environ or args.gpus <= 1:
+        # launched under torchrun (driver's N>1 form) or single rank
+        worker(args)
+        return
+    # bare `bench.py --gpus N`: self-spawn N ranks (one per GPU) so the
+    # scaling curve needs no external launcher
+    import torch.multiprocessing as mp
+
+    port = 29500 + (os.getpid() % 1000)
+    mp.spawn(_spawned, args=(args, port), nprocs=args.gpus, join=True)
 
 
 if __name__ == "__main__":

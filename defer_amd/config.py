@@ -69,6 +69,13 @@ class PipelineConfig:
     # BASELINE.json benchmark path).
     weights_dir: Optional[str] = None
 
+    # Per-layer measured-cost file (JSON written by tools/calibrate.py /
+    # defer_amd.parallel.calibrate) used by the auto-partitioner instead
+    # of the static cost model. None -> auto: use the in-tree
+    # defer_amd/calib/{model}.json profile when one exists (SURVEY §7
+    # "per-layer-cost profiler to pick cuts", measured not hand-tuned).
+    calibration_file: Optional[str] = None
+
     # --- distributed ------------------------------------------------------
     backend: str = "nccl"         # "nccl" (RCCL over xGMI) or "gloo" (CPU)
     # Whether the last stage sends results back to rank 0 (the reference's

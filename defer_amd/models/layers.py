@@ -83,6 +83,15 @@ class GlobalAvgPool(nn.Module):
         return ops.global_avg_pool(x)
 
 
+class Flatten(nn.Module):
+    """NHWC -> (N, H*W*C). Matches Keras Flatten under channels_last
+    (the reference's VGG19 head flattens 7x7x512 -> 25088); a reshape of
+    a contiguous NHWC tensor, so free on GPU."""
+
+    def forward(self, x):
+        return x.reshape(x.shape[0], -1)
+
+
 class Dense(nn.Module):
     def __init__(self, cin, cout, bias=True, act="none"):
         super().__init__()

@@ -227,6 +227,23 @@ def split_point(numel: int) -> int:
     return max(half, 1)
 
 
+def dual_active(cfg: PipelineConfig, world: int) -> bool:
+    """Single source of truth for whether dual-rail rings are built:
+    needs a third rank to route through and a fixed-size wire (zfp+lz4
+    is variable-size). The partitioner's bandwidth credit
+    (dual_bw_boost) derives from this so the two can't drift apart."""
+    return bool(cfg.dual_rail) and world > 2 \
+        and cfg.compression != "zfp+lz4"
+
+
+def dual_bw_boost(cfg: PipelineConfig, world: int) -> float:
+    """Effective hop-bandwidth multiplier the auto-partitioner may
+    assume. 2.0 only when EVERY hop runs dual-rail (world >= 4; at
+    world == 3 the last hop stays single-rail — see hop_via — so the
+    cut chooser stays conservative there)."""
+    return 2.0 if dual_active(cfg, world) and world >= 4 else 1.0
+
+
 def hop_via(hop: int, world: int):
     """Forwarder rank for pipeline hop `hop` (ranks hop -> hop+1), or
     None if the hop stays single-rail.
@@ -334,7 +351,15 @@ class HopForwarder:
         self._error = None
 
     def begin(self, steps: int):
-        assert self.relayed == self.steps, "previous run not drained"
+        if self.relayed != self.steps:
+            # a previous run() aborted mid-stream (feed/collect raised,
+            # or drain surfaced a forwarder error): in-flight works on
+            # the rails are unrecoverable from here — the pipeline must
+            # be rebuilt, not silently restarted on inconsistent rings
+            raise RuntimeError(
+                f"dual-rail forwarder {self.src}->{self.dst} holds "
+                f"{self.steps - self.relayed} undelivered items from an "
+                f"aborted run; rebuild the pipeline before running again")
         self.steps = steps
         self.posted = 0
         self.relayed = 0

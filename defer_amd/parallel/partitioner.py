@@ -151,6 +151,7 @@ def auto_partition(model, num_stages: int, input_shape=(1, 224, 224, 3),
                    bytes_per_elem: float = 2.0,
                    eff_tflops: float = EFF_TFLOPS,
                    link_gbps: float = XGMI_LINK_GBPS,
+                   measured_us: dict = None,
                    ) -> Tuple[List[str], List[GraphModel]]:
     """Choose num_stages-1 cuts minimizing the pipeline bottleneck.
 
@@ -158,6 +159,10 @@ def auto_partition(model, num_stages: int, input_shape=(1, 224, 224, 3),
         stage_time_i = max(sum(flops_i)/eff, in_bytes_i/bw, out_bytes_i/bw)
     Minimized exactly by binary search on the bottleneck + greedy
     feasibility check over the valid articulation points.
+
+    `measured_us` ({node_name: us per image}, from parallel.calibrate)
+    replaces the static per-node time estimates with on-GPU measured
+    ones; nodes absent from the dict keep their model estimate.
     """
     gm = as_graph_model(model)
     graph = gm.graph
@@ -165,6 +170,8 @@ def auto_partition(model, num_stages: int, input_shape=(1, 224, 224, 3),
         return [], [gm]
     flops, out_bytes, time_us = node_times(graph, input_shape,
                                            bytes_per_elem)
+    if measured_us:
+        time_us = {n: measured_us.get(n, t) for n, t in time_us.items()}
     cuts_avail = graph.valid_cut_points()
     if len(cuts_avail) < num_stages - 1:
         raise ValueError(

@@ -25,8 +25,9 @@ import torch.distributed as dist
 from defer_amd.config import PipelineConfig
 from defer_amd.graph import GraphModel
 from defer_amd.parallel.comm import (Codec, DualRailRing, HopForwarder,
-                                     P2PRing, dtype_bytes, hop_via,
-                                     make_ring, split_point)
+                                     P2PRing, dtype_bytes, dual_active,
+                                     dual_bw_boost, hop_via, make_ring,
+                                     split_point)
 from defer_amd.parallel.partitioner import (as_graph_model, auto_partition,
                                             partition_model)
 
@@ -149,11 +150,29 @@ class DEFER:
                   input_stream: "queue.Queue", output_stream: "queue.Queue"):
         cfg = self.cfg
         n = len(self.computeNodes)
+        if partition_layers is None and cfg.weights_dir and n > 1:
+            # partition exactly as the checkpoint was partitioned (same
+            # contract as DistPipeline): auto cuts that differ from the
+            # saved cuts would otherwise fail load with a raw
+            # state_dict key mismatch
+            from defer_amd import checkpoint
+
+            mani = checkpoint.load_manifest(cfg.weights_dir)
+            if mani["num_stages"] != n:
+                raise ValueError(
+                    f"checkpoint has {mani['num_stages']} stages but "
+                    f"{n} compute nodes were given")
+            partition_layers = list(mani["cut_points"])
         if partition_layers is None:
+            from defer_amd.parallel.calibrate import find_calibration
+
+            gm0 = as_graph_model(model)
             partition_layers, stages = auto_partition(
-                model, n,
+                gm0, n,
                 input_shape=tuple(cfg.input_shape)
-                if cfg.input_shape else (1, 224, 224, 3))
+                if cfg.input_shape else (1, 224, 224, 3),
+                measured_us=find_calibration(gm0.model_name,
+                                             cfg.calibration_file))
         else:
             stages = partition_model(model, partition_layers)
         if len(stages) != n:
@@ -279,18 +298,17 @@ class DistPipeline:
             cuts = list(cfg.partition_layers)
             stages = partition_model(gm, cuts)
         else:
+            from defer_amd.parallel.calibrate import find_calibration
             from defer_amd.parallel.partitioner import XGMI_LINK_GBPS
 
-            # dual-rail doubles effective hop bandwidth on every hop for
-            # world >= 4 (world == 3 leaves the last hop single-rail, so
-            # stay conservative there)
-            dual = (cfg.dual_rail and self.world >= 4
-                    and cfg.compression != "zfp+lz4")
             cuts, stages = auto_partition(
                 gm, self.world,
                 input_shape=(1,) + self.batch_shape[1:],
                 bytes_per_elem=self._wire_bytes_per_elem(),
-                link_gbps=XGMI_LINK_GBPS * (2.0 if dual else 1.0))
+                link_gbps=XGMI_LINK_GBPS
+                * dual_bw_boost(cfg, self.world),
+                measured_us=find_calibration(gm.model_name,
+                                             cfg.calibration_file))
         if len(stages) != self.world:
             raise ValueError(f"{len(stages)} stages != world {self.world}")
         self.cuts = cuts
@@ -315,10 +333,7 @@ class DistPipeline:
         self.send_ring = None
         self.result_ring = None
         self.fwd = None
-        # dual-rail applies to fixed-size data hops only (zfp+lz4 is
-        # variable-size) and needs a third rank to route through
-        dual = (cfg.dual_rail and self.world > 2
-                and cfg.compression != "zfp+lz4")
+        dual = dual_active(cfg, self.world)
         if cfg.dual_rail and not dual and self.rank == 0:
             import sys
 
@@ -327,22 +342,23 @@ class DistPipeline:
                      if cfg.compression == "zfp+lz4"
                      else "(needs world > 2)"), file=sys.stderr)
         if self.world > 1:
+            # a wire too small to split (codec.wire_numel < 16) stays
+            # single-rail: a zero/one-element second half has
+            # backend-dependent isend/irecv behavior
+            def _ring(codec, hop):
+                via = hop_via(hop, self.world) if dual else None
+                if via is not None and codec.wire_numel >= 16:
+                    return DualRailRing(codec, cfg.ring_depth, via)
+                return make_ring(codec, cfg.ring_depth)
+
             if self.rank > 0:
                 self.in_codec = Codec(cfg, self.in_shape, self.dtype,
                                       self.device)
-                via = hop_via(self.rank - 1, self.world) if dual else None
-                self.recv_ring = (
-                    DualRailRing(self.in_codec, cfg.ring_depth, via)
-                    if via is not None
-                    else make_ring(self.in_codec, cfg.ring_depth))
+                self.recv_ring = _ring(self.in_codec, self.rank - 1)
             if self.rank < self.world - 1:
                 self.out_codec = Codec(cfg, self.out_shape, self.dtype,
                                        self.device)
-                via = hop_via(self.rank, self.world) if dual else None
-                self.send_ring = (
-                    DualRailRing(self.out_codec, cfg.ring_depth, via)
-                    if via is not None
-                    else make_ring(self.out_codec, cfg.ring_depth))
+                self.send_ring = _ring(self.out_codec, self.rank)
             if dual:
                 # this rank's forwarding duty (at most one hop routes
                 # through any given rank — see comm.hop_via)
@@ -352,6 +368,9 @@ class DistPipeline:
                 if fwd_hop is not None:
                     c = Codec(cfg, self._chain_shapes[fwd_hop],
                               self.dtype, self.device)
+                if fwd_hop is not None and c.wire_numel >= 16:
+                    # same small-wire guard as _ring: a hop that stayed
+                    # single-rail has no second half to relay
                     self.fwd = HopForwarder(
                         c.wire_numel - split_point(c.wire_numel),
                         c.wire_dtype, self.device, cfg.ring_depth,

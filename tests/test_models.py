@@ -38,18 +38,33 @@ def test_resnet50_partition_equivalence():
 
 
 def test_vgg19_shapes_and_partition():
+    # faithful Keras VGG19 head: Flatten(7x7x512)->fc4096 needs the real
+    # 224x224 input (fc1 is the published 25088x4096 GEMM)
     m = vgg19(num_classes=10)
-    x = torch.randn(1, 64, 64, 3)
+    p = dict(m.named_parameters())
+    assert p["layers.fc1.weight"].shape == (4096, 7 * 7 * 512)
+    x = torch.randn(1, 224, 224, 3)
     with torch.no_grad():
         want = m(x)
     assert want.shape == (1, 10)
-    cuts, stages = auto_partition(m, 4, input_shape=(1, 64, 64, 3))
+    cuts, stages = auto_partition(m, 4, input_shape=(1, 224, 224, 3))
     assert len(stages) == 4
     with torch.no_grad():
         z = x
         for s in stages:
             z = s(z)
     assert torch.equal(z, want)
+
+
+def test_vgg19_gap_variant():
+    from defer_amd.models import vgg19_gap
+
+    m = vgg19_gap(num_classes=10)
+    assert dict(m.named_parameters())["layers.fc1.weight"].shape == (4096, 512)
+    x = torch.randn(1, 64, 64, 3)  # GAP head is input-size independent
+    with torch.no_grad():
+        y = m(x)
+    assert y.shape == (1, 10)
 
 
 def test_auto_partition_resnet_8():
