@@ -60,20 +60,31 @@ const float* zeros_buf() {
 // mismatch). Unbounded but tiny (one entry per stem conv weight);
 // in-place weight mutation is not detected (inference engine — the
 // reference also ships weights exactly once, dispatcher.py:57).
+struct PrepEntry {
+    Tensor src;       // pinning src keeps its pointer unique
+    Tensor prepped;
+    uint32_t version; // src._version at prep time: in-place mutation of
+                      // the source weight (e.g. checkpoint reload into
+                      // the same tensors) bumps it and invalidates the
+                      // cached repack
+};
+
 Tensor cached_weight_prep(const Tensor& w, int64_t variant,
                           const std::function<Tensor()>& make) {
-    static std::unordered_map<uint64_t, std::pair<Tensor, Tensor>> cache;
+    static std::unordered_map<uint64_t, PrepEntry> cache;
     static std::mutex mu;
     uint64_t key = (uint64_t)(uintptr_t)w.data_ptr() * 31 +
                    (uint64_t)w.numel() * 7 + (uint64_t)variant;
+    uint32_t ver = w.unsafeGetTensorImpl()->version_counter().current_version();
     std::lock_guard<std::mutex> g(mu);
     auto it = cache.find(key);
     if (it != cache.end() &&
-        it->second.first.data_ptr() == w.data_ptr() &&
-        it->second.first.numel() == w.numel())
-        return it->second.second;
+        it->second.src.data_ptr() == w.data_ptr() &&
+        it->second.src.numel() == w.numel() &&
+        it->second.version == ver)
+        return it->second.prepped;
     Tensor t = make();
-    cache[key] = {w, t};     // pinning w keeps its pointer unique
+    cache[key] = {w, t, ver};
     return t;
 }
 
@@ -376,6 +387,32 @@ Tensor lz4_compress(Tensor x) {
     return o.narrow(0, 0, 4 * (2 + nb + 1) + (long)total);
 }
 
+// Fully async zfp+lz4 hop encode: compress `x` (u8, the ZFP wire) into
+// the caller's preallocated worst-case `out` ring slot and write the
+// total wire byte count into `len_out` (device int64) — no host sync.
+// The hop sends `len_out` as the size message straight from the device
+// and reads the host copy one item later (comm.VarP2PRing), keeping the
+// sender's host loop ahead of the device (the round-1 version's
+// per-item .to(kCPU) sync serialized the pipeline).
+void lz4_compress_into(Tensor x, Tensor scratch, Tensor out,
+                       Tensor len_out) {
+    TORCH_CHECK(x.is_cuda() && x.is_contiguous()
+                && x.scalar_type() == at::kByte, "x must be GPU u8 contig");
+    long n = x.numel();
+    TORCH_CHECK(n > 0, "empty input");
+    TORCH_CHECK(scratch.numel() >= defer_hip::lz4_scratch_bytes(n)
+                && scratch.is_cuda(), "scratch too small");
+    TORCH_CHECK(out.numel() >= defer_hip::lz4_max_compressed(n)
+                && out.is_cuda() && out.scalar_type() == at::kByte,
+                "out too small");
+    TORCH_CHECK(len_out.is_cuda() && len_out.numel() == 1
+                && len_out.scalar_type() == at::kLong, "bad len_out");
+    hipStream_t s = cur_stream();
+    defer_hip::launch_lz4_compress(bptr(x), n, bptr_mut(scratch),
+                                   bptr_mut(out), s);
+    defer_hip::launch_lz4_wire_len(bptr(out), n, bptr_mut(len_out), s);
+}
+
 Tensor lz4_decompress(Tensor comp, int64_t raw_len) {
     TORCH_CHECK(comp.is_cuda() && comp.is_contiguous()
                 && comp.scalar_type() == at::kByte, "bad comp buffer");
@@ -403,5 +440,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("out") = py::none(), py::arg("phases") = 3);
     m.def("zfp_decode", &zfp_decode);
     m.def("lz4_compress", &lz4_compress);
+    m.def("lz4_compress_into", &lz4_compress_into);
+    m.def("lz4_scratch_bytes",
+          [](int64_t n) { return defer_hip::lz4_scratch_bytes(n); });
     m.def("lz4_decompress", &lz4_decompress);
 }

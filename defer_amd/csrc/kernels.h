@@ -70,6 +70,10 @@ long lz4_max_compressed(long n);
 long lz4_scratch_bytes(long n);
 void launch_lz4_compress(const void* in, long n, void* scratch, void* out,
                          hipStream_t s);
+// writes the total wire length (header + payload bytes) of a stream
+// produced by launch_lz4_compress into *len_out (device int64), async
+void launch_lz4_wire_len(const void* out_stream, long n, void* len_out,
+                         hipStream_t s);
 void launch_lz4_decompress(const void* comp, void* out, long raw_len,
                            hipStream_t s);
 

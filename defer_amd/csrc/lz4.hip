@@ -248,6 +248,17 @@ lz4_decompress_kernel(const u8* __restrict__ comp, u8* __restrict__ out,
     for (int i = lane; i < blen; i += WAVE) out[base + i] = dst[i];
 }
 
+// ---- wire length: total stream bytes as a device-side int64 -------------
+// Lets the hop ship the size message straight from device memory (RCCL
+// send of a device tensor) with no host round-trip on the critical path.
+__global__ void
+lz4_wire_len_kernel(const u32* __restrict__ header, int nblocks,
+                    long long* __restrict__ len_out) {
+    if (threadIdx.x == 0)
+        *len_out = (long long)LZ_HDR(nblocks) +
+                   (long long)header[2 + nblocks];
+}
+
 // ---------------------------------------------------------------------------
 namespace defer_hip {
 
@@ -274,6 +285,13 @@ void launch_lz4_compress(const void* in, long n, void* scratch, void* out,
     int gather_grid = nb < 4096 ? nb : 4096;
     hipLaunchKernelGGL(lz4_gather_kernel, dim3(gather_grid), dim3(64), 0,
                        s, scr, (const u32*)out, (u8*)out, nb);
+}
+
+void launch_lz4_wire_len(const void* out_stream, long n, void* len_out,
+                         hipStream_t s) {
+    int nb = (int)((n + LZ_BLK - 1) / LZ_BLK);
+    hipLaunchKernelGGL(lz4_wire_len_kernel, dim3(1), dim3(1), 0, s,
+                       (const u32*)out_stream, nb, (long long*)len_out);
 }
 
 void launch_lz4_decompress(const void* comp, void* out, long raw_len,

@@ -148,24 +148,75 @@ class P2PRing:
         self.set_work(k, dist.isend(wire, dst=dst))
         return wire.numel() * dtype_bytes(wire.dtype)
 
+    def flush(self) -> int:
+        """Issue any deferred tail send (variable-size rings); no-op
+        for fixed-size rings whose sends are issued eagerly."""
+        return 0
+
 
 class VarP2PRing(P2PRing):
     """Variable-size hop (zfp+lz4): each item is an 8-byte size message
     followed by the exact-size payload — the RCCL analogue of the
     reference's length-prefixed TCP framing (node_state.py:44-54).
 
-    RCCL p2p matches send/recv strictly in issue order per (src,dst)
-    pair (no tags), so the receiver must alternate size/payload recvs in
-    the same order the sender issues them; size recvs are posted one
-    item ahead (right after the previous payload recv), never `depth`
-    ahead."""
+    RCCL and gloo match p2p ops strictly in issue order per (src,dst)
+    pair (no tags), so BOTH sides follow one interleave for K items:
+
+        s_0, s_1, p_0, s_2, p_1, ..., s_{K-1}, p_{K-2}, p_{K-1}
+
+    (s_k = size of item k, p_k = its exact-size payload). The payload
+    send of item k is deferred until item k+1's encode has been issued
+    (or flush()): on RCCL the size message is sent straight from device
+    memory — lz4_compress_into writes the device int64, no host
+    round-trip — and the host reads a pinned copy one item later, so
+    the sender's host loop never blocks on the codec kernels of the
+    item it is currently issuing (the round-1 per-item .to(kCPU) sync
+    serialized the 8-stage compressed config).
+
+    The deferred interleave requires that no rank host-blocks between
+    issuing s_k and p_k. True on RCCL (Work.wait is stream-level) but
+    NOT on gloo: rank0's host-blocking result wait for item k would
+    deadlock against its own deferred p_k. gloo (CPU) therefore uses
+    the eager interleave s_0, p_0, s_1, p_1, ... (sizes are known
+    synchronously there anyway); both sides of a hop share a backend,
+    so the orders always agree. DEFER_AMD_VARRING_DEFER=1 forces the
+    deferred order on gloo for protocol tests (only safe without a
+    result-return cycle)."""
 
     def __init__(self, codec: Codec, depth: int):
-        super().__init__(codec, depth)
+        # the one-item payload deferral needs >= 2 slots (at depth 1 the
+        # undelivered payload's slot would be re-encoded)
+        super().__init__(codec, max(depth, 2))
+        self.depth = max(depth, 2)
         self.sizes = [torch.zeros(1, dtype=torch.long, device=codec.device)
-                      for _ in range(depth)]
-        self.size_works: List[Optional[dist.Work]] = [None] * depth
+                      for _ in range(self.depth)]
+        self.size_works: List[Optional[dist.Work]] = [None] * self.depth
         self._last_wire = 0
+        self.cuda = torch.device(codec.device).type == "cuda"
+        import os
+
+        self.defer = self.cuda or \
+            os.environ.get("DEFER_AMD_VARRING_DEFER") == "1"
+        # sender-side deferral state
+        self._pending: Optional[int] = None   # item awaiting payload send
+        self._dst: Optional[int] = None
+        self._host_n = [0] * self.depth       # CPU path: size known now
+        if self.cuda:
+            from defer_amd import ops as _ops
+
+            self._hip = _ops._load_hip()
+            if self._hip is None:
+                raise RuntimeError("HIP extension missing for zfp+lz4 hop")
+            zb = codec.zfp_bytes
+            self._zfp_buf = torch.empty(zb, dtype=torch.uint8,
+                                        device=codec.device)
+            self._scratch = torch.empty(
+                int(self._hip.lz4_scratch_bytes(zb)), dtype=torch.uint8,
+                device=codec.device)
+            self._pinned = [torch.zeros(1, dtype=torch.long,
+                                        pin_memory=True)
+                            for _ in range(self.depth)]
+            self._events = [torch.cuda.Event() for _ in range(self.depth)]
 
     def _wait_size(self, k: int):
         w = self.size_works[k % self.depth]
@@ -181,10 +232,14 @@ class VarP2PRing(P2PRing):
     def wait_recv(self, k: int, steps: int, src: int) -> torch.Tensor:
         i = k % self.depth
         self._wait_size(k)
-        n = int(self.sizes[i].item())
+        n = int(self.sizes[i].item())     # host sync: size arrival
+        if self.defer and k + 1 < steps:  # s_{k+1} precedes p_k
+            j = (k + 1) % self.depth
+            self.size_works[j] = dist.irecv(self.sizes[j], src=src)
         view = self.bufs[i][:n]
-        dist.recv(view, src=src)          # exact-size, in-order payload
-        if k + 1 < steps:                 # lookahead: next item's size
+        w = dist.irecv(view, src=src)     # exact-size, in-order payload
+        w.wait()    # stream-level on RCCL (host moves on); host on gloo
+        if not self.defer and k + 1 < steps:  # eager: s_{k+1} after p_k
             j = (k + 1) % self.depth
             self.size_works[j] = dist.irecv(self.sizes[j], src=src)
         self._last_wire = n
@@ -196,15 +251,47 @@ class VarP2PRing(P2PRing):
     # -- sender side
     def send_encoded(self, k: int, y: torch.Tensor, dst: int) -> int:
         i = k % self.depth
-        self.wait(k)                      # payload slot free
-        self._wait_size(k)
-        wire = self.codec.encode(y)
-        n = wire.numel()
-        self.sizes[i].fill_(n)
-        self.size_works[i] = dist.isend(self.sizes[i], dst=dst)
-        buf = self.bufs[i][:n]
-        buf.copy_(wire)
-        self.set_work(k, dist.isend(buf, dst=dst))
+        self._dst = dst
+        self.wait(k)                      # payload slot free (k-depth)
+        self._wait_size(k)                # size slot free (k-depth)
+        if self.cuda:
+            from defer_amd.ops import codec as zc
+
+            zc.zfp_encode(y, self.codec.rate, out=self._zfp_buf)
+            self._hip.lz4_compress_into(self._zfp_buf, self._scratch,
+                                        self.bufs[i], self.sizes[i])
+            self.size_works[i] = dist.isend(self.sizes[i], dst=dst)
+            self._pinned[i].copy_(self.sizes[i], non_blocking=True)
+            self._events[i].record()
+        else:
+            wire = self.codec.encode(y)
+            n = wire.numel()
+            self.bufs[i][:n].copy_(wire)
+            self.sizes[i].fill_(n)
+            self._host_n[i] = n
+            self.size_works[i] = dist.isend(self.sizes[i], dst=dst)
+        if not self.defer:                # eager: p_k right after s_k
+            self._pending = k
+            return self.flush()
+        sent = self.flush()               # deferred p_{k-1}
+        self._pending = k
+        return sent
+
+    def flush(self) -> int:
+        """Issue the deferred payload send (item issued last). Called
+        between items (p_{k-1} after s_k) and by the pipeline's drain
+        for the final item."""
+        if self._pending is None:
+            return 0
+        j = self._pending
+        i = j % self.depth
+        if self.cuda:
+            self._events[i].synchronize()   # encode j + D2H done
+            n = int(self._pinned[i].item())
+        else:
+            n = self._host_n[i]
+        self.set_work(j, dist.isend(self.bufs[i][:n], dst=self._dst))
+        self._pending = None
         return n
 
 

@@ -559,6 +559,7 @@ class DistPipeline:
         # run() call of a warmup+timed sequence at world_size > 1)
         for ring in (self.send_ring, self.result_ring):
             if ring is not None:
+                self.stats.bytes_out += ring.flush()  # deferred tail send
                 for i, w in enumerate(ring.works):
                     if w is not None:
                         w.wait()

@@ -343,3 +343,90 @@ def test_dual_rail_with_fp8_wire():
     ride separate rails)."""
     _run(4, ["add_4", "add_8", "add_12"], steps=3, compression="fp8",
          tol=1e-6, dual_rail=True)
+
+
+# ---------------------------------------------------------------------------
+# deferred variable-size interleave (the RCCL-order protocol, forced on gloo)
+# ---------------------------------------------------------------------------
+
+def _defer_worker(rank, world, port, q, steps):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    # force the CUDA-order interleave (s_0, s_1, p_0, s_2, p_1, ...) on
+    # gloo; safe only without the rank0 result-return cycle, so
+    # return_results=False and the last rank collects
+    os.environ["DEFER_AMD_VARRING_DEFER"] = "1"
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        torch.manual_seed(0)
+        from defer_amd.parallel.pipeline import DistPipeline
+
+        model = resnet50()
+        cuts = ["add_4", "add_12"]
+        cfg = PipelineConfig(device="cpu", dtype="fp32",
+                             partition_layers=cuts, ring_depth=2,
+                             compression="zfp+lz4", zfp_rate_bits=14,
+                             backend="gloo", return_results=False)
+        B = 1
+        pipe = DistPipeline(model, cfg, (B, 64, 64, 3))
+        torch.manual_seed(1234)
+        inputs = [torch.randn(B, 64, 64, 3) for _ in range(steps)]
+        res1, res2 = {}, {}
+        feed = (lambda k: inputs[k]) if rank == 0 else None
+        pipe.run(steps, feed=feed,
+                 collect=lambda k, y: res1.__setitem__(k, y.clone()))
+        # second run on the same rings: deferral state must reset
+        pipe.run(steps, feed=feed,
+                 collect=lambda k, y: res2.__setitem__(k, y.clone()))
+        if rank == world - 1:
+            from defer_amd.parallel.comm import Codec
+            from defer_amd.parallel.partitioner import partition_model
+            from defer_amd.parallel.pipeline import StageExecutor
+
+            stages = partition_model(model, cuts)
+            execs = [StageExecutor(s, "cpu", torch.float32)
+                     for s in stages]
+
+            def fwd(x):
+                z = x
+                for i, ex in enumerate(execs):
+                    with torch.no_grad():
+                        z = ex.run(z)
+                    if i < len(execs) - 1:
+                        c = Codec(cfg, tuple(z.shape), torch.float32,
+                                  "cpu")
+                        z = c.decode(c.encode(z))
+                return z
+
+            for k in range(steps):
+                want = fwd(inputs[k])
+                e1 = (res1[k] - want).abs().max().item()
+                e2 = (res2[k] - want).abs().max().item()
+                q.put(("err", k, max(e1, e2)))
+            q.put(("done", rank, None))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_var_ring_deferred_interleave_gloo():
+    """The deferred size/payload interleave used on RCCL, forced on gloo
+    (world 3, two zfp+lz4 hops, no result return): outputs must match
+    the codec-simulated chain exactly, across two sequential runs."""
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    steps = 4
+    procs = [ctx.Process(target=_defer_worker,
+                         args=(r, 3, 29721, q, steps))
+             for r in range(3)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=300)
+        assert p.exitcode == 0, f"worker failed: {p.exitcode}"
+    msgs = []
+    while not q.empty():
+        msgs.append(q.get())
+    errs = [m[2] for m in msgs if m[0] == "err"]
+    assert len(errs) == steps
+    for e in errs:
+        assert e < 1e-6, f"deferred interleave corrupted payloads: {e}"
