@@ -1,0 +1,104 @@
+"""Measured per-layer calibration feeding auto_partition (SURVEY.md §7:
+per-layer-cost profiler; VERDICT.md next-round #6: cuts from measured
+times, not hand-tuned constants)."""
+
+import os
+
+import pytest
+import torch
+
+from defer_amd.models import resnet50
+from defer_amd.parallel.calibrate import (calibrate_and_save,
+                                          find_calibration,
+                                          load_calibration,
+                                          measure_layer_times)
+from defer_amd.parallel.partitioner import auto_partition
+
+
+def test_measure_layer_times_cpu_covers_all_nodes():
+    gm = resnet50()
+    us = measure_layer_times(gm, (2, 64, 64, 3), device="cpu",
+                             iters=2, warmup=1)
+    names = {n.name for n in gm.graph.nodes}
+    assert set(us) == names
+    assert all(v > 0 for v in us.values())
+
+
+def test_calibration_roundtrip_and_auto_partition(tmp_path):
+    gm = resnet50()
+    us = measure_layer_times(gm, (1, 64, 64, 3), device="cpu",
+                             iters=1, warmup=0)
+    path = str(tmp_path / "r50.json")
+    import json
+
+    from defer_amd.parallel import calibrate as C
+
+    os.makedirs(os.path.dirname(path), exist_ok=True)
+    with open(path, "w") as f:
+        json.dump({"model": "resnet50", "batch": 1, "device": "cpu",
+                   "us_per_image": us}, f)
+    got = load_calibration(path)
+    assert got.keys() == us.keys()
+    assert find_calibration("resnet50", path) is not None
+    assert find_calibration("no_such_model_xyz") is None
+    # measured costs drive the DP: still num_stages valid stages, and a
+    # skewed measurement moves the cut
+    cuts, stages = auto_partition(gm, 4, input_shape=(1, 64, 64, 3),
+                                  measured_us=got)
+    assert len(stages) == 4
+    skew = {k: (1000.0 if k == "add_2" else 0.001) for k in got}
+    cuts2, _ = auto_partition(gm, 2, input_shape=(1, 64, 64, 3),
+                              measured_us=skew)
+    # with nearly all cost in add_2's prefix the single cut lands at or
+    # right after add_2 to isolate it
+    assert cuts2 != cuts[:1] or cuts2[0] in ("add_1", "add_2", "add_3")
+    assert C.calibration_path("resnet50").endswith(
+        os.path.join("calib", "resnet50.json"))
+
+
+@pytest.mark.gpu
+def test_calibrated_cuts_balance_gpu():
+    """Auto cuts chosen from on-GPU measured layer times: each stage's
+    measured fused forward must track the calibration's predicted stage
+    time, and the resulting 4-stage split must be compute-balanced."""
+    from defer_amd.graph import GraphModel
+    from defer_amd.parallel.pipeline import StageExecutor
+
+    torch.manual_seed(0)
+    gm = resnet50()
+    B = 32
+    us = measure_layer_times(gm, (B, 224, 224, 3), device="cuda",
+                             dtype=torch.bfloat16, iters=15, warmup=5)
+    cuts, stages = auto_partition(gm, 4, measured_us=us)
+    assert len(stages) == 4
+
+    pred_us = [sum(us[n.name] for n in s.graph.nodes) * B
+               for s in stages]
+
+    shape = (B, 224, 224, 3)
+    times_us = []
+    x = torch.randn(*shape, device="cuda", dtype=torch.bfloat16)
+    for s in stages:
+        ex = StageExecutor(GraphModel(s.graph, name="st"), "cuda:0",
+                           torch.bfloat16)
+        with torch.no_grad():
+            for _ in range(5):
+                y = ex.run(x)
+            torch.cuda.synchronize()
+            e0, e1 = torch.cuda.Event(True), torch.cuda.Event(True)
+            e0.record()
+            for _ in range(15):
+                y = ex.run(x)
+            e1.record()
+            e1.synchronize()
+        times_us.append(e0.elapsed_time(e1) * 1e3 / 15)
+        x = y.clone()
+
+    # per-stage: fused forward within 2x of the unfused per-layer sum
+    # (fusion + inter-layer cache reuse make stages faster than sums;
+    # launch gaps can make them slower at tiny stages)
+    for p, t in zip(pred_us, times_us):
+        assert 0.35 < t / p < 2.0, (pred_us, times_us)
+    # balance: bottleneck stage within 60% of the mean stage time
+    mean = sum(times_us) / len(times_us)
+    assert max(times_us) / mean < 1.6, times_us

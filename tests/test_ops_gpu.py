@@ -68,7 +68,66 @@ def test_conv2d_bn_act(case):
                             residual=rg)
     assert got.shape == want.shape
     e = _relerr(got, want)
-    assert e < 0.03, f"conv rel err {e}"
+    assert e < 0.005, f"conv rel err {e}"
+
+
+def _rand_conv_cases(n=24, seed=20260914):
+    """Seeded random shape sweep across the conv dispatch paths (igemm,
+    gemm-mode 1x1, window, stem/channel-pad) — same generator as
+    tools/tolcheck.py so measured envelopes map 1:1 to cases."""
+    g = torch.Generator().manual_seed(seed)
+
+    def ri(lo, hi):
+        return int(torch.randint(lo, hi + 1, (1,), generator=g))
+
+    cases = []
+    for _ in range(n):
+        R = [1, 3, 3, 7][ri(0, 3)]
+        stride = ri(1, 2)
+        N = ri(1, 3)
+        H = ri(7, 48)
+        W = ri(7, 48)
+        cin_pool = [3, 8, 16, 24, 40, 64, 72, 96, 128, 192, 256]
+        Cin = cin_pool[ri(0, len(cin_pool) - 1)]
+        Cout = 8 * ri(1, 48)
+        pad = R // 2 if ri(0, 1) else 0
+        if H + 2 * pad < R or W + 2 * pad < R:
+            pad = R // 2
+            if H + 2 * pad < R:
+                H = R
+            if W + 2 * pad < R:
+                W = R
+        act = "relu" if ri(0, 1) else "none"
+        has_res = bool(ri(0, 1)) and R == 1 and stride == 1 and pad == 0
+        cases.append((N, H, W, Cin, Cout, R, stride, pad, act, has_res))
+    return cases
+
+
+@requires_gpu
+@pytest.mark.parametrize("case", _rand_conv_cases(),
+                         ids=[f"s{i}" for i in range(24)])
+def test_conv_shape_sweep(case):
+    """Property-style shape sweep: all dispatch paths, tolerance set
+    from the measured envelope (tools/tolcheck.py), ~5x headroom."""
+    N, H, W, Cin, Cout, R, stride, pad, act, has_res = case
+    torch.manual_seed(hash((N, H, W, Cin, Cout, R, stride)) % 2**31)
+    x = torch.randn(N, H, W, Cin)
+    w = torch.randn(Cout, R, R, Cin) * (2.0 / (Cin * R * R)) ** 0.5
+    scale = torch.rand(Cout) + 0.5
+    bias = torch.randn(Cout) * 0.1
+    OH = (H + 2 * pad - R) // stride + 1
+    OW = (W + 2 * pad - R) // stride + 1
+    res = torch.randn(N, OH, OW, Cout) if has_res else None
+    xg, wg = _to_dev_bf16(x, w)
+    rg = _to_dev_bf16(res)[0] if has_res else None
+    want = ref.conv2d_bn_act(xg.cpu(), wg.cpu(), scale, bias, stride,
+                             pad, act, rg.cpu() if has_res else None)
+    got = ops.conv2d_bn_act(xg, wg, scale.to(DEV), bias.to(DEV),
+                            stride=stride, padding=pad, act=act,
+                            residual=rg)
+    assert got.shape == want.shape
+    e = _relerr(got, want)
+    assert e < 0.005, f"conv sweep rel err {e} for {case}"
 
 
 @requires_gpu
@@ -80,7 +139,7 @@ def test_conv_zero_padding_boundary():
     xg, wg = _to_dev_bf16(x, w)
     want = ref.conv2d_bn_act(xg.cpu(), wg.cpu(), None, None, 1, 1, "none")
     got = ops.conv2d_bn_act(xg, wg, None, None, stride=1, padding=1)
-    assert _relerr(got, want) < 0.02
+    assert _relerr(got, want) < 0.005
 
 
 @requires_gpu
@@ -107,7 +166,7 @@ def test_global_avg_pool():
     (xg,) = _to_dev_bf16(x)
     want = ref.global_avg_pool(xg.cpu())
     got = ops.global_avg_pool(xg)
-    assert _relerr(got, want) < 0.02
+    assert _relerr(got, want) < 0.01
 
 
 @requires_gpu
@@ -118,7 +177,41 @@ def test_linear():
     xg, wg = _to_dev_bf16(x, w)
     want = ref.linear(xg.cpu(), wg.cpu(), b)
     got = ops.linear(xg, wg, b.to(DEV))
-    assert _relerr(got, want) < 0.03
+    assert _relerr(got, want) < 0.005
+
+
+@requires_gpu
+def test_linear_large_k_vgg_fc1():
+    """The faithful VGG19 fc1 (25088x4096): large-K accumulation path of
+    the MFMA GEMM (VERDICT.md next-round #5)."""
+    x = torch.randn(8, 25088)
+    w = torch.randn(4096, 25088) * (1 / 25088) ** 0.5
+    b = torch.randn(4096)
+    xg, wg = _to_dev_bf16(x, w)
+    want = ref.linear(xg.cpu(), wg.cpu(), b)
+    got = ops.linear(xg, wg, b.to(DEV))
+    assert _relerr(got, want) < 0.005
+
+
+@requires_gpu
+def test_vgg19_full_forward_vs_cpu():
+    """Whole faithful-head VGG19 bf16 GPU forward vs fp32 CPU reference
+    (flatten -> 25088x4096 fc1 in the real graph)."""
+    from defer_amd.graph import GraphModel
+    from defer_amd.models import vgg19
+    from defer_amd.parallel.pipeline import StageExecutor
+
+    torch.manual_seed(0)
+    m = vgg19()
+    x = torch.randn(2, 224, 224, 3)
+    with torch.no_grad():
+        want = m(x).float()
+    ex = StageExecutor(GraphModel(m.graph), DEV, torch.bfloat16,
+                       use_graph=False)
+    with torch.no_grad():
+        got = ex.run(x.to(DEV, torch.bfloat16)).float().cpu()
+    cos = torch.nn.functional.cosine_similarity(got, want, dim=-1)
+    assert (cos > 0.98).all(), f"cosine {cos}"
 
 
 @requires_gpu
