@@ -191,6 +191,9 @@ def worker(args):
     el = elapsed.item()
     images = args.steps * B
     ips = images / el
+    import copy
+
+    timed_stats = copy.copy(pipe.stats)   # before the latency pass runs
 
     # ---- untimed: serialized per-item end-to-end latency (rank0 feed ->
     # rank0 result, host-synced per item; the reference's batch-1
@@ -226,7 +229,7 @@ def worker(args):
 
     if args.stats:
         import sys
-        st = pipe.stats
+        st = timed_stats
         cms = st.compute_ms / max(st.items, 1)
         busy = st.compute_ms / 1e3 / el * 100 if el > 0 else 0.0
         print(f"[stage {pipe.rank}] items={st.items} images={st.images} "

@@ -10,6 +10,11 @@ measured stage times on MI355X, not hand-tuned constants."""
 
 import argparse
 
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
+
 import torch
 
 from defer_amd.models import MODELS

@@ -9,6 +9,8 @@ Prints PROBE_OK or the failure mode."""
 import os
 import sys
 
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
+
 import torch
 import torch.distributed as dist
 import torch.multiprocessing as mp

@@ -5,7 +5,10 @@ fixed op-test cases plus a seeded random conv shape sweep. Output feeds
 the tolerance choices in tests/test_ops_gpu.py (set
 asserts at ~2-5x the measured envelope, not 3%)."""
 
+import os
 import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 
 import torch
 
