@@ -147,7 +147,19 @@ struct BitReader {
 // executes it changed.
 #define CGRP 256                     // blocks per workgroup round
 
-template <bool BF16_IN, int PH = 3>   // PH: bit0=transform, bit1=serialize
+// Panel staging (PANEL=true, requires b2 % 16 == 0 so a 16-block run
+// shares one (bi,bj) row tile): the per-lane direct gather reads 16
+// scattered 8-byte segments per block — ~1/8 of each cache line used,
+// which capped encode at ~130 GB/s (profiles/zfp_codec_throughput.txt).
+// Instead each wave owns 64 CONSECUTIVE blocks per round, staged as 4
+// panels of 16 blocks: a panel's footprint is 16 rows x 64 contiguous
+// d2-values, loaded with fully dense 128-byte row reads into an LDS
+// slab and re-read block-wise (row stride padded to dodge bank
+// conflicts). All boundary tensors the pipeline ships have C >= 64, so
+// the panel path is the hot one; b2 % 16 != 0 falls back to the
+// direct-gather path.
+template <bool BF16_IN, int PH = 3,   // PH: bit0=transform, bit1=serialize
+          bool PANEL = false>
 __global__ __launch_bounds__(CGRP, 2) void zfp_encode_kernel(
     const void* __restrict__ xv, u32* __restrict__ out, int d0, int d1,
     int d2, int b0, int b1, int b2, int rate) {
@@ -161,6 +173,9 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_encode_kernel(
 
     __shared__ u64 s_planes[CGRP][PLANES + 1];  // +1: LDS bank pad
     __shared__ u32 s_hdr[CGRP];                 // 0 = zero block
+    using elem_t = typename std::conditional<BF16_IN, bf16, float>::type;
+    constexpr int SLAB_W = BF16_IN ? 66 : 65;   // pad: rows spread banks
+    __shared__ elem_t s_slab[4][16][PANEL ? SLAB_W : 1];
 
     // incremental block coordinates for this wave's stride-nwaves walk
     // (no runtime division in the loop) and a PF-deep gather pipeline so
@@ -168,8 +183,62 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_encode_kernel(
     constexpr int PF = 4;
     for (long base = (long)blockIdx.x * CGRP; base < nblocks;
          base += (long)gridDim.x * CGRP) {
-        // ---- phase 1: each wave transforms CGRP/nwaves blocks
-        if (PH & 1) {
+        // ---- phase 1 (panel): 4 panels of 16 consecutive blocks/wave
+        if (PANEL && (PH & 1)) {
+            for (int p = 0; p < 64 / 16; ++p) {
+                long pb = base + wavei * 64 + p * 16;
+                if (pb >= nblocks) break;
+                int npan = (int)(nblocks - pb < 16 ? nblocks - pb : 16);
+                int bk0 = (int)(pb % b2);
+                long t = pb / b2;
+                int bj = (int)(t % b1), bi = (int)(t / b1);
+                // dense row reads: 64 contiguous values per row
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    int gi = bi * 4 + (r >> 2); if (gi > d0 - 1) gi = d0 - 1;
+                    int gj = bj * 4 + (r & 3);  if (gj > d1 - 1) gj = d1 - 1;
+                    int gk = bk0 * 4 + lane;    if (gk > d2 - 1) gk = d2 - 1;
+                    long idx = ((long)gi * d1 + gj) * d2 + gk;
+                    s_slab[wavei][r][lane] =
+                        ((const elem_t*)xv)[idx];
+                }
+                for (int b = 0; b < npan; ++b) {
+                    int s = wavei * 64 + p * 16 + b;
+                    elem_t raw = s_slab[wavei][lane >> 2]
+                                       [b * 4 + (lane & 3)];
+                    float v = BF16_IN ? bf2f(*(bf16*)&raw)
+                                      : *(float*)&raw;
+                    float av = fabsf(v);
+#pragma unroll
+                    for (int off = 32; off > 0; off >>= 1)
+                        av = fmaxf(av, __shfl_xor(av, off));
+                    if (!(av > 0.f) || !isfinite(av)) {
+                        if (lane == 0) s_hdr[s] = 0;
+                        continue;
+                    }
+                    int emax;
+                    frexpf(av, &emax);
+                    int q = (int)rintf(v * ldexpf(1.0f, QBITS - emax));
+                    q = fwd_axis(q, lane, 1);
+                    q = fwd_axis(q, lane, 4);
+                    q = fwd_axis(q, lane, 16);
+                    q = __shfl(q, ZPERM[lane]);
+                    u32 u = ((u32)q + NBMASK) ^ NBMASK;
+                    if (lane == 0)
+                        s_hdr[s] = (1u << 15)
+                                   | ((u32)(emax + 256) & 0x1FFu);
+                    u64 myw = 0;
+#pragma unroll
+                    for (int pl = 0; pl < PLANES; ++pl) {
+                        u64 x = __ballot((u >> pl) & 1);
+                        if (lane == pl) myw = x;
+                    }
+                    if (lane < PLANES) s_planes[s][lane] = myw;
+                }
+            }
+        }
+        // ---- phase 1 (direct gather): each wave strides CGRP/nwaves
+        if (!PANEL && (PH & 1)) {
             long blk0 = base + wavei;
             int bk = (int)(blk0 % b2);
             long t = blk0 / b2;
