@@ -8,6 +8,8 @@
 // is one __ballot, and lane 0 serializes the group-tested bit stream.
 // Blocks are independent and fixed-size (rate*8 bytes), so encode and
 // decode are embarrassingly parallel across the tensor.
+#include <type_traits>
+
 #include "common.h"
 #include "kernels.h"
 
@@ -351,7 +353,7 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_encode_kernel(
 // parsers); phase 2 runs the inverse transform wave-per-block from LDS.
 // Truncation endgame differs from the per-bit reference only in zero
 // bits / internal n, so reconstruction is identical.
-template <bool BF16_OUT>
+template <bool BF16_OUT, bool PANEL = false>
 __global__ __launch_bounds__(CGRP, 2) void zfp_decode_kernel(
     const u32* __restrict__ wire, void* __restrict__ yv, int d0, int d1,
     int d2, int b0, int b1, int b2, int rate) {
@@ -365,6 +367,9 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_decode_kernel(
 
     __shared__ u64 s_planes[CGRP][PLANES + 1];  // +1: LDS bank pad
     __shared__ u32 s_hdr[CGRP];
+    using elem_t = typename std::conditional<BF16_OUT, bf16, float>::type;
+    constexpr int SLAB_W = BF16_OUT ? 66 : 65;
+    __shared__ elem_t s_slab[4][16][PANEL ? SLAB_W : 1];
 
     for (long base = (long)blockIdx.x * CGRP; base < nblocks;
          base += (long)gridDim.x * CGRP) {
@@ -420,7 +425,61 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_decode_kernel(
             }
         }
         __syncthreads();
-        // ---- phase 2: each wave reconstructs CGRP/nwaves blocks
+        // ---- phase 2 (panel): inverse of the encoder's staging — the
+        // wave reconstructs 16 consecutive blocks into the LDS slab,
+        // then writes 16 dense 64-value rows (the direct per-lane
+        // scatter wastes ~7/8 of every written cache line)
+        if (PANEL) {
+            for (int p = 0; p < 64 / 16; ++p) {
+                long pb = base + wavei * 64 + p * 16;
+                if (pb >= nblocks) break;
+                int npan = (int)(nblocks - pb < 16 ? nblocks - pb : 16);
+                int bk0 = (int)(pb % b2);
+                long t = pb / b2;
+                int bj = (int)(t % b1), bi = (int)(t / b1);
+                for (int b = 0; b < npan; ++b) {
+                    int s = wavei * 64 + p * 16 + b;
+                    u32 hdr = s_hdr[s];
+                    float outv = 0.f;
+                    if (hdr >> 15) {
+                        int emax = (int)(hdr & 0x1FFu) - 256;
+                        u32 u = 0;
+#pragma unroll
+                        for (int pl = 0; pl < PLANES; ++pl)
+                            u |= (u32)((s_planes[s][pl] >> lane) & 1)
+                                 << pl;
+                        int q = (int)((u ^ NBMASK) - NBMASK);
+                        q = __shfl(q, ZIPERM[lane]);
+                        q = inv_axis(q, lane, 16);
+                        q = inv_axis(q, lane, 4);
+                        q = inv_axis(q, lane, 1);
+                        outv = ldexpf((float)q, emax - QBITS);
+                    }
+                    if (BF16_OUT) {
+                        bf16 h = f2bf(outv);
+                        s_slab[wavei][lane >> 2][b * 4 + (lane & 3)] =
+                            *(elem_t*)&h;
+                    } else {
+                        s_slab[wavei][lane >> 2][b * 4 + (lane & 3)] =
+                            *(elem_t*)&outv;
+                    }
+                }
+                // dense row writes (partial rows predicated per lane)
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    int gi = bi * 4 + (r >> 2);
+                    int gj = bj * 4 + (r & 3);
+                    if (gi >= d0 || gj >= d1) continue;
+                    int gk = bk0 * 4 + lane;
+                    if (gk >= d2 || gk >= (bk0 + npan) * 4) continue;
+                    long idx = ((long)gi * d1 + gj) * d2 + gk;
+                    ((elem_t*)yv)[idx] = s_slab[wavei][r][lane];
+                }
+            }
+            __syncthreads();           // LDS reused next round
+            continue;
+        }
+        // ---- phase 2 (direct): each wave reconstructs CGRP/nwaves blocks
         for (int s = wavei; s < CGRP; s += nwaves) {
             long blk = base + s;
             if (blk >= nblocks) break;
@@ -472,40 +531,52 @@ void launch_zfp_encode(const void* x, void* out, bool bf16_in, int d0,
                        int phases) {
     int b0 = (d0 + 3) / 4, b1 = (d1 + 3) / 4, b2 = (d2 + 3) / 4;
     long nblocks = (long)b0 * b1 * b2;
+    bool panel = (b2 % 16 == 0);      // NHWC boundary tensors: C >= 64
     dim3 grid(codec_grid(nblocks, 256)), block(256);
+#define ENC_1(BF, PH, PAN)                                                 \
+    hipLaunchKernelGGL((zfp_encode_kernel<BF, PH, PAN>), grid, block, 0,   \
+                       s, x, (u32*)out, d0, d1, d2, b0, b1, b2, rate)
 #define ENC_DISPATCH(BF)                                                   \
     do {                                                                   \
-        if (phases == 1)                                                   \
-            hipLaunchKernelGGL((zfp_encode_kernel<BF, 1>), grid, block, 0, \
-                               s, x, (u32*)out, d0, d1, d2, b0, b1, b2,    \
-                               rate);                                      \
-        else if (phases == 2)                                              \
-            hipLaunchKernelGGL((zfp_encode_kernel<BF, 2>), grid, block, 0, \
-                               s, x, (u32*)out, d0, d1, d2, b0, b1, b2,    \
-                               rate);                                      \
-        else                                                               \
-            hipLaunchKernelGGL((zfp_encode_kernel<BF, 3>), grid, block, 0, \
-                               s, x, (u32*)out, d0, d1, d2, b0, b1, b2,    \
-                               rate);                                      \
+        if (phases == 1) {                                                 \
+            if (panel) ENC_1(BF, 1, true); else ENC_1(BF, 1, false);       \
+        } else if (phases == 2) {                                          \
+            if (panel) ENC_1(BF, 2, true); else ENC_1(BF, 2, false);       \
+        } else {                                                           \
+            if (panel) ENC_1(BF, 3, true); else ENC_1(BF, 3, false);       \
+        }                                                                  \
     } while (0)
     if (bf16_in) ENC_DISPATCH(true);
     else ENC_DISPATCH(false);
 #undef ENC_DISPATCH
+#undef ENC_1
 }
 
 void launch_zfp_decode(const void* wire, void* y, bool bf16_out, int d0,
                        int d1, int d2, int rate, hipStream_t s) {
     int b0 = (d0 + 3) / 4, b1 = (d1 + 3) / 4, b2 = (d2 + 3) / 4;
     long nblocks = (long)b0 * b1 * b2;
+    bool panel = (b2 % 16 == 0);
     dim3 grid(codec_grid(nblocks, 256)), block(256);
-    if (bf16_out)
-        hipLaunchKernelGGL((zfp_decode_kernel<true>), grid, block, 0, s,
-                           (const u32*)wire, y, d0, d1, d2, b0, b1, b2,
-                           rate);
-    else
-        hipLaunchKernelGGL((zfp_decode_kernel<false>), grid, block, 0, s,
-                           (const u32*)wire, y, d0, d1, d2, b0, b1, b2,
-                           rate);
+    if (bf16_out) {
+        if (panel)
+            hipLaunchKernelGGL((zfp_decode_kernel<true, true>), grid,
+                               block, 0, s, (const u32*)wire, y, d0, d1,
+                               d2, b0, b1, b2, rate);
+        else
+            hipLaunchKernelGGL((zfp_decode_kernel<true, false>), grid,
+                               block, 0, s, (const u32*)wire, y, d0, d1,
+                               d2, b0, b1, b2, rate);
+    } else {
+        if (panel)
+            hipLaunchKernelGGL((zfp_decode_kernel<false, true>), grid,
+                               block, 0, s, (const u32*)wire, y, d0, d1,
+                               d2, b0, b1, b2, rate);
+        else
+            hipLaunchKernelGGL((zfp_decode_kernel<false, false>), grid,
+                               block, 0, s, (const u32*)wire, y, d0, d1,
+                               d2, b0, b1, b2, rate);
+    }
 }
 
 }  // namespace defer_hip

@@ -58,7 +58,12 @@ def test_gpu_encode_bitexact_vs_reference():
         pytest.skip("no GPU")
     rng = np.random.default_rng(2)
     for shape, rate in [((8, 16, 32), 8), ((2, 6, 10, 24), 12),
-                        ((64, 1000), 8), ((5, 5, 7), 6)]:
+                        ((64, 1000), 8), ((5, 5, 7), 6),
+                        # panel-staged path (b2 % 16 == 0, C >= 64):
+                        # dense-row LDS staging must stay bit-exact,
+                        # incl. d0/d1 edge clamping and d2 tails
+                        ((4, 16, 64), 8), ((2, 8, 8, 256), 8),
+                        ((3, 7, 5, 64), 10), ((2, 6, 6, 253), 12)]:
         a = (rng.standard_normal(shape) * 3).astype(np.float32)
         want = zfp_ref.encode(a, rate)
         xg = torch.from_numpy(a).cuda()
