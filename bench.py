@@ -58,11 +58,17 @@ def parse_args(argv=None):
                     help="capture stage forwards in hipGraphs")
     ap.add_argument("--device", default="cuda", choices=["cuda", "cpu"])
     ap.add_argument("--ring-depth", type=int, default=4)
-    ap.add_argument("--dual-rail", action="store_true",
+    ap.add_argument("--dual-rail", dest="dual_rail", default=None,
+                    action="store_true",
                     help="split each fixed-size hop across two xGMI "
                          "rails (direct + via an idle third GPU); "
                          "doubles effective hop bandwidth for "
-                         "boundaries above one link's 153 GB/s")
+                         "boundaries above one link's 153 GB/s. "
+                         "Default: ON at world >= 4 (every hop dual, "
+                         "the xGMI-topology-aware design point), OFF "
+                         "below")
+    ap.add_argument("--no-dual-rail", dest="dual_rail",
+                    action="store_false")
     ap.add_argument("--no-return-results", action="store_true")
     ap.add_argument("--latency-items", type=int, default=32,
                     help="serialized items for the post-run latency "
@@ -130,6 +136,11 @@ def worker(args):
     if B % M:
         raise SystemExit(f"--batch {B} not divisible by --micro-batch {M}")
     per_step = B // M
+    if args.dual_rail is None:
+        # default: dual-rail when EVERY hop can run dual (world >= 4);
+        # the auto-partitioner's bandwidth credit follows via
+        # comm.dual_bw_boost
+        args.dual_rail = world >= 4 and args.compression != "zfp+lz4"
 
     cfg = PipelineConfig(
         partition_layers=cuts, num_stages=world, device=args.device,

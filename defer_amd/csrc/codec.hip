@@ -214,6 +214,10 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_encode_kernel(
 #pragma unroll
                     for (int off = 32; off > 0; off >>= 1)
                         av = fmaxf(av, __shfl_xor(av, off));
+                    if (PH == 4) {       // perf bisect: stage+reduce only
+                        if (av == 12345.678f) s_hdr[s] = 1;
+                        continue;
+                    }
                     if (!(av > 0.f) || !isfinite(av)) {
                         if (lane == 0) s_hdr[s] = 0;
                         continue;
@@ -226,6 +230,10 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_encode_kernel(
                     q = fwd_axis(q, lane, 16);
                     q = __shfl(q, ZPERM[lane]);
                     u32 u = ((u32)q + NBMASK) ^ NBMASK;
+                    if (PH == 5) {       // perf bisect: + lift, no ballots
+                        if (u == 0xdeadbeefu) s_hdr[s] = u;
+                        continue;
+                    }
                     if (lane == 0)
                         s_hdr[s] = (1u << 15)
                                    | ((u32)(emax + 256) & 0x1FFu);
@@ -540,6 +548,10 @@ void launch_zfp_encode(const void* x, void* out, bool bf16_in, int d0,
     do {                                                                   \
         if (phases == 1) {                                                 \
             if (panel) ENC_1(BF, 1, true); else ENC_1(BF, 1, false);       \
+        } else if (phases == 4 && panel) {   /* bisect: stage+reduce */    \
+            ENC_1(BF, 4, true);                                            \
+        } else if (phases == 5 && panel) {   /* bisect: + lift */          \
+            ENC_1(BF, 5, true);                                            \
         } else if (phases == 2) {                                          \
             if (panel) ENC_1(BF, 2, true); else ENC_1(BF, 2, false);       \
         } else {                                                           \

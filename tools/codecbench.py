@@ -31,7 +31,7 @@ def bench(shape, rate, iters=30):
     dec_us = t1.elapsed_time(t2) * 1e3 / iters
     # phase bisection: transform-only / serialize-only encodes
     ph = {}
-    for phase in (1, 2):
+    for phase in (1, 2, 4, 5):
         torch.cuda.synchronize()
         a, b = torch.cuda.Event(True), torch.cuda.Event(True)
         a.record()
@@ -48,4 +48,4 @@ for name, shape in SHAPES:
         e, d, mb, wmb, ph = bench(shape, rate)
         print(f"{name:20s} rate{rate:2d}: enc {e:7.1f} us ({mb/e*1e3:6.0f} GB/s) "
               f"dec {d:7.1f} us ({mb/d*1e3:6.0f} GB/s)  {mb:.0f}->{wmb:.0f} MB"
-              f"  [p1 {ph[1]:.1f}us p2 {ph[2]:.1f}us]")
+              f"  [p1 {ph[1]:.1f}us p2 {ph[2]:.1f}us stage {ph[4]:.1f}us lift {ph[5]:.1f}us]")
