@@ -65,6 +65,30 @@ __device__ __forceinline__ int inv_axis(int q, int lane, int s) {
     return q;
 }
 
+// ---- 64x64 bit-matrix transpose across the wave -------------------------
+// lane L holds row L (u64); returns column L. Six butterfly steps, each
+// swapping bit j of the row index with bit j of the bit index (steps
+// commute). Replaces the 30x __ballot plane loop of the encoder (and
+// the 30x ds_read_b64 gather of the decoder): ~85% of the transform
+// phase was that loop (profiles/zfp_codec_throughput.txt bisect).
+__device__ __forceinline__ u64 bit_transpose64(u64 x, int lane) {
+    static const u64 M[6] = {
+        0x00000000FFFFFFFFull, 0x0000FFFF0000FFFFull,
+        0x00FF00FF00FF00FFull, 0x0F0F0F0F0F0F0F0Full,
+        0x3333333333333333ull, 0x5555555555555555ull};
+#pragma unroll
+    for (int i = 0; i < 6; ++i) {
+        const int j = 32 >> i;
+        const u64 m = M[i];
+        u64 y = __shfl_xor(x, j);
+        if ((lane & j) == 0)
+            x ^= (((x >> j) ^ y) & m) << j;
+        else
+            x ^= ((y >> j) ^ x) & m;
+    }
+    return x;
+}
+
 // ---- lane-0 bit stream ---------------------------------------------------
 struct BitWriter {
     u32* out;       // current word pointer
@@ -237,12 +261,9 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_encode_kernel(
                     if (lane == 0)
                         s_hdr[s] = (1u << 15)
                                    | ((u32)(emax + 256) & 0x1FFu);
-                    u64 myw = 0;
-#pragma unroll
-                    for (int pl = 0; pl < PLANES; ++pl) {
-                        u64 x = __ballot((u >> pl) & 1);
-                        if (lane == pl) myw = x;
-                    }
+                    // butterfly transpose: lane p ends with plane p's
+                    // 64-value word (one coalesced 30-lane ds_write)
+                    u64 myw = bit_transpose64((u64)u, lane);
                     if (lane < PLANES) s_planes[s][lane] = myw;
                 }
             }
@@ -304,16 +325,12 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_encode_kernel(
                 u32 u = ((u32)q + NBMASK) ^ NBMASK;
                 if (lane == 0)
                     s_hdr[s] = (1u << 15) | ((u32)(emax + 256) & 0x1FFu);
-                // lane p collects plane p's ballot word; ONE coalesced
-                // 30-lane ds_write replaces 30 single-lane writes (the
-                // single-lane stores serialized the whole phase on the
-                // LDS pipe: measured 725us -> see profiles/README.md)
-                u64 myw = 0;
-#pragma unroll
-                for (int p = 0; p < PLANES; ++p) {
-                    u64 x = __ballot((u >> p) & 1);
-                    if (lane == p) myw = x;
-                }
+                // butterfly transpose: lane p ends with plane p's word;
+                // ONE coalesced 30-lane ds_write (the 30x __ballot loop
+                // this replaces was ~85% of the transform phase, and
+                // before that 30 single-lane stores serialized it on
+                // the LDS pipe — both measured, profiles/README.md)
+                u64 myw = bit_transpose64((u64)u, lane);
                 if (lane < PLANES) s_planes[s][lane] = myw;
             }
         }
@@ -451,11 +468,10 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_decode_kernel(
                     float outv = 0.f;
                     if (hdr >> 15) {
                         int emax = (int)(hdr & 0x1FFu) - 256;
-                        u32 u = 0;
-#pragma unroll
-                        for (int pl = 0; pl < PLANES; ++pl)
-                            u |= (u32)((s_planes[s][pl] >> lane) & 1)
-                                 << pl;
+                        // one ds_read_b64 per lane + butterfly back
+                        // (inverse of the encoder's transpose)
+                        u64 w = (lane < PLANES) ? s_planes[s][lane] : 0;
+                        u32 u = (u32)bit_transpose64(w, lane);
                         int q = (int)((u ^ NBMASK) - NBMASK);
                         q = __shfl(q, ZIPERM[lane]);
                         q = inv_axis(q, lane, 16);
@@ -504,10 +520,8 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_decode_kernel(
             float outv = 0.f;
             if (hdr >> 15) {
                 int emax = (int)(hdr & 0x1FFu) - 256;
-                u32 u = 0;
-#pragma unroll
-                for (int p = 0; p < PLANES; ++p)
-                    u |= (u32)((s_planes[s][p] >> lane) & 1) << p;
+                u64 w = (lane < PLANES) ? s_planes[s][lane] : 0;
+                u32 u = (u32)bit_transpose64(w, lane);
                 int q = (int)((u ^ NBMASK) - NBMASK);  // negabinary inv
                 q = __shfl(q, ZIPERM[lane]);
                 q = inv_axis(q, lane, 16);
