@@ -28,9 +28,16 @@ def main():
     from defer_amd.parallel.partitioner import (XGMI_LINK_GBPS,
                                                 auto_partition, node_times)
 
+    from defer_amd.parallel.calibrate import find_calibration
+
     m = MODELS[args.model]()
     g = m.graph
     _, ob, tu = node_times(g, (1, 224, 224, 3))
+    cal = find_calibration(m.model_name)
+    if cal:
+        tu = {n: cal.get(n, v) for n, v in tu.items()}
+        print(f"(using measured calibration defer_amd/calib/"
+              f"{m.model_name}.json)")
     names = [n.name for n in g.nodes]
     pos = {nm: i for i, nm in enumerate(names)}
     t = [tu[nm] for nm in names]
@@ -40,7 +47,8 @@ def main():
     for ns in (int(s) for s in args.stages.split(",")):
         mult = 2.0 if args.dual_rail and ns >= 4 else 1.0
         cuts, _ = auto_partition(g, ns,
-                                 link_gbps=XGMI_LINK_GBPS * mult)
+                                 link_gbps=XGMI_LINK_GBPS * mult,
+                                 measured_us=cal)
         bounds = sorted(pos[c] for c in cuts) + [len(names) - 1]
         st, start = [], 0
         for e in bounds:
