@@ -33,11 +33,12 @@ import torch
 def parse_args(argv=None):
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=400,
+    ap.add_argument("--steps", type=int, default=800,
                     help="timed steps; one step = one --batch global batch "
-                         "(default sized so the timed region is ~2.5 s at "
-                         "1 GPU and longer at N>1: rocm-smi sampling can "
-                         "see the GPU busy)")
+                         "(default sized so the timed region is >=5 s at "
+                         "1 GPU: the driver's rocm-smi sampling can see "
+                         "the GPU busy, and ms_per_step is robust to "
+                         "clock ramp)")
     ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--model", default="resnet50",
                     choices=["resnet50", "resnet101", "resnet152",
