@@ -541,7 +541,105 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_decode_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// fp8 (e4m3fn) wire codec: amax-scaled cast, 1 B/value + 4-byte scale.
+// Wire layout (identical to the torch fallback in parallel/comm.py):
+// bytes [0,4) = fp32 amax bits, bytes [4, 4+n) = e4m3 of x*448/amax.
+// Three tiny kernels (init + atomic amax + cast) replace the torch
+// path's ~5 dispatches and its extra fp32 materialization pass; the
+// scale never visits the host.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ u8 f32_to_e4m3(float f) {
+    u32 b = __float_as_uint(f);
+    u8 sign = (u8)((b >> 24) & 0x80);
+    u32 ab = b & 0x7fffffffu;
+    if (ab > 0x7f800000u) return sign | 0x7f;          // NaN
+    if (ab < 0x3C800000u) {                            // |x| < 2^-6
+        // subnormal: m = rne(|x| * 2^9); m == 8 lands exactly on the
+        // min-normal encoding, so the boundary needs no special case
+        u32 m = (u32)rintf(__uint_as_float(ab) * 512.f);
+        return sign | (u8)m;
+    }
+    // normal: RNE to 3 mantissa bits in the integer domain
+    u32 r = ab + 0x0007FFFFu + ((ab >> 20) & 1u);
+    u32 out = (r >> 20) - ((127u - 7u) << 3);
+    if (out > 0x7e) out = 0x7e;                        // saturate to 448
+    return sign | (u8)out;
+}
+
+__device__ __forceinline__ float e4m3_to_f32(u8 v) {
+    u32 sign = (u32)(v & 0x80) << 24;
+    u32 exp = (v >> 3) & 0xf;
+    u32 man = v & 0x7;
+    if (exp == 0xf && man == 0x7)
+        return __uint_as_float(sign | 0x7fc00000u);    // NaN
+    float m;
+    if (exp == 0)
+        m = (float)man * (1.f / 512.f);                // subnormal 2^-9
+    else
+        m = ldexpf(8.f + (float)man, (int)exp - 10);   // (1+man/8)*2^(e-7)
+    return (v & 0x80) ? -m : m;
+}
+
+__global__ void fp8_init_kernel(float* amax) { *amax = 1e-12f; }
+
+__global__ void fp8_amax_kernel(const bf16* __restrict__ x, long n,
+                                float* __restrict__ amax) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    float m = 0.f;
+    for (; i < n; i += (long)gridDim.x * blockDim.x) {
+        float v = fabsf(bf2f(x[i]));
+        if (isfinite(v)) m = fmaxf(m, v);
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        m = fmaxf(m, __shfl_xor(m, off));
+    // non-negative floats compare correctly as uints
+    if ((threadIdx.x % WAVE) == 0)
+        atomicMax((unsigned*)amax, __float_as_uint(m));
+}
+
+__global__ void fp8_cast_kernel(const bf16* __restrict__ x, long n,
+                                const float* __restrict__ amax,
+                                u8* __restrict__ out) {
+    float s = 448.f / *amax;
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    for (; i < n; i += (long)gridDim.x * blockDim.x)
+        out[i] = f32_to_e4m3(bf2f(x[i]) * s);
+}
+
+__global__ void fp8_decode_kernel(const u8* __restrict__ wire, long n,
+                                  const float* __restrict__ amax,
+                                  bf16* __restrict__ y) {
+    float s = *amax / 448.f;
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    for (; i < n; i += (long)gridDim.x * blockDim.x)
+        y[i] = f2bf(e4m3_to_f32(wire[i]) * s);
+}
+
+// ---------------------------------------------------------------------------
 namespace defer_hip {
+
+void launch_fp8_encode(const void* x, long n, void* out, hipStream_t s) {
+    float* amax = (float*)out;                 // wire bytes [0, 4)
+    u8* payload = (u8*)out + 4;
+    hipLaunchKernelGGL(fp8_init_kernel, dim3(1), dim3(1), 0, s, amax);
+    int grid = (int)((n + 255) / 256);
+    if (grid > 4096) grid = 4096;
+    hipLaunchKernelGGL(fp8_amax_kernel, dim3(grid), dim3(256), 0, s,
+                       (const bf16*)x, n, amax);
+    hipLaunchKernelGGL(fp8_cast_kernel, dim3(grid), dim3(256), 0, s,
+                       (const bf16*)x, n, amax, payload);
+}
+
+void launch_fp8_decode(const void* wire, long n, void* y, hipStream_t s) {
+    const float* amax = (const float*)wire;
+    const u8* payload = (const u8*)wire + 4;
+    int grid = (int)((n + 255) / 256);
+    if (grid > 4096) grid = 4096;
+    hipLaunchKernelGGL(fp8_decode_kernel, dim3(grid), dim3(256), 0, s,
+                       payload, n, amax, (bf16*)y);
+}
 
 static int codec_grid(long nblocks, int blocks_per_wg) {
     long wgs = (nblocks + blocks_per_wg - 1) / blocks_per_wg;
