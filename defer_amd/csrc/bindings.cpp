@@ -368,6 +368,29 @@ Tensor zfp_decode(Tensor wire, std::vector<int64_t> shape, int64_t rate,
     return o;
 }
 
+// fp8 e4m3fn wire: one fused HIP path (init + atomic amax + cast, the
+// scale never visits the host) replacing the torch fallback's ~5
+// dispatches. Wire layout matches comm.Codec exactly.
+void fp8_encode(Tensor x, Tensor out) {
+    check_bf16(x, "x");
+    long n = x.numel();
+    TORCH_CHECK(out.is_cuda() && out.is_contiguous()
+                && out.scalar_type() == at::kByte
+                && out.numel() == n + 4, "bad fp8 wire buffer");
+    defer_hip::launch_fp8_encode(bptr(x), n, bptr_mut(out),
+                                 cur_stream());
+}
+
+Tensor fp8_decode(Tensor wire, std::vector<int64_t> shape) {
+    TORCH_CHECK(wire.is_cuda() && wire.is_contiguous()
+                && wire.scalar_type() == at::kByte, "bad wire");
+    auto y = at::empty(shape, wire.options().dtype(at::kBFloat16));
+    TORCH_CHECK(y.numel() + 4 == wire.numel(), "wire/shape mismatch");
+    defer_hip::launch_fp8_decode(bptr(wire), y.numel(), bptr_mut(y),
+                                 cur_stream());
+    return y;
+}
+
 Tensor lz4_compress(Tensor x) {
     TORCH_CHECK(x.is_cuda() && x.is_contiguous()
                 && x.scalar_type() == at::kByte, "x must be GPU u8 contig");
@@ -439,6 +462,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("zfp_encode", &zfp_encode, py::arg("x"), py::arg("rate"),
           py::arg("out") = py::none(), py::arg("phases") = 3);
     m.def("zfp_decode", &zfp_decode);
+    m.def("fp8_encode", &fp8_encode);
+    m.def("fp8_decode", &fp8_decode);
     m.def("lz4_compress", &lz4_compress);
     m.def("lz4_compress_into", &lz4_compress_into);
     m.def("lz4_scratch_bytes",

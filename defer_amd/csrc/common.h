@@ -15,6 +15,7 @@ typedef __bf16 bf16x4 __attribute__((ext_vector_type(4)));
 typedef __bf16 bf16x2 __attribute__((ext_vector_type(2)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 typedef float f32x16 __attribute__((ext_vector_type(16)));
+typedef unsigned char u8;
 typedef unsigned int u32;
 typedef unsigned long long u64;
 typedef short s16x8 __attribute__((ext_vector_type(8)));

@@ -65,6 +65,11 @@ void launch_zfp_encode(const void* x, void* out, bool bf16_in, int d0,
 void launch_zfp_decode(const void* wire, void* y, bool bf16_out, int d0,
                        int d1, int d2, int rate, hipStream_t s);
 
+// fp8 e4m3fn wire codec: [4-byte fp32 amax][n bytes e4m3 of x*448/amax]
+// (bit-compatible with the torch fallback in parallel/comm.py)
+void launch_fp8_encode(const void* x, long n, void* out, hipStream_t s);
+void launch_fp8_decode(const void* wire, long n, void* y, hipStream_t s);
+
 // LZ4-style block compressor (see csrc/lz4.hip / ops/lz4_ref.py)
 long lz4_max_compressed(long n);
 long lz4_scratch_bytes(long n);

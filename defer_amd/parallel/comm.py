@@ -71,6 +71,16 @@ class Codec:
         if self.mode == "fp8":
             if out is None:
                 out = self.alloc_wire()
+            if x.is_cuda and x.dtype == torch.bfloat16:
+                # fused HIP path (init + atomic amax + cast); wire
+                # layout identical to the torch fallback below
+                from defer_amd import ops as _ops
+
+                m = _ops._load_hip()
+                if m is None:
+                    raise RuntimeError("HIP extension missing for fp8")
+                m.fp8_encode(x.contiguous(), out)
+                return out
             amax = x.detach().abs().amax().float().clamp_min(1e-12)
             q = (x.float() * (448.0 / amax)).to(torch.float8_e4m3fn)
             out[:4].copy_(amax.reshape(1).view(torch.uint8))
@@ -85,6 +95,14 @@ class Codec:
         if self.mode == "none":
             return wire.view(self.shape)
         if self.mode == "fp8":
+            if wire.is_cuda and self.dtype == torch.bfloat16:
+                from defer_amd import ops as _ops
+
+                m = _ops._load_hip()
+                if m is None:
+                    raise RuntimeError("HIP extension missing for fp8")
+                return m.fp8_decode(wire.contiguous(),
+                                    list(self.shape))
             amax = wire[:4].view(torch.float32)
             vals = wire[4:].view(torch.float8_e4m3fn).to(torch.float32)
             return (vals * (amax / 448.0)).to(self.dtype) \

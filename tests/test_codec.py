@@ -243,3 +243,45 @@ def test_fp8_wire_codec_property_sweep():
         assert torch.equal(y == 0, x == 0) or (x == 0).sum() == 0
 
     check()
+
+
+@pytest.mark.gpu
+def test_fp8_wire_hip_matches_torch_cast():
+    """The fused HIP fp8 wire (csrc/codec.hip fp8_encode/decode) must be
+    bit-identical to the torch-fallback wire in parallel/comm.py: same
+    amax bytes, same e4m3fn payload (RNE + saturate-to-448 + subnormals),
+    and the decode must reproduce torch's dequantization in bf16."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    import defer_amd.ops as _ops
+
+    m = _ops._load_hip()
+    assert m is not None
+    torch.manual_seed(3)
+    cases = [
+        torch.randn(4096) * 10,
+        torch.randn(2, 56, 56, 64) * 0.01,          # subnormal-heavy
+        torch.linspace(-600, 600, 8192),            # saturation range
+        torch.zeros(256),
+        torch.full((64,), 4.375e-3),                # near min-normal/8
+    ]
+    for x in cases:
+        xg = x.to("cuda", torch.bfloat16).contiguous()
+        n = xg.numel()
+        out = torch.empty(n + 4, dtype=torch.uint8, device="cuda")
+        m.fp8_encode(xg, out)
+        # torch-fallback wire
+        amax = xg.detach().abs().amax().float().clamp_min(1e-12)
+        q = (xg.float() * (448.0 / amax)).to(torch.float8_e4m3fn)
+        want = torch.empty_like(out)
+        want[:4] = amax.reshape(1).view(torch.uint8)
+        want[4:] = q.view(torch.uint8).reshape(-1)
+        mism = (out != want).sum().item()
+        assert mism == 0, (
+            f"{mism}/{n} byte mismatches; first "
+            f"{(out != want).nonzero()[:5].flatten().tolist()}")
+        dec = m.fp8_decode(out, list(xg.shape))
+        vals = want[4:].view(torch.float8_e4m3fn).to(torch.float32)
+        want_dec = (vals * (amax / 448.0)).to(torch.bfloat16) \
+            .view(xg.shape)
+        assert torch.equal(dec, want_dec)
