@@ -29,7 +29,7 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--model", default="resnet50",
                     choices=["resnet50", "resnet101", "resnet152",
-                             "vgg19"])
+                             "vgg19", "vgg19_gap"])
     ap.add_argument("--batch", type=int, default=64)
     ap.add_argument("--cuts", default="auto")
     ap.add_argument("--items", type=int, default=0,
@@ -40,6 +40,10 @@ def main():
                     choices=["none", "fp8", "zfp", "zfp+lz4"])
     ap.add_argument("--device", default="cuda", choices=["cuda", "cpu"])
     ap.add_argument("--dual-rail", action="store_true")
+    ap.add_argument("--calibration", default=None,
+                    help="per-layer measured-cost file for auto cuts "
+                         "(tools/calibrate.py; default: the in-tree "
+                         "defer_amd/calib profile for the model)")
     ap.add_argument("--weights-dir", default=None,
                     help="per-stage checkpoint dir (checkpoint."
                          "save_stages); cuts come from its manifest")
@@ -75,6 +79,7 @@ def main():
         batch_size=args.batch, use_hip_graphs=False,
         compression=args.compression, dual_rail=args.dual_rail,
         weights_dir=args.weights_dir,
+        calibration_file=args.calibration,
         backend=backend, return_results=True)
     dev = (torch.device("cuda", local_rank) if args.device == "cuda"
            else torch.device("cpu"))

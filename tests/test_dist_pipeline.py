@@ -430,3 +430,25 @@ def test_var_ring_deferred_interleave_gloo():
     assert len(errs) == steps
     for e in errs:
         assert e < 1e-6, f"deferred interleave corrupted payloads: {e}"
+
+
+def test_bench_self_spawn_two_ranks_cpu():
+    """The driver's bare `bench.py --gpus N` form: bench self-spawns N
+    ranks via mp.spawn (no torchrun) and prints one JSON line."""
+    import json
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, "bench.py"), "--gpus", "2",
+         "--device", "cpu", "--batch", "4", "--micro-batch", "2",
+         "--steps", "2", "--warmup", "1", "--latency-items", "1"],
+        capture_output=True, text=True, timeout=420, cwd=root)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [ln for ln in out.stdout.splitlines() if ln.startswith("{")]
+    assert len(line) == 1, out.stdout
+    d = json.loads(line[0])
+    assert d["n_gpus"] == 2 and d["steps"] == 2
+    assert d["config"]["parallelism"] == "pp2"
+    assert d["latency_ms"]["items"] == 1
