@@ -602,10 +602,12 @@ __global__ void fp8_amax_kernel(const bf16* __restrict__ x, long n,
 __global__ void fp8_cast_kernel(const bf16* __restrict__ x, long n,
                                 const float* __restrict__ amax,
                                 u8* __restrict__ out) {
-    // correctly-rounded divide: the default v_rcp sequence is 1-2 ulp
-    // off, which moves boundary values across the e4m3 rounding point
-    // (measured: 4.9% byte mismatches vs the torch wire)
-    float s = __fdiv_rn(448.f, *amax);
+    // correctly-rounded divide: fp32 '/' (and __fdiv_rn) lower to the
+    // ~1-ulp v_rcp sequence here, which moved boundary values across
+    // the e4m3 rounding point (measured: 4.9% byte mismatches vs the
+    // torch wire). Device double division IS correctly rounded; the
+    // once-per-tensor scale can afford it.
+    float s = (float)(448.0 / (double)*amax);
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     for (; i < n; i += (long)gridDim.x * blockDim.x)
         out[i] = f32_to_e4m3(bf2f(x[i]) * s);
@@ -614,7 +616,7 @@ __global__ void fp8_cast_kernel(const bf16* __restrict__ x, long n,
 __global__ void fp8_decode_kernel(const u8* __restrict__ wire, long n,
                                   const float* __restrict__ amax,
                                   bf16* __restrict__ y) {
-    float s = __fdiv_rn(*amax, 448.f);
+    float s = (float)((double)*amax / 448.0);
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     for (; i < n; i += (long)gridDim.x * blockDim.x)
         y[i] = f2bf(e4m3_to_f32(wire[i]) * s);
