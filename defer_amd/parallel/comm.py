@@ -82,7 +82,11 @@ class Codec:
                 m.fp8_encode(x.contiguous(), out)
                 return out
             amax = x.detach().abs().amax().float().clamp_min(1e-12)
-            q = (x.float() * (448.0 / amax)).to(torch.float8_e4m3fn)
+            # scale via double division then fp32: torch's scalar/tensor
+            # 448.0/amax lowers to reciprocal-multiply (1 ulp off true
+            # RN); this form is bit-identical to the HIP kernel's
+            scale = (448.0 / amax.double()).float()
+            q = (x.float() * scale).to(torch.float8_e4m3fn)
             out[:4].copy_(amax.reshape(1).view(torch.uint8))
             out[4:].copy_(q.view(torch.uint8).reshape(-1))
             return out
@@ -105,7 +109,8 @@ class Codec:
                                     list(self.shape))
             amax = wire[:4].view(torch.float32)
             vals = wire[4:].view(torch.float8_e4m3fn).to(torch.float32)
-            return (vals * (amax / 448.0)).to(self.dtype) \
+            scale = (amax.double() / 448.0).float()
+            return (vals * scale).to(self.dtype) \
                 .view(self.shape)
         from defer_amd.ops import codec as zc
         if self.variable:

@@ -270,9 +270,10 @@ def test_fp8_wire_hip_matches_torch_cast():
         n = xg.numel()
         out = torch.empty(n + 4, dtype=torch.uint8, device="cuda")
         m.fp8_encode(xg, out)
-        # torch-fallback wire
+        # torch-fallback wire (double-division scale, see comm.Codec)
         amax = xg.detach().abs().amax().float().clamp_min(1e-12)
-        q = (xg.float() * (448.0 / amax)).to(torch.float8_e4m3fn)
+        q = (xg.float() * (448.0 / amax.double()).float()) \
+            .to(torch.float8_e4m3fn)
         want = torch.empty_like(out)
         want[:4] = amax.reshape(1).view(torch.uint8)
         want[4:] = q.view(torch.uint8).reshape(-1)
@@ -282,6 +283,6 @@ def test_fp8_wire_hip_matches_torch_cast():
             f"{(out != want).nonzero()[:5].flatten().tolist()}")
         dec = m.fp8_decode(out, list(xg.shape))
         vals = want[4:].view(torch.float8_e4m3fn).to(torch.float32)
-        want_dec = (vals * (amax / 448.0)).to(torch.bfloat16) \
-            .view(xg.shape)
+        want_dec = (vals * (amax.double() / 448.0).float()) \
+            .to(torch.bfloat16).view(xg.shape)
         assert torch.equal(dec, want_dec)
