@@ -43,6 +43,31 @@ def bench(shape, rate, iters=30):
     mb = x.numel() * 2 / 1e6
     return enc_us, dec_us, mb, w.numel() / 1e6, ph
 
+def bench_fp8(shape, iters=50):
+    x = torch.randn(*shape, device="cuda", dtype=torch.bfloat16)
+    n = x.numel()
+    w = torch.empty(n + 4, dtype=torch.uint8, device="cuda")
+    hip.fp8_encode(x, w)
+    y = hip.fp8_decode(w, list(shape))
+    torch.cuda.synchronize()
+    t0, t1, t2 = (torch.cuda.Event(True) for _ in range(3))
+    t0.record()
+    for _ in range(iters):
+        hip.fp8_encode(x, w)
+    t1.record()
+    for _ in range(iters):
+        y = hip.fp8_decode(w, list(shape))
+    t2.record()
+    torch.cuda.synchronize()
+    enc = t0.elapsed_time(t1) * 1e3 / iters
+    dec = t1.elapsed_time(t2) * 1e3 / iters
+    mb = n * 2 / 1e6
+    print(f"fp8  {str(shape):20s}: enc {enc:7.1f} us ({mb/enc*1e3:6.0f} GB/s) "
+          f"dec {dec:7.1f} us ({mb/dec*1e3:6.0f} GB/s)  {mb:.0f}->{(n+4)/1e6:.0f} MB")
+
+
+for name, shape in SHAPES:
+    bench_fp8(shape)
 for name, shape in SHAPES:
     for rate in (4, 8):
         e, d, mb, wmb, ph = bench(shape, rate)
