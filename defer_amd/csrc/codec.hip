@@ -594,9 +594,19 @@ __global__ void fp8_amax_kernel(const bf16* __restrict__ x, long n,
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1)
         m = fmaxf(m, __shfl_xor(m, off));
-    // non-negative floats compare correctly as uints
-    if ((threadIdx.x % WAVE) == 0)
+    // LDS-reduce the block's waves to ONE atomic per block: per-wave
+    // atomics to the single global word serialized at ~12 ns each —
+    // 16K of them put a ~200 us floor under encode (measured)
+    __shared__ float wmax[8];
+    const int wv = threadIdx.x / WAVE;
+    if ((threadIdx.x % WAVE) == 0) wmax[wv] = m;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        const int nw = blockDim.x / WAVE;
+        for (int w = 1; w < nw; ++w) m = fmaxf(m, wmax[w]);
+        // non-negative floats compare correctly as uints
         atomicMax((unsigned*)amax, __float_as_uint(m));
+    }
 }
 
 __global__ void fp8_cast_kernel(const bf16* __restrict__ x, long n,
@@ -630,9 +640,9 @@ void launch_fp8_encode(const void* x, long n, void* out, hipStream_t s) {
     u8* payload = (u8*)out + 4;
     hipLaunchKernelGGL(fp8_init_kernel, dim3(1), dim3(1), 0, s, amax);
     int grid = (int)((n + 255) / 256);
-    if (grid > 4096) grid = 4096;
+    if (grid > 2048) grid = 2048;   // one atomic per block after the
     hipLaunchKernelGGL(fp8_amax_kernel, dim3(grid), dim3(256), 0, s,
-                       (const bf16*)x, n, amax);
+                       (const bf16*)x, n, amax);   // LDS reduce
     hipLaunchKernelGGL(fp8_cast_kernel, dim3(grid), dim3(256), 0, s,
                        (const bf16*)x, n, amax, payload);
 }
