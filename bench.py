@@ -184,9 +184,56 @@ def worker(args):
     # (lazy per-pair init ordering is then irrelevant on RCCL)
     barrier_sync()
 
+    # ---- dual-rail stall watchdog around warmup+barrier. The dual-rail
+    # relay is gloo-proven but its first-ever RCCL execution is the
+    # round-end multi-GPU run itself; if an RCCL-specific stall shows up
+    # there it would show up in the warmup, so every rank re-execs
+    # itself single-rail (same argv + --no-dual-rail; fd 1 restored for
+    # the one-JSON-line contract; env rendezvous state survives exec)
+    # instead of hanging the whole scaling sweep.
+    import sys as _sys
+    import threading
+
+    warm_done = threading.Event()
+    wd_s = float(os.environ.get("DEFER_BENCH_WATCHDOG_S", "300"))
+    arm = args.dual_rail and world > 2 and \
+        (args.device == "cuda"
+         or os.environ.get("DEFER_BENCH_TEST_STALL") == "1")
+
+    def _watchdog():
+        if not warm_done.wait(timeout=wd_s):
+            print(f"[bench rank {rank}] dual-rail warmup stalled "
+                  f">{wd_s:.0f}s; re-exec single-rail", flush=True)
+            os.environ.pop("DEFER_BENCH_TEST_STALL", None)
+            # fresh rendezvous on a derived port: under torchrun the
+            # original store lives in the (still running) agent and a
+            # second init on it would collide on rendezvous keys; the
+            # re-exec'd rank 0 hosts the new store itself
+            os.environ["MASTER_ADDR"] = "127.0.0.1"
+            os.environ["MASTER_PORT"] = str(
+                int(os.environ.get("MASTER_PORT", "29500")) + 17)
+            # TORCHELASTIC_USE_AGENT_STORE=True would make every rank a
+            # store CLIENT on the new port (nobody hosts) — drop the
+            # torchrun-agent markers so rank 0 hosts the fresh store
+            for k in list(os.environ):
+                if k.startswith("TORCHELASTIC"):
+                    os.environ.pop(k)
+            os.dup2(real_stdout, 1)
+            argv = [a for a in _sys.argv[1:]
+                    if a not in ("--dual-rail", "--no-dual-rail")]
+            os.execv(_sys.executable,
+                     [_sys.executable, os.path.abspath(__file__)]
+                     + argv + ["--no-dual-rail"])
+
+    if arm:
+        threading.Thread(target=_watchdog, daemon=True).start()
+    if os.environ.get("DEFER_BENCH_TEST_STALL") == "1" and arm:
+        time.sleep(wd_s + 30)   # simulate a hung warmup (tests only)
+
     # ---- warmup (fills pipeline, triggers graph capture)
     pipe.run(max(args.warmup, 2) * per_step, feed=feed, collect=collect)
     barrier_sync()
+    warm_done.set()
     pipe.reset_stats()
 
     # ---- timed region: exactly --steps global batches
