@@ -228,7 +228,7 @@ def worker(args):
     if arm:
         threading.Thread(target=_watchdog, daemon=True).start()
     if os.environ.get("DEFER_BENCH_TEST_STALL") == "1" and arm:
-        time.sleep(wd_s + 30)   # simulate a hung warmup (tests only)
+        time.sleep(wd_s * 3 + 10)   # simulate a hung warmup (tests)
 
     # ---- warmup (fills pipeline, triggers graph capture)
     pipe.run(max(args.warmup, 2) * per_step, feed=feed, collect=collect)

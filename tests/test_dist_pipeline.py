@@ -452,3 +452,29 @@ def test_bench_self_spawn_two_ranks_cpu():
     assert d["n_gpus"] == 2 and d["steps"] == 2
     assert d["config"]["parallelism"] == "pp2"
     assert d["latency_ms"]["items"] == 1
+
+
+def test_bench_dual_rail_watchdog_fallback_cpu():
+    """If the dual-rail warmup ever stalls (the one RCCL-only unknown),
+    every rank re-execs itself single-rail on a fresh rendezvous port
+    instead of hanging the driver's scaling sweep. Simulated via the
+    DEFER_BENCH_TEST_STALL knob on gloo."""
+    import json
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, DEFER_BENCH_TEST_STALL="1",
+               DEFER_BENCH_WATCHDOG_S="2")
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, "bench.py"), "--gpus", "3",
+         "--device", "cpu", "--batch", "4", "--micro-batch", "2",
+         "--steps", "2", "--warmup", "1", "--latency-items", "0",
+         "--dual-rail"],
+        capture_output=True, text=True, timeout=600, cwd=root, env=env)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert out.stderr.count("re-exec single-rail") == 3
+    d = json.loads([ln for ln in out.stdout.splitlines()
+                    if ln.startswith("{")][0])
+    assert d["config"]["dual_rail"] is False   # fell back
+    assert d["n_gpus"] == 3
