@@ -228,6 +228,15 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_encode_kernel(
                     s_slab[wavei][r][lane] =
                         ((const elem_t*)xv)[idx];
                 }
+                // straightline block transform (writes predicated on
+                // `active` instead of early continues) unrolled 2x:
+                // the lift+transpose is a serial cross-lane dependency
+                // chain and the phase is latency-bound at 2 waves/SIMD,
+                // so two independent chains in flight per wave halve
+                // the exposed latency. Zero blocks waste a transform
+                // (uniform-rare in activation data) but stay bit-exact
+                // — their planes are simply not written.
+#pragma unroll 2
                 for (int b = 0; b < npan; ++b) {
                     int s = wavei * 64 + p * 16 + b;
                     elem_t raw = s_slab[wavei][lane >> 2]
@@ -242,10 +251,7 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_encode_kernel(
                         if (av == 12345.678f) s_hdr[s] = 1;
                         continue;
                     }
-                    if (!(av > 0.f) || !isfinite(av)) {
-                        if (lane == 0) s_hdr[s] = 0;
-                        continue;
-                    }
+                    const bool active = (av > 0.f) && isfinite(av);
                     int emax;
                     frexpf(av, &emax);
                     int q = (int)rintf(v * ldexpf(1.0f, QBITS - emax));
@@ -259,12 +265,13 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_encode_kernel(
                         continue;
                     }
                     if (lane == 0)
-                        s_hdr[s] = (1u << 15)
-                                   | ((u32)(emax + 256) & 0x1FFu);
+                        s_hdr[s] = active
+                            ? ((1u << 15) | ((u32)(emax + 256) & 0x1FFu))
+                            : 0u;
                     // butterfly transpose: lane p ends with plane p's
                     // 64-value word (one coalesced 30-lane ds_write)
                     u64 myw = bit_transpose64((u64)u, lane);
-                    if (lane < PLANES) s_planes[s][lane] = myw;
+                    if (active && lane < PLANES) s_planes[s][lane] = myw;
                 }
             }
         }
