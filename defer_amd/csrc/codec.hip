@@ -65,28 +65,45 @@ __device__ __forceinline__ int inv_axis(int q, int lane, int s) {
     return q;
 }
 
-// ---- 64x64 bit-matrix transpose across the wave -------------------------
-// lane L holds row L (u64); returns column L. Six butterfly steps, each
-// swapping bit j of the row index with bit j of the bit index (steps
-// commute). Replaces the 30x __ballot plane loop of the encoder (and
-// the 30x ds_read_b64 gather of the decoder): ~85% of the transform
-// phase was that loop (profiles/zfp_codec_throughput.txt bisect).
-__device__ __forceinline__ u64 bit_transpose64(u64 x, int lane) {
-    static const u64 M[6] = {
-        0x00000000FFFFFFFFull, 0x0000FFFF0000FFFFull,
-        0x00FF00FF00FF00FFull, 0x0F0F0F0F0F0F0F0Full,
-        0x3333333333333333ull, 0x5555555555555555ull};
+// ---- wave bit-plane transpose -------------------------------------------
+// Replaces the encoder's 30x __ballot plane-collect loop and the
+// decoder's 30x ds_read_b64 gather (measured ~85% of the transform
+// phase). The matrix is 64 values x 30
+// planes, so 32-bit rows suffice — five u32 butterfly steps transpose
+// the two 32x32 halves (lane-xor j < 32 stays inside each half) and one
+// cross-half shuffle stitches the u64 plane words. Roughly half the
+// VALU ops of the u64 version (the transform phase is issue-bound).
+__device__ __forceinline__ u32 bt32_steps(u32 x, int lane) {
+    static const u32 M5[5] = {0x0000FFFFu, 0x00FF00FFu, 0x0F0F0F0Fu,
+                              0x33333333u, 0x55555555u};
 #pragma unroll
-    for (int i = 0; i < 6; ++i) {
-        const int j = 32 >> i;
-        const u64 m = M[i];
-        u64 y = __shfl_xor(x, j);
+    for (int i = 0; i < 5; ++i) {
+        const int j = 16 >> i;
+        const u32 m = M5[i];
+        u32 y = __shfl_xor(x, j);
         if ((lane & j) == 0)
             x ^= (((x >> j) ^ y) & m) << j;
         else
             x ^= ((y >> j) ^ x) & m;
     }
     return x;
+}
+
+// encode: lane L holds u32 plane-bits of value L -> lanes p < 32 hold
+// the u64 word of plane p (bit L = value L's bit p); lanes >= 32 return
+// garbage (only lanes < PLANES store).
+__device__ __forceinline__ u64 planes_from_values(u32 u, int lane) {
+    u32 x = bt32_steps(u, lane);
+    u32 hi = __shfl_xor(x, 32);      // lane l < 32 reads lane l+32
+    return (u64)x | ((u64)hi << 32);
+}
+
+// decode: lanes p < 32 hold plane word w_p (lanes 30/31 zero) -> every
+// lane L gets its value's u32 plane-bits.
+__device__ __forceinline__ u32 values_from_planes(u64 w, int lane) {
+    u64 other = __shfl_xor(w, 32);
+    u32 x = (lane < 32) ? (u32)w : (u32)(other >> 32);
+    return bt32_steps(x, lane);
 }
 
 // ---- lane-0 bit stream ---------------------------------------------------
@@ -270,7 +287,7 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_encode_kernel(
                             : 0u;
                     // butterfly transpose: lane p ends with plane p's
                     // 64-value word (one coalesced 30-lane ds_write)
-                    u64 myw = bit_transpose64((u64)u, lane);
+                    u64 myw = planes_from_values(u, lane);
                     if (active && lane < PLANES) s_planes[s][lane] = myw;
                 }
             }
@@ -337,7 +354,7 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_encode_kernel(
                 // this replaces was ~85% of the transform phase, and
                 // before that 30 single-lane stores serialized it on
                 // the LDS pipe — both measured, profiles/README.md)
-                u64 myw = bit_transpose64((u64)u, lane);
+                u64 myw = planes_from_values(u, lane);
                 if (lane < PLANES) s_planes[s][lane] = myw;
             }
         }
@@ -478,7 +495,7 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_decode_kernel(
                         // one ds_read_b64 per lane + butterfly back
                         // (inverse of the encoder's transpose)
                         u64 w = (lane < PLANES) ? s_planes[s][lane] : 0;
-                        u32 u = (u32)bit_transpose64(w, lane);
+                        u32 u = values_from_planes(w, lane);
                         int q = (int)((u ^ NBMASK) - NBMASK);
                         q = __shfl(q, ZIPERM[lane]);
                         q = inv_axis(q, lane, 16);
@@ -528,7 +545,7 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_decode_kernel(
             if (hdr >> 15) {
                 int emax = (int)(hdr & 0x1FFu) - 256;
                 u64 w = (lane < PLANES) ? s_planes[s][lane] : 0;
-                u32 u = (u32)bit_transpose64(w, lane);
+                u32 u = values_from_planes(w, lane);
                 int q = (int)((u ^ NBMASK) - NBMASK);  // negabinary inv
                 q = __shfl(q, ZIPERM[lane]);
                 q = inv_axis(q, lane, 16);
