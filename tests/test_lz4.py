@@ -120,3 +120,69 @@ def test_spec_roundtrip_property_sweep():
         assert np.array_equal(out, data)
 
     check()
+
+
+@pytest.mark.gpu
+def test_lz4_compress_into_matches_lz4_compress():
+    """The async hop encode (lz4_compress_into: preallocated out +
+    device-side wire length, comm.VarP2PRing's CUDA path) must produce
+    byte-identical streams to the allocating lz4_compress binding."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    import defer_amd.ops as _ops
+
+    m = _ops._load_hip()
+    assert m is not None
+    torch.manual_seed(11)
+    for n in (4096, 40000, 1 << 20, (1 << 20) + 37):
+        # compressible: repeated byte runs + random tail
+        a = torch.randint(0, 4, (n,), dtype=torch.uint8, device="cuda")
+        a[: n // 2] = 7
+        ref = m.lz4_compress(a)
+        scratch = torch.empty(int(m.lz4_scratch_bytes(n)),
+                              dtype=torch.uint8, device="cuda")
+        out = torch.empty(ref.numel() + n, dtype=torch.uint8,
+                          device="cuda")
+        ln = torch.zeros(1, dtype=torch.long, device="cuda")
+        m.lz4_compress_into(a, scratch, out, ln)
+        k = int(ln.item())
+        assert k == ref.numel(), (k, ref.numel())
+        assert torch.equal(out[:k], ref)
+        # and the stream round-trips
+        back = m.lz4_decompress(out[:k].contiguous(), n)
+        assert torch.equal(back, a)
+
+
+@pytest.mark.gpu
+def test_var_hop_cuda_encode_roundtrip():
+    """The full CUDA zfp+lz4 hop encode sequence (zfp_encode into a
+    reused buffer -> lz4_compress_into a ring slot -> device wire
+    length), decoded through comm.Codec.decode — the exact kernel
+    sequence VarP2PRing.send_encoded issues per item."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    import defer_amd.ops as _ops
+    from defer_amd.config import PipelineConfig
+    from defer_amd.ops import codec as zc
+    from defer_amd.parallel.comm import Codec
+
+    m = _ops._load_hip()
+    torch.manual_seed(5)
+    shape = (2, 28, 28, 512)
+    cfg = PipelineConfig(compression="zfp+lz4", zfp_rate_bits=8)
+    c = Codec(cfg, shape, torch.bfloat16, "cuda")
+    x = (torch.randn(*shape) * 2).to("cuda", torch.bfloat16)
+    zfp_buf = torch.empty(c.zfp_bytes, dtype=torch.uint8, device="cuda")
+    zc.zfp_encode(x, c.rate, out=zfp_buf)
+    slot = c.alloc_wire()
+    scratch = torch.empty(int(m.lz4_scratch_bytes(c.zfp_bytes)),
+                          dtype=torch.uint8, device="cuda")
+    ln = torch.zeros(1, dtype=torch.long, device="cuda")
+    m.lz4_compress_into(zfp_buf, scratch, slot, ln)
+    n = int(ln.item())
+    assert 0 < n <= c.wire_numel
+    y = c.decode(slot[:n])
+    assert y.shape == shape and y.dtype == torch.bfloat16
+    # lz4 is lossless over zfp: equals the zfp-only roundtrip exactly
+    want = zc.zfp_decode(zfp_buf, shape, c.rate, dtype=torch.bfloat16)
+    assert torch.equal(y, want)
