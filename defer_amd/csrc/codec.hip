@@ -106,6 +106,61 @@ __device__ __forceinline__ u32 values_from_planes(u64 w, int lane) {
     return bt32_steps(x, lane);
 }
 
+// ---- two-block interleaved forms ----------------------------------------
+// The transform phase is 60% SQ_WAIT at 2 waves/SIMD (LDS-capped
+// occupancy; profiles/zfp_codec_throughput.txt PMC): the cross-lane
+// chain's latency is exposed. Convergent ops execute in program order,
+// so the only way to overlap two blocks' chains is to interleave the
+// statements by hand — each helper advances two independent chains
+// alternately, halving exposed latency per block pair.
+__device__ __forceinline__ void fwd_axis2(int& q0, int& q1, int lane,
+                                          int s) {
+    int r = (lane / s) & 3;
+    int o0 = __shfl_xor(q0, s);
+    int o1 = __shfl_xor(q1, s);
+    q0 = ((r & 1) == 0) ? ((q0 + o0) >> 1) : (o0 - q0);
+    q1 = ((r & 1) == 0) ? ((q1 + o1) >> 1) : (o1 - q1);
+    o0 = __shfl_xor(q0, 2 * s);
+    o1 = __shfl_xor(q1, 2 * s);
+    if ((r & 1) == 0) {
+        q0 = (r == 0) ? ((q0 + o0) >> 1) : (o0 - q0);
+        q1 = (r == 0) ? ((q1 + o1) >> 1) : (o1 - q1);
+    }
+    static const int MAPV[4] = {0, 2, 1, 3};
+    int dst = lane - r * s + MAPV[r] * s;
+    q0 = __shfl(q0, dst);
+    q1 = __shfl(q1, dst);
+}
+
+__device__ __forceinline__ void bt32_steps2(u32& x0, u32& x1, int lane) {
+    static const u32 M5[5] = {0x0000FFFFu, 0x00FF00FFu, 0x0F0F0F0Fu,
+                              0x33333333u, 0x55555555u};
+#pragma unroll
+    for (int i = 0; i < 5; ++i) {
+        const int j = 16 >> i;
+        const u32 m = M5[i];
+        u32 y0 = __shfl_xor(x0, j);
+        u32 y1 = __shfl_xor(x1, j);
+        if ((lane & j) == 0) {
+            x0 ^= (((x0 >> j) ^ y0) & m) << j;
+            x1 ^= (((x1 >> j) ^ y1) & m) << j;
+        } else {
+            x0 ^= ((y0 >> j) ^ x0) & m;
+            x1 ^= ((y1 >> j) ^ x1) & m;
+        }
+    }
+}
+
+__device__ __forceinline__ void planes_from_values2(u32 u0, u32 u1,
+                                                    int lane, u64& w0,
+                                                    u64& w1) {
+    bt32_steps2(u0, u1, lane);
+    u32 h0 = __shfl_xor(u0, 32);
+    u32 h1 = __shfl_xor(u1, 32);
+    w0 = (u64)u0 | ((u64)h0 << 32);
+    w1 = (u64)u1 | ((u64)h1 << 32);
+}
+
 // ---- lane-0 bit stream ---------------------------------------------------
 struct BitWriter {
     u32* out;       // current word pointer
@@ -245,15 +300,94 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_encode_kernel(
                     s_slab[wavei][r][lane] =
                         ((const elem_t*)xv)[idx];
                 }
-                // straightline block transform (writes predicated on
-                // `active` instead of early continues) unrolled 2x:
-                // the lift+transpose is a serial cross-lane dependency
-                // chain and the phase is latency-bound at 2 waves/SIMD,
-                // so two independent chains in flight per wave halve
-                // the exposed latency. Zero blocks waste a transform
-                // (uniform-rare in activation data) but stay bit-exact
-                // — their planes are simply not written.
-#pragma unroll 2
+                // hand-interleaved block PAIRS (PH 1/3): two
+                // independent cross-lane chains advanced alternately —
+                // convergent ops execute in program order, so source
+                // interleave is the only way to overlap the chains'
+                // LDS-pipe latency (60% SQ_WAIT at 2 waves/SIMD). Zero
+                // blocks waste a transform (uniform-rare in activation
+                // data) but stay bit-exact — planes just not written.
+                if (PH != 4 && PH != 5) {
+                    for (int b = 0; b + 1 < npan; b += 2) {
+                        int s0 = wavei * 64 + p * 16 + b;
+                        int s1 = s0 + 1;
+                        elem_t ra = s_slab[wavei][lane >> 2]
+                                          [b * 4 + (lane & 3)];
+                        elem_t rb = s_slab[wavei][lane >> 2]
+                                          [(b + 1) * 4 + (lane & 3)];
+                        float v0 = BF16_IN ? bf2f(*(bf16*)&ra)
+                                           : *(float*)&ra;
+                        float v1 = BF16_IN ? bf2f(*(bf16*)&rb)
+                                           : *(float*)&rb;
+                        float a0 = fabsf(v0), a1 = fabsf(v1);
+#pragma unroll
+                        for (int off = 32; off > 0; off >>= 1) {
+                            a0 = fmaxf(a0, __shfl_xor(a0, off));
+                            a1 = fmaxf(a1, __shfl_xor(a1, off));
+                        }
+                        const bool act0 = (a0 > 0.f) && isfinite(a0);
+                        const bool act1 = (a1 > 0.f) && isfinite(a1);
+                        int e0, e1;
+                        frexpf(a0, &e0);
+                        frexpf(a1, &e1);
+                        int q0 = (int)rintf(v0 * ldexpf(1.0f,
+                                                        QBITS - e0));
+                        int q1 = (int)rintf(v1 * ldexpf(1.0f,
+                                                        QBITS - e1));
+                        fwd_axis2(q0, q1, lane, 1);
+                        fwd_axis2(q0, q1, lane, 4);
+                        fwd_axis2(q0, q1, lane, 16);
+                        q0 = __shfl(q0, ZPERM[lane]);
+                        q1 = __shfl(q1, ZPERM[lane]);
+                        u32 u0 = ((u32)q0 + NBMASK) ^ NBMASK;
+                        u32 u1 = ((u32)q1 + NBMASK) ^ NBMASK;
+                        if (lane == 0) {
+                            s_hdr[s0] = act0
+                                ? ((1u << 15)
+                                   | ((u32)(e0 + 256) & 0x1FFu)) : 0u;
+                            s_hdr[s1] = act1
+                                ? ((1u << 15)
+                                   | ((u32)(e1 + 256) & 0x1FFu)) : 0u;
+                        }
+                        u64 w0, w1;
+                        planes_from_values2(u0, u1, lane, w0, w1);
+                        if (act0 && lane < PLANES)
+                            s_planes[s0][lane] = w0;
+                        if (act1 && lane < PLANES)
+                            s_planes[s1][lane] = w1;
+                    }
+                    if (npan & 1) {     // odd tail (unreachable under
+                        int b = npan - 1;            // PANEL alignment)
+                        int s = wavei * 64 + p * 16 + b;
+                        elem_t raw = s_slab[wavei][lane >> 2]
+                                           [b * 4 + (lane & 3)];
+                        float v = BF16_IN ? bf2f(*(bf16*)&raw)
+                                          : *(float*)&raw;
+                        float av = fabsf(v);
+#pragma unroll
+                        for (int off = 32; off > 0; off >>= 1)
+                            av = fmaxf(av, __shfl_xor(av, off));
+                        const bool active = (av > 0.f) && isfinite(av);
+                        int emax;
+                        frexpf(av, &emax);
+                        int q = (int)rintf(v * ldexpf(1.0f,
+                                                      QBITS - emax));
+                        q = fwd_axis(q, lane, 1);
+                        q = fwd_axis(q, lane, 4);
+                        q = fwd_axis(q, lane, 16);
+                        q = __shfl(q, ZPERM[lane]);
+                        u32 u = ((u32)q + NBMASK) ^ NBMASK;
+                        if (lane == 0)
+                            s_hdr[s] = active
+                                ? ((1u << 15)
+                                   | ((u32)(emax + 256) & 0x1FFu)) : 0u;
+                        u64 myw = planes_from_values(u, lane);
+                        if (active && lane < PLANES)
+                            s_planes[s][lane] = myw;
+                    }
+                    continue;           // next panel
+                }
+                // scalar loop: perf-bisect probes (PH 4/5) only
                 for (int b = 0; b < npan; ++b) {
                     int s = wavei * 64 + p * 16 + b;
                     elem_t raw = s_slab[wavei][lane >> 2]
