@@ -113,52 +113,52 @@ __device__ __forceinline__ u32 values_from_planes(u64 w, int lane) {
 // so the only way to overlap two blocks' chains is to interleave the
 // statements by hand — each helper advances two independent chains
 // alternately, halving exposed latency per block pair.
-__device__ __forceinline__ void fwd_axis2(int& q0, int& q1, int lane,
-                                          int s) {
+template <int NC>
+__device__ __forceinline__ void fwd_axisN(int q[NC], int lane, int s) {
     int r = (lane / s) & 3;
-    int o0 = __shfl_xor(q0, s);
-    int o1 = __shfl_xor(q1, s);
-    q0 = ((r & 1) == 0) ? ((q0 + o0) >> 1) : (o0 - q0);
-    q1 = ((r & 1) == 0) ? ((q1 + o1) >> 1) : (o1 - q1);
-    o0 = __shfl_xor(q0, 2 * s);
-    o1 = __shfl_xor(q1, 2 * s);
-    if ((r & 1) == 0) {
-        q0 = (r == 0) ? ((q0 + o0) >> 1) : (o0 - q0);
-        q1 = (r == 0) ? ((q1 + o1) >> 1) : (o1 - q1);
-    }
+    int o[NC];
+#pragma unroll
+    for (int c = 0; c < NC; ++c) o[c] = __shfl_xor(q[c], s);
+#pragma unroll
+    for (int c = 0; c < NC; ++c)
+        q[c] = ((r & 1) == 0) ? ((q[c] + o[c]) >> 1) : (o[c] - q[c]);
+#pragma unroll
+    for (int c = 0; c < NC; ++c) o[c] = __shfl_xor(q[c], 2 * s);
+#pragma unroll
+    for (int c = 0; c < NC; ++c)
+        if ((r & 1) == 0)
+            q[c] = (r == 0) ? ((q[c] + o[c]) >> 1) : (o[c] - q[c]);
     static const int MAPV[4] = {0, 2, 1, 3};
     int dst = lane - r * s + MAPV[r] * s;
-    q0 = __shfl(q0, dst);
-    q1 = __shfl(q1, dst);
+#pragma unroll
+    for (int c = 0; c < NC; ++c) q[c] = __shfl(q[c], dst);
 }
 
-__device__ __forceinline__ void bt32_steps2(u32& x0, u32& x1, int lane) {
+template <int NC>
+__device__ __forceinline__ void planes_from_valuesN(u32 u[NC], int lane,
+                                                    u64 w[NC]) {
     static const u32 M5[5] = {0x0000FFFFu, 0x00FF00FFu, 0x0F0F0F0Fu,
                               0x33333333u, 0x55555555u};
 #pragma unroll
     for (int i = 0; i < 5; ++i) {
         const int j = 16 >> i;
         const u32 m = M5[i];
-        u32 y0 = __shfl_xor(x0, j);
-        u32 y1 = __shfl_xor(x1, j);
-        if ((lane & j) == 0) {
-            x0 ^= (((x0 >> j) ^ y0) & m) << j;
-            x1 ^= (((x1 >> j) ^ y1) & m) << j;
-        } else {
-            x0 ^= ((y0 >> j) ^ x0) & m;
-            x1 ^= ((y1 >> j) ^ x1) & m;
+        u32 y[NC];
+#pragma unroll
+        for (int c = 0; c < NC; ++c) y[c] = __shfl_xor(u[c], j);
+#pragma unroll
+        for (int c = 0; c < NC; ++c) {
+            if ((lane & j) == 0)
+                u[c] ^= (((u[c] >> j) ^ y[c]) & m) << j;
+            else
+                u[c] ^= ((y[c] >> j) ^ u[c]) & m;
         }
     }
-}
-
-__device__ __forceinline__ void planes_from_values2(u32 u0, u32 u1,
-                                                    int lane, u64& w0,
-                                                    u64& w1) {
-    bt32_steps2(u0, u1, lane);
-    u32 h0 = __shfl_xor(u0, 32);
-    u32 h1 = __shfl_xor(u1, 32);
-    w0 = (u64)u0 | ((u64)h0 << 32);
-    w1 = (u64)u1 | ((u64)h1 << 32);
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
+        u32 h = __shfl_xor(u[c], 32);
+        w[c] = (u64)u[c] | ((u64)h << 32);
+    }
 }
 
 // ---- lane-0 bit stream ---------------------------------------------------
@@ -308,57 +308,60 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_encode_kernel(
                 // blocks waste a transform (uniform-rare in activation
                 // data) but stay bit-exact — planes just not written.
                 if (PH != 4 && PH != 5) {
-                    for (int b = 0; b + 1 < npan; b += 2) {
-                        int s0 = wavei * 64 + p * 16 + b;
-                        int s1 = s0 + 1;
-                        elem_t ra = s_slab[wavei][lane >> 2]
-                                          [b * 4 + (lane & 3)];
-                        elem_t rb = s_slab[wavei][lane >> 2]
-                                          [(b + 1) * 4 + (lane & 3)];
-                        float v0 = BF16_IN ? bf2f(*(bf16*)&ra)
-                                           : *(float*)&ra;
-                        float v1 = BF16_IN ? bf2f(*(bf16*)&rb)
-                                           : *(float*)&rb;
-                        float a0 = fabsf(v0), a1 = fabsf(v1);
+                    constexpr int NC = 4;
+                    int b = 0;
+                    for (; b + NC - 1 < npan; b += NC) {
+                        float v[NC], a[NC];
 #pragma unroll
-                        for (int off = 32; off > 0; off >>= 1) {
-                            a0 = fmaxf(a0, __shfl_xor(a0, off));
-                            a1 = fmaxf(a1, __shfl_xor(a1, off));
+                        for (int c = 0; c < NC; ++c) {
+                            elem_t raw = s_slab[wavei][lane >> 2]
+                                               [(b + c) * 4 + (lane & 3)];
+                            v[c] = BF16_IN ? bf2f(*(bf16*)&raw)
+                                           : *(float*)&raw;
+                            a[c] = fabsf(v[c]);
                         }
-                        const bool act0 = (a0 > 0.f) && isfinite(a0);
-                        const bool act1 = (a1 > 0.f) && isfinite(a1);
-                        int e0, e1;
-                        frexpf(a0, &e0);
-                        frexpf(a1, &e1);
-                        int q0 = (int)rintf(v0 * ldexpf(1.0f,
-                                                        QBITS - e0));
-                        int q1 = (int)rintf(v1 * ldexpf(1.0f,
-                                                        QBITS - e1));
-                        fwd_axis2(q0, q1, lane, 1);
-                        fwd_axis2(q0, q1, lane, 4);
-                        fwd_axis2(q0, q1, lane, 16);
-                        q0 = __shfl(q0, ZPERM[lane]);
-                        q1 = __shfl(q1, ZPERM[lane]);
-                        u32 u0 = ((u32)q0 + NBMASK) ^ NBMASK;
-                        u32 u1 = ((u32)q1 + NBMASK) ^ NBMASK;
-                        if (lane == 0) {
-                            s_hdr[s0] = act0
-                                ? ((1u << 15)
-                                   | ((u32)(e0 + 256) & 0x1FFu)) : 0u;
-                            s_hdr[s1] = act1
-                                ? ((1u << 15)
-                                   | ((u32)(e1 + 256) & 0x1FFu)) : 0u;
+#pragma unroll
+                        for (int off = 32; off > 0; off >>= 1)
+#pragma unroll
+                            for (int c = 0; c < NC; ++c)
+                                a[c] = fmaxf(a[c],
+                                             __shfl_xor(a[c], off));
+                        bool act[NC];
+                        int e[NC], q[NC];
+#pragma unroll
+                        for (int c = 0; c < NC; ++c) {
+                            act[c] = (a[c] > 0.f) && isfinite(a[c]);
+                            frexpf(a[c], &e[c]);
+                            q[c] = (int)rintf(
+                                v[c] * ldexpf(1.0f, QBITS - e[c]));
                         }
-                        u64 w0, w1;
-                        planes_from_values2(u0, u1, lane, w0, w1);
-                        if (act0 && lane < PLANES)
-                            s_planes[s0][lane] = w0;
-                        if (act1 && lane < PLANES)
-                            s_planes[s1][lane] = w1;
+                        fwd_axisN<NC>(q, lane, 1);
+                        fwd_axisN<NC>(q, lane, 4);
+                        fwd_axisN<NC>(q, lane, 16);
+                        u32 u[NC];
+#pragma unroll
+                        for (int c = 0; c < NC; ++c) {
+                            q[c] = __shfl(q[c], ZPERM[lane]);
+                            u[c] = ((u32)q[c] + NBMASK) ^ NBMASK;
+                        }
+                        if (lane == 0)
+#pragma unroll
+                            for (int c = 0; c < NC; ++c)
+                                s_hdr[wavei * 64 + p * 16 + b + c] =
+                                    act[c]
+                                    ? ((1u << 15)
+                                       | ((u32)(e[c] + 256) & 0x1FFu))
+                                    : 0u;
+                        u64 w[NC];
+                        planes_from_valuesN<NC>(u, lane, w);
+#pragma unroll
+                        for (int c = 0; c < NC; ++c)
+                            if (act[c] && lane < PLANES)
+                                s_planes[wavei * 64 + p * 16 + b + c]
+                                        [lane] = w[c];
                     }
-                    if (npan & 1) {     // odd tail (unreachable under
-                        int b = npan - 1;            // PANEL alignment)
-                        int s = wavei * 64 + p * 16 + b;
+                    for (; b < npan; ++b) {   // tail (unreachable under
+                        int s = wavei * 64 + p * 16 + b;  // PANEL align)
                         elem_t raw = s_slab[wavei][lane >> 2]
                                            [b * 4 + (lane & 3)];
                         float v = BF16_IN ? bf2f(*(bf16*)&raw)
