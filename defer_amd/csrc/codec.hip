@@ -161,6 +161,57 @@ __device__ __forceinline__ void planes_from_valuesN(u32 u[NC], int lane,
     }
 }
 
+template <int NC>
+__device__ __forceinline__ void inv_axisN(int q[NC], int lane, int s) {
+    int r = (lane / s) & 3;
+    static const int MAPV[4] = {0, 2, 1, 3};
+    int dst = lane - r * s + MAPV[r] * s;
+#pragma unroll
+    for (int c = 0; c < NC; ++c) q[c] = __shfl(q[c], dst);
+    int o[NC];
+#pragma unroll
+    for (int c = 0; c < NC; ++c) o[c] = __shfl_xor(q[c], 2 * s);
+#pragma unroll
+    for (int c = 0; c < NC; ++c)
+        if ((r & 1) == 0)
+            q[c] = (r == 0) ? ((q[c] - (o[c] >> 1)) + o[c])
+                            : (o[c] - (q[c] >> 1));
+#pragma unroll
+    for (int c = 0; c < NC; ++c) o[c] = __shfl_xor(q[c], s);
+#pragma unroll
+    for (int c = 0; c < NC; ++c)
+        q[c] = ((r & 1) == 0) ? ((q[c] - (o[c] >> 1)) + o[c])
+                              : (o[c] - (q[c] >> 1));
+}
+
+template <int NC>
+__device__ __forceinline__ void values_from_planesN(u64 w[NC], int lane,
+                                                    u32 u[NC]) {
+    static const u32 M5[5] = {0x0000FFFFu, 0x00FF00FFu, 0x0F0F0F0Fu,
+                              0x33333333u, 0x55555555u};
+    u64 other[NC];
+#pragma unroll
+    for (int c = 0; c < NC; ++c) other[c] = __shfl_xor(w[c], 32);
+#pragma unroll
+    for (int c = 0; c < NC; ++c)
+        u[c] = (lane < 32) ? (u32)w[c] : (u32)(other[c] >> 32);
+#pragma unroll
+    for (int i = 0; i < 5; ++i) {
+        const int j = 16 >> i;
+        const u32 m = M5[i];
+        u32 y[NC];
+#pragma unroll
+        for (int c = 0; c < NC; ++c) y[c] = __shfl_xor(u[c], j);
+#pragma unroll
+        for (int c = 0; c < NC; ++c) {
+            if ((lane & j) == 0)
+                u[c] ^= (((u[c] >> j) ^ y[c]) & m) << j;
+            else
+                u[c] ^= ((y[c] >> j) ^ u[c]) & m;
+        }
+    }
+}
+
 // ---- lane-0 bit stream ---------------------------------------------------
 struct BitWriter {
     u32* out;       // current word pointer
@@ -623,14 +674,56 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_decode_kernel(
                 int bk0 = (int)(pb % b2);
                 long t = pb / b2;
                 int bj = (int)(t % b1), bi = (int)(t / b1);
-                for (int b = 0; b < npan; ++b) {
-                    int s = wavei * 64 + p * 16 + b;
+                // four inverse chains hand-interleaved per wave (same
+                // latency argument as the encoder); dead-header blocks
+                // compute garbage that the flag masks to 0
+                constexpr int NC = 4;
+                int b = 0;
+                for (; b + NC - 1 < npan; b += NC) {
+                    u64 w[NC];
+                    u32 u[NC];
+                    int q[NC], em[NC];
+                    bool live[NC];
+#pragma unroll
+                    for (int c = 0; c < NC; ++c) {
+                        int s = wavei * 64 + p * 16 + b + c;
+                        u32 hdr = s_hdr[s];
+                        live[c] = (hdr >> 15) != 0;
+                        em[c] = (int)(hdr & 0x1FFu) - 256;
+                        w[c] = (lane < PLANES) ? s_planes[s][lane] : 0;
+                    }
+                    values_from_planesN<NC>(w, lane, u);
+#pragma unroll
+                    for (int c = 0; c < NC; ++c)
+                        q[c] = (int)((u[c] ^ NBMASK) - NBMASK);
+#pragma unroll
+                    for (int c = 0; c < NC; ++c)
+                        q[c] = __shfl(q[c], ZIPERM[lane]);
+                    inv_axisN<NC>(q, lane, 16);
+                    inv_axisN<NC>(q, lane, 4);
+                    inv_axisN<NC>(q, lane, 1);
+#pragma unroll
+                    for (int c = 0; c < NC; ++c) {
+                        float outv = live[c]
+                            ? ldexpf((float)q[c], em[c] - QBITS) : 0.f;
+                        if (BF16_OUT) {
+                            bf16 h = f2bf(outv);
+                            s_slab[wavei][lane >> 2]
+                                  [(b + c) * 4 + (lane & 3)] =
+                                *(elem_t*)&h;
+                        } else {
+                            s_slab[wavei][lane >> 2]
+                                  [(b + c) * 4 + (lane & 3)] =
+                                *(elem_t*)&outv;
+                        }
+                    }
+                }
+                for (; b < npan; ++b) {   // tail (unreachable: PANEL
+                    int s = wavei * 64 + p * 16 + b;    // keeps npan 16)
                     u32 hdr = s_hdr[s];
                     float outv = 0.f;
                     if (hdr >> 15) {
                         int emax = (int)(hdr & 0x1FFu) - 256;
-                        // one ds_read_b64 per lane + butterfly back
-                        // (inverse of the encoder's transpose)
                         u64 w = (lane < PLANES) ? s_planes[s][lane] : 0;
                         u32 u = values_from_planes(w, lane);
                         int q = (int)((u ^ NBMASK) - NBMASK);
