@@ -52,18 +52,39 @@ def find_calibration(model_name: str,
 @torch.no_grad()
 def measure_layer_times(gm, input_shape=(64, 224, 224, 3),
                         device="cuda", dtype=torch.bfloat16,
-                        iters: int = 30, warmup: int = 8
-                        ) -> Dict[str, float]:
+                        iters: int = 30, warmup: int = 8,
+                        reps: int = 3) -> Dict[str, float]:
     """Time each graph node's layer standalone on `device` at the given
     batch; returns {node_name: us per IMAGE}. GPU-only by intent (the
     partitioner's costs describe the HIP kernels); CPU works for tests
-    with wall-clock timing."""
+    with wall-clock timing.
+
+    Each layer is timed as the MIN over `reps` independent event-timed
+    runs: one-off multi-ms stalls (allocator slow paths, DPM dips on a
+    fresh box) otherwise land in a single layer's mean and poison the
+    cut choice — observed as ~47 us/img outliers on elementwise nodes,
+    114x their real cost."""
     graph: LayerGraph = gm.graph
     dev = torch.device(device)
     cuda = dev.type == "cuda"
     if not cuda:
         dtype = torch.float32
     batch = int(input_shape[0])
+
+    if cuda:
+        # sustained load to ramp DPM clocks before any timed region
+        xw = torch.randn(max(batch, 16), 56, 56, 64, device=dev,
+                         dtype=dtype)
+        ww = (torch.randn(64, 3, 3, 64) * 0.05).to(dev, dtype)
+        from defer_amd import ops as _ops
+
+        t0 = __import__("time").perf_counter()
+        while __import__("time").perf_counter() - t0 < 0.6:
+            for _ in range(50):
+                _ops.conv2d_bn_act(xw, ww, None, None, stride=1,
+                                   padding=1)
+            torch.cuda.synchronize(dev)
+        del xw, ww
 
     # shape trace on CPU to learn each node's input shapes
     shapes = {LayerGraph.INPUT: tuple(input_shape)}
@@ -89,23 +110,28 @@ def measure_layer_times(gm, input_shape=(64, 224, 224, 3),
                for p in n.inputs]
         for _ in range(warmup):
             layer(*ins, **n.kwargs)
-        if cuda:
-            torch.cuda.synchronize(dev)
-            ev0, ev1 = torch.cuda.Event(True), torch.cuda.Event(True)
-            ev0.record()
-            for _ in range(iters):
-                layer(*ins, **n.kwargs)
-            ev1.record()
-            ev1.synchronize()
-            ms = ev0.elapsed_time(ev1)
-        else:
-            import time
+        it = max(iters // reps, 1)
+        best_ms = None
+        for _ in range(reps):
+            if cuda:
+                torch.cuda.synchronize(dev)
+                ev0 = torch.cuda.Event(True)
+                ev1 = torch.cuda.Event(True)
+                ev0.record()
+                for _ in range(it):
+                    layer(*ins, **n.kwargs)
+                ev1.record()
+                ev1.synchronize()
+                ms = ev0.elapsed_time(ev1)
+            else:
+                import time
 
-            t0 = time.perf_counter()
-            for _ in range(iters):
-                layer(*ins, **n.kwargs)
-            ms = (time.perf_counter() - t0) * 1e3
-        times_us[n.name] = ms * 1e3 / iters / batch
+                t0 = time.perf_counter()
+                for _ in range(it):
+                    layer(*ins, **n.kwargs)
+                ms = (time.perf_counter() - t0) * 1e3
+            best_ms = ms if best_ms is None else min(best_ms, ms)
+        times_us[n.name] = best_ms * 1e3 / it / batch
         del ins, layer
     return times_us
 
