@@ -286,3 +286,61 @@ def test_fp8_wire_hip_matches_torch_cast():
         want_dec = (vals * (amax.double() / 448.0).float()) \
             .to(torch.bfloat16).view(xg.shape)
         assert torch.equal(dec, want_dec)
+
+
+@pytest.mark.gpu
+def test_var_ring_cuda_sender_bookkeeping(monkeypatch):
+    """VarP2PRing's CUDA sender (deferred payload, device-side size
+    message, pinned readback one item later) without a peer: stub
+    dist.isend and verify the issued sequence s_0,s_1,p_0,s_2,p_1,...
+    with correct device-computed lengths and decodable payloads."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    import torch.distributed as dist
+
+    from defer_amd.config import PipelineConfig
+    from defer_amd.parallel.comm import Codec, VarP2PRing
+
+    sent = []
+
+    class _W:
+        def wait(self):
+            pass
+
+    def fake_isend(t, dst=None, **kw):
+        torch.cuda.synchronize()          # value must be final by now
+        sent.append(t.detach().clone())
+        return _W()
+
+    monkeypatch.setattr(dist, "isend", fake_isend)
+
+    shape = (2, 14, 14, 256)
+    cfg = PipelineConfig(compression="zfp+lz4", zfp_rate_bits=8,
+                         ring_depth=2)
+    c = Codec(cfg, shape, torch.bfloat16, "cuda")
+    ring = VarP2PRing(c, cfg.ring_depth)
+    assert ring.defer and ring.cuda
+    torch.manual_seed(9)
+    xs = [(torch.randn(*shape) * 2).to("cuda", torch.bfloat16)
+          for _ in range(4)]
+    for k, x in enumerate(xs):
+        ring.send_encoded(k, x, dst=1)
+    ring.flush()
+    # sequence: s0, s1, p0, s2, p1, s3, p2, p3
+    kinds = ["s", "s", "p", "s", "p", "s", "p", "p"]
+    items = [0, 1, 0, 2, 1, 3, 2, 3]
+    assert len(sent) == len(kinds)
+    from defer_amd.ops import codec as zc
+
+    sizes = {}
+    for t, kind, it in zip(sent, kinds, items):
+        if kind == "s":
+            assert t.dtype == torch.long and t.numel() == 1
+            sizes[it] = int(t.item())
+        else:
+            n = sizes[it]
+            assert t.numel() == n, (it, t.numel(), n)
+            want = zc.zfp_decode(zc.zfp_encode(xs[it], c.rate),
+                                 shape, c.rate, dtype=torch.bfloat16)
+            got = c.decode(t)
+            assert torch.equal(got, want), f"item {it} corrupted"
