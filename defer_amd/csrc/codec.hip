@@ -359,7 +359,7 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_encode_kernel(
                 // blocks waste a transform (uniform-rare in activation
                 // data) but stay bit-exact — planes just not written.
                 if (PH != 4 && PH != 5) {
-                    constexpr int NC = 4;
+                    constexpr int NC = 8;
                     int b = 0;
                     for (; b + NC - 1 < npan; b += NC) {
                         float v[NC], a[NC];
@@ -677,7 +677,7 @@ __global__ __launch_bounds__(CGRP, 2) void zfp_decode_kernel(
                 // four inverse chains hand-interleaved per wave (same
                 // latency argument as the encoder); dead-header blocks
                 // compute garbage that the flag masks to 0
-                constexpr int NC = 4;
+                constexpr int NC = 8;
                 int b = 0;
                 for (; b + NC - 1 < npan; b += NC) {
                     u64 w[NC];
