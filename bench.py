@@ -51,7 +51,7 @@ def parse_args(argv=None):
                          "1 stage; 64 at >1 stage so 20 driver steps are "
                          ">=80 items and pipeline fill/drain stays <10%%)")
     ap.add_argument("--compression", default="none",
-                    choices=["none", "fp8", "zfp", "zfp+lz4"])
+                    choices=["none", "fp8", "zfp", "zfp+lz4", "auto"])
     ap.add_argument("--zfp-bits", type=int, default=8)
     ap.add_argument("--cuts", default="auto",
                     help='"auto", "defer8", or comma-separated layer names')

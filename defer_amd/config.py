@@ -41,7 +41,11 @@ class PipelineConfig:
     # lz4(zfp(x)) (dispatcher.py:81-84). "none" ships raw bf16 over xGMI;
     # "fp8" ships a cast-only e4m3 wire (1 B/value, ~free); "zfp" ships
     # fixed-rate ZFP blocks; "zfp+lz4" adds the LZ4 stage.
-    compression: str = "none"     # "none" | "fp8" | "zfp" | "zfp+lz4"
+    # "auto" picks the wire PER HOP: fp8 on hops whose raw-bf16 relay
+    # would exceed the slowest stage's compute (the pipeline bottleneck),
+    # lossless bf16 elsewhere (comm.choose_hop_modes — deterministic
+    # from the calibration + cuts, so all ranks agree).
+    compression: str = "none"  # "none"|"fp8"|"zfp"|"zfp+lz4"|"auto"
     zfp_rate_bits: int = 8        # fixed-rate bits per value (ZFP)
 
     # Depth of the per-stage device-resident activation ring buffers (the
@@ -91,10 +95,11 @@ class PipelineConfig:
     extra: dict = field(default_factory=dict)
 
     def __post_init__(self):
-        if self.compression not in ("none", "fp8", "zfp", "zfp+lz4"):
+        if self.compression not in ("none", "fp8", "zfp", "zfp+lz4",
+                                    "auto"):
             raise ValueError(
                 f"unknown compression {self.compression!r} "
-                "(none|fp8|zfp|zfp+lz4)")
+                "(none|fp8|zfp|zfp+lz4|auto)")
         if self.dtype not in ("bf16", "fp16", "fp32"):
             raise ValueError(f"unknown dtype {self.dtype!r} "
                              "(bf16|fp16|fp32)")

@@ -37,7 +37,7 @@ def main():
     ap.add_argument("--report", type=int, default=64,
                     help="rank 0 prints throughput every N items")
     ap.add_argument("--compression", default="none",
-                    choices=["none", "fp8", "zfp", "zfp+lz4"])
+                    choices=["none", "fp8", "zfp", "zfp+lz4", "auto"])
     ap.add_argument("--device", default="cuda", choices=["cuda", "cpu"])
     ap.add_argument("--dual-rail", action="store_true")
     ap.add_argument("--calibration", default=None,

@@ -17,11 +17,32 @@ import torch.distributed as dist
 from defer_amd.config import PipelineConfig
 
 
+def choose_hop_modes(stage_us, hop_bytes, link_gbps, dual_boosts=None):
+    """Per-hop wire mode for compression="auto": a hop whose raw-bf16
+    relay time would exceed the slowest stage's compute becomes the
+    pipeline bottleneck — ship THAT hop as fp8 (2x fewer bytes, codec
+    ~free at 1.0/2.6 TB/s) and keep every other hop lossless. Pure
+    function of (per-stage us, per-hop bytes, link rate): every rank
+    computes the same answer, so send/recv codecs always agree.
+
+    stage_us: per-stage compute (us/item); hop_bytes: bf16 bytes/item
+    per hop; dual_boosts: per-hop effective-bandwidth multiplier
+    (comm.dual_bw_boost). Returns ["none"|"fp8"] per hop."""
+    bottleneck = max(stage_us) if stage_us else 0.0
+    modes = []
+    for i, b in enumerate(hop_bytes):
+        boost = dual_boosts[i] if dual_boosts else 1.0
+        hop_us = b / (link_gbps * 1e3 * boost)
+        modes.append("fp8" if hop_us > bottleneck else "none")
+    return modes
+
+
 class Codec:
     """Encode/decode boundary activations. "none" = raw tensor."""
 
-    def __init__(self, cfg: PipelineConfig, shape, dtype, device):
-        self.mode = cfg.compression
+    def __init__(self, cfg: PipelineConfig, shape, dtype, device,
+                 mode: str = None):
+        self.mode = mode if mode is not None else cfg.compression
         self.shape = tuple(shape)
         self.dtype = dtype
         self.device = device

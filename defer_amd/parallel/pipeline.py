@@ -319,6 +319,37 @@ class DistPipeline:
                            cut_points=cuts)
         # shape-trace on CPU BEFORE weights move to the GPU
         self.in_shape, self.out_shape = self._boundary_shapes(stages)
+        # compression="auto": per-hop wire choice (fp8 on would-be
+        # bottleneck hops, lossless bf16 elsewhere) — deterministic from
+        # calibration + cuts, identical on every rank
+        self._hop_modes = None
+        if cfg.compression == "auto" and self.world > 1:
+            from defer_amd.parallel.calibrate import find_calibration
+            from defer_amd.parallel.comm import choose_hop_modes
+            from defer_amd.parallel.partitioner import (XGMI_LINK_GBPS,
+                                                        node_times)
+
+            cal = find_calibration(gm.model_name, cfg.calibration_file)
+            _, _, tu = node_times(gm.graph,
+                                  (1,) + self.batch_shape[1:])
+            if cal:
+                tu = {n: cal.get(n, v) for n, v in tu.items()}
+            stage_us = [sum(tu[n.name] for n in s.graph.nodes)
+                        for s in stages]
+            bpe = 2.0 if self.dtype == torch.bfloat16 else 4.0
+            hop_bytes = [float(torch.tensor(shp[1:]).prod()) * bpe
+                         for shp in self._chain_shapes[:-1]]
+            d_act = dual_active(cfg, self.world)
+            boosts = [2.0 if d_act
+                      and hop_via(i, self.world) is not None else 1.0
+                      for i in range(self.world - 1)]
+            self._hop_modes = choose_hop_modes(
+                stage_us, hop_bytes, XGMI_LINK_GBPS, boosts)
+            if self.rank == 0 and cfg.log_stage_stats:
+                import sys
+
+                print(f"defer_amd: auto wire modes {self._hop_modes}",
+                      file=sys.stderr)
         if cfg.weights_dir:
             # per-rank stage weights only (reference: per-node weight
             # shipping, dispatcher.py:57) — before fusion/pre-cast
@@ -351,13 +382,19 @@ class DistPipeline:
                     return DualRailRing(codec, cfg.ring_depth, via)
                 return make_ring(codec, cfg.ring_depth)
 
+            def _mode(hop):
+                return (self._hop_modes[hop] if self._hop_modes
+                        else None)
+
             if self.rank > 0:
                 self.in_codec = Codec(cfg, self.in_shape, self.dtype,
-                                      self.device)
+                                      self.device,
+                                      mode=_mode(self.rank - 1))
                 self.recv_ring = _ring(self.in_codec, self.rank - 1)
             if self.rank < self.world - 1:
                 self.out_codec = Codec(cfg, self.out_shape, self.dtype,
-                                       self.device)
+                                       self.device,
+                                       mode=_mode(self.rank))
                 self.send_ring = _ring(self.out_codec, self.rank)
             if dual:
                 # this rank's forwarding duty (at most one hop routes
@@ -367,7 +404,8 @@ class DistPipeline:
                                None)
                 if fwd_hop is not None:
                     c = Codec(cfg, self._chain_shapes[fwd_hop],
-                              self.dtype, self.device)
+                              self.dtype, self.device,
+                              mode=_mode(fwd_hop))
                 if fwd_hop is not None and c.wire_numel >= 16:
                     # same small-wire guard as _ring: a hop that stayed
                     # single-rail has no second half to relay

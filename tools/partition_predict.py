@@ -22,6 +22,9 @@ def main():
                     help="assume dual-rail hops (2x link bandwidth, "
                          "world >= 4; matches DistPipeline's cut "
                          "choice with cfg.dual_rail)")
+    ap.add_argument("--auto-wire", action="store_true",
+                    help="model compression=auto: hops past the "
+                         "bottleneck stage ship fp8 (half the bytes)")
     args = ap.parse_args()
 
     from defer_amd.models import MODELS
@@ -55,6 +58,14 @@ def main():
             st.append(sum(t[start:e + 1]))
             start = e + 1
         hops = [ob[c] / (XGMI_LINK_GBPS * mult * 1e3) for c in cuts]
+        if args.auto_wire:
+            from defer_amd.parallel.comm import choose_hop_modes
+
+            modes = choose_hop_modes(st, [ob[c] for c in cuts],
+                                     XGMI_LINK_GBPS * mult)
+            hops = [h / 2 if m == "fp8" else h
+                    for h, m in zip(hops, modes)]
+            print(f"  auto wire: {modes}")
         bot = max(st + hops)
         kind = "compute" if bot in st else "xGMI hop"
         print(f"pp{ns}: cuts={cuts}")
