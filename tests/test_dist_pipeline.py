@@ -585,3 +585,11 @@ def test_dist_pipeline_auto_forces_fp8_with_tiny_calibration(tmp_path):
     assert len(errs) == steps
     for e in errs:
         assert e < 1e-6, e
+
+
+def test_dual_rail_with_auto_compression_world4():
+    """auto wire + dual-rail combined (gloo world 4): the hop-mode
+    decision must fold in the dual bandwidth boost and the split rails
+    must carry whatever per-hop wire was chosen."""
+    _run(4, ["add_4", "add_8", "add_12"], steps=4, compression="auto",
+         dual_rail=True, tol=0.0)
