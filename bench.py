@@ -42,7 +42,7 @@ def parse_args(argv=None):
     ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--model", default="resnet50",
                     choices=["resnet50", "resnet101", "resnet152",
-                             "vgg19", "vgg19_gap"])
+                             "vgg19", "vgg19_gap", "densenet121"])
     ap.add_argument("--batch", type=int, default=256,
                     help="images per step (global batch; throughput "
                          "saturates ~128-256, profiles/README.md)")
@@ -302,7 +302,8 @@ def worker(args):
             "metric": "images/sec (whole node) "
                       + {"resnet50": "ResNet50", "resnet101": "ResNet101",
                          "resnet152": "ResNet152", "vgg19": "VGG19",
-                         "vgg19_gap": "VGG19-GAP"}[args.model]
+                         "vgg19_gap": "VGG19-GAP",
+                         "densenet121": "DenseNet121"}[args.model]
                       + " pipeline",
             "value": round(ips, 1),
             "unit": "images/sec",

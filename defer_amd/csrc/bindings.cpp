@@ -312,6 +312,52 @@ Tensor maxpool2d(Tensor x, int64_t kernel, int64_t stride, int64_t pad) {
     return out;
 }
 
+Tensor avgpool2d(Tensor x, int64_t kernel, int64_t stride, int64_t pad) {
+    check_bf16(x, "x");
+    TORCH_CHECK(x.dim() == 4, "x must be NHWC");
+    int NB = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+    TORCH_CHECK(C % 8 == 0, "C must be a multiple of 8");
+    int OH = (int)((H + 2 * pad - kernel) / stride + 1);
+    int OW = (int)((W + 2 * pad - kernel) / stride + 1);
+    auto out = at::empty({NB, OH, OW, C}, x.options());
+    defer_hip::launch_avgpool(bptr(x), bptr_mut(out), NB, H, W, C, OH,
+                              OW, (int)kernel, (int)stride, (int)pad,
+                              cur_stream());
+    return out;
+}
+
+// Concat along the LAST (channel) dim of NHWC tensors: pure data
+// movement, one strided device-to-device hipMemcpy2DAsync per input
+// (row = one pixel's channels) — no torch compute kernels in the path
+// (DenseNet's dense blocks are concat-heavy).
+Tensor concat_lastdim(std::vector<Tensor> xs) {
+    TORCH_CHECK(!xs.empty(), "concat of nothing");
+    auto base = xs[0].sizes().vec();
+    int nd = (int)base.size();
+    long rows = 1;
+    for (int i = 0; i + 1 < nd; ++i) rows *= base[i];
+    long ctot = 0;
+    for (auto& t : xs) {
+        check_bf16(t, "concat input");
+        TORCH_CHECK((int)t.sizes().size() == nd, "rank mismatch");
+        for (int i = 0; i + 1 < nd; ++i)
+            TORCH_CHECK(t.size(i) == base[i], "leading-dim mismatch");
+        ctot += t.size(nd - 1);
+    }
+    base[nd - 1] = ctot;
+    auto out = at::empty(base, xs[0].options());
+    hipStream_t s = cur_stream();
+    long coff = 0;
+    for (auto& t : xs) {
+        long cj = t.size(nd - 1);
+        hipMemcpy2DAsync((char*)bptr_mut(out) + coff * 2, ctot * 2,
+                         bptr(t), cj * 2, cj * 2, rows,
+                         hipMemcpyDeviceToDevice, s);
+        coff += cj;
+    }
+    return out;
+}
+
 Tensor global_avg_pool(Tensor x) {
     check_bf16(x, "x");
     TORCH_CHECK(x.dim() == 4, "x must be NHWC");
@@ -458,6 +504,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("relu", &relu);
     m.def("softmax", &softmax);
     m.def("maxpool2d", &maxpool2d);
+    m.def("avgpool2d", &avgpool2d);
+    m.def("concat_lastdim", &concat_lastdim);
     m.def("global_avg_pool", &global_avg_pool);
     m.def("zfp_encode", &zfp_encode, py::arg("x"), py::arg("rate"),
           py::arg("out") = py::none(), py::arg("phases") = 3);

@@ -53,6 +53,9 @@ void launch_softmax(const void* x, void* y, int rows, int cols,
 void launch_maxpool(const void* x, void* y, int NB, int H, int W, int C,
                     int OH, int OW, int k, int stride, int pad,
                     hipStream_t s);
+void launch_avgpool(const void* x, void* y, int NB, int H, int W, int C,
+                    int OH, int OW, int k, int stride, int pad,
+                    hipStream_t s);
 void launch_gap(const void* x, void* y, int NB, int HW, int C,
                 hipStream_t s);
 

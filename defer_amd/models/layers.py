@@ -78,6 +78,69 @@ class MaxPool(nn.Module):
         return ops.maxpool2d(x, self.kernel, self.stride, self.padding)
 
 
+class AvgPool(nn.Module):
+    """Average pooling (DenseNet transition 2x2/2)."""
+
+    def __init__(self, kernel=2, stride=2, padding=0):
+        super().__init__()
+        self.kernel, self.stride, self.padding = kernel, stride, padding
+
+    def forward(self, x):
+        return ops.avgpool2d(x, self.kernel, self.stride, self.padding)
+
+
+class Concat(nn.Module):
+    """Channel concat of NHWC inputs (DenseNet dense connections).
+    GPU path is strided device copies, no torch compute kernels."""
+
+    def forward(self, *xs):
+        return ops.concat_channels(xs)
+
+
+class BNActConv(nn.Module):
+    """Pre-activation composite BN -> ReLU -> Conv (DenseNet layer
+    ordering; torchvision DenseLayer norm/relu/conv). Inference BN is a
+    folded per-channel scale/bias applied by the standalone bn_act HIP
+    kernel; the conv runs without epilogue BN."""
+
+    def __init__(self, cin, cout, kernel=1, stride=1, padding=0):
+        super().__init__()
+        self.cin, self.cout = cin, cout
+        self.kernel, self.stride, self.padding = kernel, stride, padding
+        fan_in = cin * kernel * kernel
+        w = torch.randn(cout, kernel, kernel, cin) \
+            * math.sqrt(2.0 / fan_in)
+        self.weight = nn.Parameter(w)
+        self.scale = nn.Parameter(torch.ones(cin))
+        self.bias = nn.Parameter(torch.zeros(cin))
+
+    def forward(self, x):
+        z = ops.batchnorm_apply(x, self.scale, self.bias, act="relu")
+        return ops.conv2d_bn_act(z, self.weight.to(x.dtype), None, None,
+                                 stride=self.stride,
+                                 padding=self.padding, act="none")
+
+    def extra_repr(self):
+        return (f"bn-relu-conv {self.cin}->{self.cout} k{self.kernel} "
+                f"s{self.stride} p{self.padding}")
+
+    def flops_per_pixel(self):
+        return 2 * self.kernel * self.kernel * self.cin * self.cout
+
+
+class BNAct(nn.Module):
+    """Standalone folded-BN + activation (DenseNet final norm5)."""
+
+    def __init__(self, c, act="relu"):
+        super().__init__()
+        self.act = act
+        self.scale = nn.Parameter(torch.ones(c))
+        self.bias = nn.Parameter(torch.zeros(c))
+
+    def forward(self, x):
+        return ops.batchnorm_apply(x, self.scale, self.bias, self.act)
+
+
 class GlobalAvgPool(nn.Module):
     def forward(self, x):
         return ops.global_avg_pool(x)

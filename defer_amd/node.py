@@ -29,7 +29,7 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--model", default="resnet50",
                     choices=["resnet50", "resnet101", "resnet152",
-                             "vgg19", "vgg19_gap"])
+                             "vgg19", "vgg19_gap", "densenet121"])
     ap.add_argument("--batch", type=int, default=64)
     ap.add_argument("--cuts", default="auto")
     ap.add_argument("--items", type=int, default=0,

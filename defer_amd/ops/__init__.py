@@ -89,6 +89,23 @@ def maxpool2d(x, kernel=3, stride=2, padding=1):
     return m.maxpool2d(x, kernel, stride, padding)
 
 
+def avgpool2d(x, kernel=2, stride=2, padding=0):
+    m = _backend(x)
+    if m is None:
+        return _ref.avgpool2d(x, kernel, stride, padding)
+    return m.avgpool2d(x, kernel, stride, padding)
+
+
+def concat_channels(xs):
+    """Concat NHWC tensors on the channel (last) dim — DenseNet's dense
+    connections. GPU: one strided device copy per input (no torch
+    compute kernels)."""
+    m = _backend(xs[0])
+    if m is None:
+        return _ref.concat_channels(xs)
+    return m.concat_lastdim(list(xs))
+
+
 def global_avg_pool(x):
     m = _backend(x)
     if m is None:

@@ -76,6 +76,20 @@ def maxpool2d(x, kernel=3, stride=2, padding=1):
     return y.permute(0, 2, 3, 1).to(x.dtype)
 
 
+def avgpool2d(x, kernel=2, stride=2, padding=0):
+    """Average pooling, NHWC, valid taps only (count_include_pad=False;
+    DenseNet transition layers)."""
+    xf = x.permute(0, 3, 1, 2).float()
+    y = F.avg_pool2d(xf, kernel_size=kernel, stride=stride,
+                     padding=padding, count_include_pad=False)
+    return y.permute(0, 2, 3, 1).to(x.dtype)
+
+
+def concat_channels(xs):
+    """Concat NHWC tensors on the channel dim (DenseNet)."""
+    return torch.cat(list(xs), dim=-1)
+
+
 def global_avg_pool(x):
     """[N, H, W, C] -> [N, C] (ResNet50 head)."""
     return x.float().mean(dim=(1, 2)).to(x.dtype)

@@ -140,3 +140,46 @@ def test_resnet_deep_variants():
             for s in stages:
                 z = s(z)
         assert torch.allclose(z, y)
+
+
+def test_densenet121_shapes_partition_and_cuts():
+    """DenseNet-121: concat-DAG model family — cumulative concats are
+    articulation points, so the partitioner can cut INSIDE dense blocks
+    (the reference's dag_util handles multi-parent Keras joins the same
+    way, dag_util.py:15-21)."""
+    from defer_amd.models import densenet121
+
+    torch.manual_seed(0)
+    m = densenet121(num_classes=10)
+    x = torch.randn(1, 64, 64, 3)
+    with torch.no_grad():
+        want = m(x)
+    assert want.shape == (1, 10)
+    assert torch.allclose(want.float().sum(dim=-1), torch.ones(1),
+                          atol=1e-4)
+    cuts = m.graph.valid_cut_points()
+    assert "dense2_6cat" in cuts and "trans1_pool" in cuts
+    auto_cuts, stages = auto_partition(m, 8, input_shape=(1, 64, 64, 3))
+    assert len(stages) == 8
+    assert any(c.startswith("dense") for c in auto_cuts)
+    with torch.no_grad():
+        z = x
+        for s in stages:
+            z = s(z)
+    assert torch.equal(z, want)
+
+
+def test_avgpool_and_concat_reference_ops():
+    from defer_amd import ops
+
+    x = torch.randn(2, 8, 8, 16)
+    y = ops.avgpool2d(x, 2, 2, 0)
+    assert y.shape == (2, 4, 4, 16)
+    import torch.nn.functional as F
+
+    want = F.avg_pool2d(x.permute(0, 3, 1, 2), 2, 2) \
+        .permute(0, 2, 3, 1)
+    assert torch.allclose(y, want, atol=1e-6)
+    a, b = torch.randn(2, 4, 4, 8), torch.randn(2, 4, 4, 24)
+    cat = ops.concat_channels([a, b])
+    assert torch.equal(cat, torch.cat([a, b], dim=-1))
