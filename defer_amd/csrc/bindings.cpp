@@ -347,6 +347,15 @@ Tensor concat_lastdim(std::vector<Tensor> xs) {
     base[nd - 1] = ctot;
     auto out = at::empty(base, xs[0].options());
     hipStream_t s = cur_stream();
+    if (xs.size() == 2 && xs[0].size(nd - 1) % 8 == 0 &&
+        xs[1].size(nd - 1) % 8 == 0) {
+        // the DenseNet shape class: one vectorized kernel beats the
+        // per-input small-pitch 2D copies (measured)
+        defer_hip::launch_cat2(bptr(xs[0]), bptr(xs[1]), bptr_mut(out),
+                               rows, (int)xs[0].size(nd - 1),
+                               (int)xs[1].size(nd - 1), s);
+        return out;
+    }
     long coff = 0;
     for (auto& t : xs) {
         long cj = t.size(nd - 1);

@@ -83,6 +83,31 @@ __global__ void softmax_kernel(const bf16* __restrict__ x,
         yr[c] = f2bf(__expf(bf2f(xr[c]) - m) * inv);
 }
 
+
+// Two-input channel concat, NHWC bf16 (DenseNet dense connections:
+// cat = [prev, feat]). One bf16x8 chunk per thread; consecutive
+// threads write consecutive output chunks, and each 8-chunk comes
+// whole from one source (c1 % 8 == 0), so reads stay coalesced per
+// segment — replaces per-input hipMemcpy2DAsync (small-pitch 2D
+// copies measured 10.8% of the DenseNet step).
+__global__ void cat2_kernel(const bf16* __restrict__ a,
+                            const bf16* __restrict__ b,
+                            bf16* __restrict__ y,
+                            long rows, int c1_8, int c2_8) {
+    const int ct8 = c1_8 + c2_8;
+    long total = rows * ct8;
+    long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long gstride = (long)gridDim.x * blockDim.x;
+    for (long i = i0; i < total; i += gstride) {
+        long r = i / ct8;
+        int c8 = (int)(i % ct8);
+        bf16x8 v = (c8 < c1_8)
+            ? load_bf16x8(a + (r * c1_8 + c8) * 8)
+            : load_bf16x8(b + (r * c2_8 + (c8 - c1_8)) * 8);
+        store_bf16x8(y + i * 8, v);
+    }
+}
+
 namespace defer_hip {
 
 static int grid1d(long work, int block) {
@@ -107,6 +132,14 @@ void launch_add_act(const void* a, const void* b, void* y, long total8,
 void launch_relu(const void* x, void* y, long total8, hipStream_t s) {
     hipLaunchKernelGGL(relu_kernel, dim3(grid1d(total8, 256)), dim3(256), 0,
                        s, (const bf16*)x, (bf16*)y, total8);
+}
+
+void launch_cat2(const void* a, const void* b, void* y, long rows,
+                 int c1, int c2, hipStream_t s) {
+    long total = rows * ((c1 + c2) / 8);
+    hipLaunchKernelGGL(cat2_kernel, dim3(grid1d(total, 256)), dim3(256),
+                       0, s, (const bf16*)a, (const bf16*)b, (bf16*)y,
+                       rows, c1 / 8, c2 / 8);
 }
 
 void launch_softmax(const void* x, void* y, int rows, int cols,

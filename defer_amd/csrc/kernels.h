@@ -48,6 +48,9 @@ void launch_bn_act(const void* x, const float* scale, const float* bias,
 void launch_add_act(const void* a, const void* b, void* y, long total8,
                     bool relu, hipStream_t s);
 void launch_relu(const void* x, void* y, long total8, hipStream_t s);
+// two-input NHWC channel concat (c1, c2 % 8 == 0)
+void launch_cat2(const void* a, const void* b, void* y, long rows,
+                 int c1, int c2, hipStream_t s);
 void launch_softmax(const void* x, void* y, int rows, int cols,
                     hipStream_t s);
 void launch_maxpool(const void* x, void* y, int NB, int H, int W, int C,
