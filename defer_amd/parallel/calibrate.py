@@ -15,8 +15,10 @@ Artifact format (defer_amd/calib/{model}.json or a user path):
      "us_per_image": {node_name: float, ...}}
 """
 
+import copy
 import json
 import os
+import time
 from typing import Dict, Optional
 
 import torch
@@ -78,8 +80,8 @@ def measure_layer_times(gm, input_shape=(64, 224, 224, 3),
         ww = (torch.randn(64, 3, 3, 64) * 0.05).to(dev, dtype)
         from defer_amd import ops as _ops
 
-        t0 = __import__("time").perf_counter()
-        while __import__("time").perf_counter() - t0 < 0.6:
+        t0 = time.perf_counter()
+        while time.perf_counter() - t0 < 0.6:
             for _ in range(50):
                 _ops.conv2d_bn_act(xw, ww, None, None, stride=1,
                                    padding=1)
@@ -97,8 +99,6 @@ def measure_layer_times(gm, input_shape=(64, 224, 224, 3),
 
     times_us: Dict[str, float] = {}
     for n in graph.nodes:
-        import copy
-
         layer = copy.deepcopy(n.layer).to(dev)
         if cuda:
             for p in layer.parameters():
@@ -124,8 +124,6 @@ def measure_layer_times(gm, input_shape=(64, 224, 224, 3),
                 ev1.synchronize()
                 ms = ev0.elapsed_time(ev1)
             else:
-                import time
-
                 t0 = time.perf_counter()
                 for _ in range(it):
                     layer(*ins, **n.kwargs)
