@@ -44,9 +44,10 @@ def main(argv):
         print("no bench JSON lines found", file=sys.stderr)
         return 1
     base = runs[0]
-    print(f"# {base['metric']}  ({base['config']['model']}, "
-          f"batch {base['config']['global_batch']}, {base['dtype']}, "
-          f"{base['data']})")
+    cfg = base.get("config", {})
+    print(f"# {base['metric']}  ({cfg.get('model', '?')}, "
+          f"batch {cfg.get('global_batch', '?')}, "
+          f"{base.get('dtype', '?')}, {base.get('data', '?')})")
     print(f"{'gpus':>4} {'images/sec':>12} {'ms/item':>9} "
           f"{'speedup':>8} {'efficiency':>10}  parallelism")
     for r in runs:
@@ -55,7 +56,7 @@ def main(argv):
         eff = sp / n * 100
         print(f"{n:>4} {r['value']:>12.1f} {r['ms_per_step']:>9.3f} "
               f"{sp:>7.2f}x {eff:>9.1f}%  "
-              f"{r['config'].get('parallelism', '?')}")
+              f"{r.get('config', {}).get('parallelism', '?')}")
     ref = "+53% at 8 nodes (reference, README.md:12)"
     if runs[-1]["n_gpus"] > 1:
         gain = (runs[-1]["value"] / base["value"] - 1) * 100
