@@ -60,8 +60,11 @@ class PipelineConfig:
     # bandwidth for the early fat boundaries that exceed one 153 GB/s
     # link (ResNet50 layer1: 1.6 MB/img = 10.5 us/hop, the 8-stage
     # bottleneck — profiles/README.md "Predicted pipeline scaling").
-    # Off by default pending multi-GPU validation; requires world > 2
-    # and a fixed-size wire (compression "none" or "zfp").
+    # Requires world > 2 and a fixed-size wire (not "zfp+lz4").
+    # bench.py defaults it ON at world >= 4 (with a warmup stall
+    # watchdog that re-execs single-rail, since the first RCCL
+    # execution is the round-end multi-GPU run itself); the raw
+    # PipelineConfig default stays False.
     dual_rail: bool = False
 
     # Directory of per-stage checkpoints written by
