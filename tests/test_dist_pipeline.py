@@ -644,3 +644,30 @@ def test_dist_pipeline_densenet_world3():
     assert len(errs) == steps
     for e in errs:
         assert e < 1e-5, e
+
+
+def test_choose_hop_modes_properties():
+    """Property sweep of the auto-wire rule: monotone in hop bytes,
+    fp8 never chosen when every hop is under the bottleneck, boost
+    only ever demotes fp8 -> none."""
+    from hypothesis import given, settings, strategies as st
+
+    from defer_amd.parallel.comm import choose_hop_modes
+
+    @settings(max_examples=60, deadline=None)
+    @given(st.lists(st.floats(0.5, 50.0), min_size=2, max_size=9),
+           st.floats(10.0, 400.0))
+    def check(stage_us, link):
+        hop_bytes = [1e5 * (i + 1) for i in range(len(stage_us) - 1)]
+        modes = choose_hop_modes(stage_us, hop_bytes, link)
+        boosted = choose_hop_modes(stage_us, hop_bytes, link,
+                                   [2.0] * len(hop_bytes))
+        bott = max(stage_us)
+        for i, m in enumerate(modes):
+            expect = "fp8" if hop_bytes[i] / (link * 1e3) > bott \
+                else "none"
+            assert m == expect
+            if m == "none":
+                assert boosted[i] == "none"   # boost never adds fp8
+
+    check()
