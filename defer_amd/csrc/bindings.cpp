@@ -255,6 +255,32 @@ Tensor linear(Tensor x, Tensor w, c10::optional<Tensor> bias) {
     return out;
 }
 
+Tensor conv1x1_prebn(Tensor x, Tensor w, Tensor scale, Tensor bias) {
+    // x [N,H,W,Cin] bf16, w [Cout,1,1,Cin] bf16, scale/bias fp32[Cin]:
+    // out = relu(x*scale+bias) @ w — one pass over x instead of the
+    // bn_act tensor round-trip (DenseNet's dominant cost)
+    check_bf16(x, "x");
+    check_bf16(w, "w");
+    TORCH_CHECK(x.dim() == 4 && w.dim() == 4 && w.size(1) == 1
+                && w.size(2) == 1, "conv1x1_prebn wants NHWC x, 1x1 w");
+    int NB = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
+    int Cout = w.size(0);
+    TORCH_CHECK(w.size(3) == Cin && Cin % 8 == 0, "Cin mismatch");
+    TORCH_CHECK(scale.scalar_type() == at::kFloat && scale.is_contiguous()
+                && scale.numel() == Cin, "scale must be fp32[Cin]");
+    TORCH_CHECK(bias.scalar_type() == at::kFloat && bias.is_contiguous()
+                && bias.numel() == Cin, "bias must be fp32[Cin]");
+    long M = (long)NB * H * W;
+    TORCH_CHECK(M < (1LL << 31), "M too large");
+    auto out = at::empty({NB, H, W, Cout}, x.options());
+    defer_hip::launch_gemm_prebn(bptr(x), bptr(w),
+                                 scale.data_ptr<float>(),
+                                 bias.data_ptr<float>(), zero_buf(),
+                                 bptr_mut(out), (int)M, Cin, Cout,
+                                 cur_stream());
+    return out;
+}
+
 Tensor bn_act(Tensor x, Tensor scale, Tensor bias, bool relu) {
     check_bf16(x, "x");
     int C = x.size(-1);
@@ -509,6 +535,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("stride"), py::arg("pad"), py::arg("relu"));
     m.def("linear", &linear);
     m.def("bn_act", &bn_act);
+    m.def("conv1x1_prebn", &conv1x1_prebn);
     m.def("add_act", &add_act);
     m.def("relu", &relu);
     m.def("softmax", &softmax);

@@ -115,6 +115,12 @@ class BNActConv(nn.Module):
         self.bias = nn.Parameter(torch.zeros(cin))
 
     def forward(self, x):
+        if self.kernel == 1 and self.stride == 1 and self.padding == 0:
+            # fused single-pass path (the bn_act round-trip over the
+            # concat inputs was 57% of the DenseNet step)
+            return ops.conv1x1_prebn(
+                x, self.weight.to(x.dtype),
+                self.scale.float(), self.bias.float())
         z = ops.batchnorm_apply(x, self.scale, self.bias, act="relu")
         return ops.conv2d_bn_act(z, self.weight.to(x.dtype), None, None,
                                  stride=self.stride,

@@ -61,6 +61,17 @@ def conv2d_bn_act(x, w, scale=None, bias=None, stride=1, padding=1,
                            act == "relu")
 
 
+def conv1x1_prebn(x, w, scale, bias):
+    """Fused relu(x*scale+bias) @ w for 1x1/s1/p0 convs (DenseNet
+    BNActConv): saves the standalone bn_act tensor round-trip on GPU;
+    CPU composes the reference ops."""
+    m = _backend(x)
+    if m is None:
+        z = _ref.batchnorm_apply(x, scale, bias, "relu")
+        return _ref.conv2d_bn_act(z, w, None, None, 1, 0, "none", None)
+    return m.conv1x1_prebn(x, w, scale, bias)
+
+
 def batchnorm_apply(x, scale, bias, act="none"):
     m = _backend(x)
     if m is None:

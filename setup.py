@@ -19,7 +19,7 @@ ROOT = Path(__file__).parent
 CSRC = ROOT / "defer_amd" / "csrc"
 
 sources = [str(CSRC / f) for f in
-           ["bindings.cpp", "conv.hip", "elementwise.hip", "pool.hip",
+           ["bindings.cpp", "conv.hip", "prebn.hip", "elementwise.hip", "pool.hip",
             "codec.hip", "lz4.hip"]
            if (CSRC / f).exists()]
 

@@ -424,3 +424,27 @@ def test_densenet121_full_forward_vs_cpu():
         got = ex.run(x.to(DEV, torch.bfloat16)).float().cpu()
     cos = torch.nn.functional.cosine_similarity(got, want, dim=-1)
     assert (cos > 0.98).all(), f"cosine {cos}"
+
+
+@requires_gpu
+def test_conv1x1_prebn_bitwise_vs_two_step():
+    """The fused pre-activation 1x1 (gemm_prebn_kernel) must be
+    BIT-identical to bn_act -> conv2d_bn_act: same fp32 bn + bf16
+    rounding before the MFMA, same ascending-k 16x16x32 accumulation."""
+    torch.manual_seed(4)
+    for N, H, W, Cin, Cout in [(2, 28, 28, 96, 128),    # K % 64 != 0
+                               (2, 14, 14, 256, 128),
+                               (1, 7, 7, 1024, 512),
+                               (3, 8, 8, 64, 40)]:      # Cout % 64 != 0
+        x = torch.randn(N, H, W, Cin).to(DEV, torch.bfloat16)
+        w = (torch.randn(Cout, 1, 1, Cin) * 0.05).to(DEV, torch.bfloat16)
+        sc = (torch.rand(Cin) + 0.5).to(DEV)
+        bi = (torch.randn(Cin) * 0.1).to(DEV)
+        z = ops.batchnorm_apply(x, sc, bi, act="relu")
+        want = ops.conv2d_bn_act(z, w, None, None, stride=1, padding=0,
+                                 act="none")
+        got = ops.conv1x1_prebn(x, w, sc, bi)
+        assert got.shape == want.shape
+        assert torch.equal(got, want), (
+            N, H, W, Cin, Cout,
+            (got.float() - want.float()).abs().max().item())
