@@ -255,7 +255,9 @@ Tensor linear(Tensor x, Tensor w, c10::optional<Tensor> bias) {
     return out;
 }
 
-Tensor conv1x1_prebn(Tensor x, Tensor w, Tensor scale, Tensor bias) {
+Tensor conv1x1_prebn(Tensor x, Tensor w, Tensor scale, Tensor bias,
+                     c10::optional<Tensor> out_scale,
+                     c10::optional<Tensor> out_bias) {
     // x [N,H,W,Cin] bf16, w [Cout,1,1,Cin] bf16, scale/bias fp32[Cin]:
     // out = relu(x*scale+bias) @ w — one pass over x instead of the
     // bn_act tensor round-trip (DenseNet's dominant cost)
@@ -273,11 +275,25 @@ Tensor conv1x1_prebn(Tensor x, Tensor w, Tensor scale, Tensor bias) {
     long M = (long)NB * H * W;
     TORCH_CHECK(M < (1LL << 31), "M too large");
     auto out = at::empty({NB, H, W, Cout}, x.options());
+    const float* os = nullptr;
+    const float* ob = nullptr;
+    if (out_scale) {
+        TORCH_CHECK(out_bias, "out_scale needs out_bias");
+        TORCH_CHECK(out_scale->scalar_type() == at::kFloat
+                    && out_scale->is_contiguous()
+                    && out_scale->numel() == Cout
+                    && out_bias->scalar_type() == at::kFloat
+                    && out_bias->is_contiguous()
+                    && out_bias->numel() == Cout,
+                    "out affine must be fp32[Cout]");
+        os = out_scale->data_ptr<float>();
+        ob = out_bias->data_ptr<float>();
+    }
     defer_hip::launch_gemm_prebn(bptr(x), bptr(w),
                                  scale.data_ptr<float>(),
-                                 bias.data_ptr<float>(), zero_buf(),
-                                 bptr_mut(out), (int)M, Cin, Cout,
-                                 cur_stream());
+                                 bias.data_ptr<float>(), os, ob,
+                                 zero_buf(), bptr_mut(out), (int)M, Cin,
+                                 Cout, cur_stream());
     return out;
 }
 
@@ -535,7 +551,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("stride"), py::arg("pad"), py::arg("relu"));
     m.def("linear", &linear);
     m.def("bn_act", &bn_act);
-    m.def("conv1x1_prebn", &conv1x1_prebn);
+    m.def("conv1x1_prebn", &conv1x1_prebn, py::arg("x"),
+          py::arg("w"), py::arg("scale"), py::arg("bias"),
+          py::arg("out_scale") = py::none(),
+          py::arg("out_bias") = py::none());
     m.def("add_act", &add_act);
     m.def("relu", &relu);
     m.def("softmax", &softmax);

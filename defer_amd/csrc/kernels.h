@@ -46,8 +46,9 @@ void launch_stem_repack_w(const void* w, void* wp, int Cout, int R, int S,
 // fused pre-activation 1x1 conv: out = relu(x*ps+pb) @ w^T
 // (DenseNet BNActConv; bit-identical to bn_act -> conv GEMM mode)
 void launch_gemm_prebn(const void* x, const void* w, const float* ps,
-                       const float* pb, const void* zbuf, void* out,
-                       int M, int K, int Cout, hipStream_t s);
+                       const float* pb, const float* os, const float* ob,
+                       const void* zbuf, void* out, int M, int K,
+                       int Cout, hipStream_t s);
 
 void launch_bn_act(const void* x, const float* scale, const float* bias,
                    void* y, long total8, int c8, bool relu, hipStream_t s);

@@ -32,12 +32,20 @@ __device__ __forceinline__ void pb_glds16(const bf16* src,
         16, 0, 0);
 }
 
-template <int BM, int BN>
+// OUT_AFF: apply a per-OUTPUT-channel affine + ReLU on the fp32
+// accumulator before rounding (the NEXT layer's folded BN-ReLU — in a
+// DenseNet dense layer, c1's output feeds only norm2/relu2/conv2, so
+// norm2 folds here and conv2 becomes a plain 3x3). Applied on fp32 acc
+// (no intermediate bf16 round), so it is slightly MORE accurate than
+// the two-step path, not bit-identical to it.
+template <int BM, int BN, bool OUT_AFF>
 __global__ __launch_bounds__(PB_TPB, 4) void gemm_prebn_kernel(
     const bf16* __restrict__ X,      // [M][K]
     const bf16* __restrict__ Wt,     // [Cout][K]
     const float* __restrict__ PS,    // [K] folded BN scale
     const float* __restrict__ PBb,   // [K] folded BN bias
+    const float* __restrict__ OS,    // [Cout] out scale (OUT_AFF)
+    const float* __restrict__ OB,    // [Cout] out bias (OUT_AFF)
     const bf16* __restrict__ Z,      // >=16B zeros
     bf16* __restrict__ OUT,          // [M][Cout]
     int M, int K, int Cout) {
@@ -162,6 +170,15 @@ __global__ __launch_bounds__(PB_TPB, 4) void gemm_prebn_kernel(
         }
         // ---- store (direct; the outputs here are small vs the fused
         // input pass this kernel saves)
+        float osc[NI], obi[NI];
+        if (OUT_AFF)
+#pragma unroll
+            for (int ni = 0; ni < NI; ++ni) {
+                int n = n0 + wn * WN + ni * 16 + lo16;
+                if (n >= Cout) n = Cout - 1;
+                osc[ni] = OS[n];
+                obi[ni] = OB[n];
+            }
 #pragma unroll
         for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
@@ -170,8 +187,11 @@ __global__ __launch_bounds__(PB_TPB, 4) void gemm_prebn_kernel(
                 for (int e = 0; e < 4; ++e) {
                     int m = m0 + wm * 32 + mi * 16 + hi4 * 4 + e;
                     int n = n0 + wn * WN + ni * 16 + lo16;
+                    float v = acc[mi][ni][e];
+                    if (OUT_AFF)
+                        v = fmaxf(fmaf(v, osc[ni], obi[ni]), 0.f);
                     if (m < M && n < Cout)
-                        OUT[(long)m * Cout + n] = f2bf(acc[mi][ni][e]);
+                        OUT[(long)m * Cout + n] = f2bf(v);
                 }
     }
 }
@@ -179,16 +199,25 @@ __global__ __launch_bounds__(PB_TPB, 4) void gemm_prebn_kernel(
 namespace defer_hip {
 
 void launch_gemm_prebn(const void* x, const void* w, const float* ps,
-                       const float* pb, const void* zbuf, void* out,
-                       int M, int K, int Cout, hipStream_t s) {
+                       const float* pb, const float* os, const float* ob,
+                       const void* zbuf, void* out, int M, int K,
+                       int Cout, hipStream_t s) {
     constexpr int BM = 64, BN = 64;
     int mtiles = (M + BM - 1) / BM;
     int ntiles = (Cout + BN - 1) / BN;
     int gx = mtiles < 4096 ? mtiles : 4096;
-    hipLaunchKernelGGL((gemm_prebn_kernel<BM, BN>), dim3(gx, ntiles),
-                       dim3(PB_TPB), 0, s, (const bf16*)x,
-                       (const bf16*)w, ps, pb, (const bf16*)zbuf,
-                       (bf16*)out, M, K, Cout);
+    if (os != nullptr)
+        hipLaunchKernelGGL((gemm_prebn_kernel<BM, BN, true>),
+                           dim3(gx, ntiles), dim3(PB_TPB), 0, s,
+                           (const bf16*)x, (const bf16*)w, ps, pb, os,
+                           ob, (const bf16*)zbuf, (bf16*)out, M, K,
+                           Cout);
+    else
+        hipLaunchKernelGGL((gemm_prebn_kernel<BM, BN, false>),
+                           dim3(gx, ntiles), dim3(PB_TPB), 0, s,
+                           (const bf16*)x, (const bf16*)w, ps, pb, os,
+                           ob, (const bf16*)zbuf, (bf16*)out, M, K,
+                           Cout);
 }
 
 }  // namespace defer_hip

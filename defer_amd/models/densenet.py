@@ -41,10 +41,16 @@ def densenet121(num_classes: int = 1000) -> GraphModel:
 
     for b, nlayers in enumerate(_BLOCKS, start=1):
         for j in range(1, nlayers + 1):
+            # c1 owns BOTH folded BNs: norm1 on its input staging and
+            # norm2 on its fp32 output accumulator (gemm_prebn_kernel),
+            # so conv2 runs as a plain 3x3 — no standalone bn_act pass
+            # anywhere in the dense layer
             f = N(f"dense{b}_{j}c1",
-                  BNActConv(c, _BN_SIZE * _GROWTH, 1, 1, 0), [x])
+                  BNActConv(c, _BN_SIZE * _GROWTH, 1, 1, 0,
+                            out_bn=True), [x])
             f = N(f"dense{b}_{j}c2",
-                  BNActConv(_BN_SIZE * _GROWTH, _GROWTH, 3, 1, 1), [f])
+                  ConvBNAct(_BN_SIZE * _GROWTH, _GROWTH, 3, 1, 1,
+                            act="none", bn=False), [f])
             x = N(f"dense{b}_{j}cat", Concat(), [x, f])
             c += _GROWTH
         if b < len(_BLOCKS):

@@ -103,7 +103,8 @@ class BNActConv(nn.Module):
     folded per-channel scale/bias applied by the standalone bn_act HIP
     kernel; the conv runs without epilogue BN."""
 
-    def __init__(self, cin, cout, kernel=1, stride=1, padding=0):
+    def __init__(self, cin, cout, kernel=1, stride=1, padding=0,
+                 out_bn=False):
         super().__init__()
         self.cin, self.cout = cin, cout
         self.kernel, self.stride, self.padding = kernel, stride, padding
@@ -113,18 +114,31 @@ class BNActConv(nn.Module):
         self.weight = nn.Parameter(w)
         self.scale = nn.Parameter(torch.ones(cin))
         self.bias = nn.Parameter(torch.zeros(cin))
+        # out_bn: this layer ALSO owns the next composite's folded
+        # BN-ReLU, applied to the conv output (DenseNet norm2 folds
+        # into c1's epilogue; conv2 then runs as a plain 3x3)
+        self.out_bn = out_bn
+        if out_bn:
+            self.out_scale = nn.Parameter(torch.ones(cout))
+            self.out_bias = nn.Parameter(torch.zeros(cout))
 
     def forward(self, x):
         if self.kernel == 1 and self.stride == 1 and self.padding == 0:
-            # fused single-pass path (the bn_act round-trip over the
-            # concat inputs was 57% of the DenseNet step)
+            # fused single-pass path (the standalone bn_act round-trips
+            # were 57% of the DenseNet step before fusion)
             return ops.conv1x1_prebn(
                 x, self.weight.to(x.dtype),
-                self.scale.float(), self.bias.float())
+                self.scale.float(), self.bias.float(),
+                self.out_scale.float() if self.out_bn else None,
+                self.out_bias.float() if self.out_bn else None)
         z = ops.batchnorm_apply(x, self.scale, self.bias, act="relu")
-        return ops.conv2d_bn_act(z, self.weight.to(x.dtype), None, None,
-                                 stride=self.stride,
-                                 padding=self.padding, act="none")
+        y = ops.conv2d_bn_act(z, self.weight.to(x.dtype), None, None,
+                              stride=self.stride,
+                              padding=self.padding, act="none")
+        if self.out_bn:
+            y = ops.batchnorm_apply(y, self.out_scale, self.out_bias,
+                                    act="relu")
+        return y
 
     def extra_repr(self):
         return (f"bn-relu-conv {self.cin}->{self.cout} k{self.kernel} "

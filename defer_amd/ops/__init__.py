@@ -61,15 +61,20 @@ def conv2d_bn_act(x, w, scale=None, bias=None, stride=1, padding=1,
                            act == "relu")
 
 
-def conv1x1_prebn(x, w, scale, bias):
+def conv1x1_prebn(x, w, scale, bias, out_scale=None, out_bias=None):
     """Fused relu(x*scale+bias) @ w for 1x1/s1/p0 convs (DenseNet
-    BNActConv): saves the standalone bn_act tensor round-trip on GPU;
-    CPU composes the reference ops."""
+    BNActConv), optionally followed by the NEXT layer's folded BN-ReLU
+    on the output accumulator (out_scale/out_bias): saves the
+    standalone bn_act round-trips on GPU; CPU composes the reference
+    ops."""
     m = _backend(x)
     if m is None:
         z = _ref.batchnorm_apply(x, scale, bias, "relu")
-        return _ref.conv2d_bn_act(z, w, None, None, 1, 0, "none", None)
-    return m.conv1x1_prebn(x, w, scale, bias)
+        y = _ref.conv2d_bn_act(z, w, None, None, 1, 0, "none", None)
+        if out_scale is not None:
+            y = _ref.batchnorm_apply(y, out_scale, out_bias, "relu")
+        return y
+    return m.conv1x1_prebn(x, w, scale, bias, out_scale, out_bias)
 
 
 def batchnorm_apply(x, scale, bias, act="none"):

@@ -448,3 +448,24 @@ def test_conv1x1_prebn_bitwise_vs_two_step():
         assert torch.equal(got, want), (
             N, H, W, Cin, Cout,
             (got.float() - want.float()).abs().max().item())
+
+
+@requires_gpu
+def test_conv1x1_prebn_out_affine():
+    """The output-affine variant (next layer's folded BN-ReLU on the
+    fp32 accumulator): compare against the two-step composition with a
+    tolerance (the fused form skips the intermediate bf16 rounding, so
+    it is slightly MORE accurate, not bit-equal)."""
+    torch.manual_seed(6)
+    N, H, W, Cin, Cout = 2, 14, 14, 160, 128
+    x = torch.randn(N, H, W, Cin).to(DEV, torch.bfloat16)
+    w = (torch.randn(Cout, 1, 1, Cin) * 0.05).to(DEV, torch.bfloat16)
+    sc = (torch.rand(Cin) + 0.5).to(DEV)
+    bi = (torch.randn(Cin) * 0.1).to(DEV)
+    osc = (torch.rand(Cout) + 0.5).to(DEV)
+    obi = (torch.randn(Cout) * 0.1).to(DEV)
+    y1 = ops.conv1x1_prebn(x, w, sc, bi)
+    want = ops.batchnorm_apply(y1, osc, obi, act="relu")
+    got = ops.conv1x1_prebn(x, w, sc, bi, osc, obi)
+    assert got.shape == want.shape
+    assert _relerr(got, want) < 0.005
