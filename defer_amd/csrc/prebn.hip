@@ -58,10 +58,9 @@ __global__ __launch_bounds__(PB_TPB, 4) void gemm_prebn_kernel(
     constexpr int ACH = BM * 8 / PB_TPB;
     constexpr int BCH = BN * 8 / PB_TPB;
 
-    __shared__ __attribute__((aligned(16)))
-    bf16 lds[2 * (BM + BN) * PB_BK];
-    bf16* A0 = lds;                      // [2][BM][BK] ping-pong
-    bf16* B0 = lds + 2 * BM * PB_BK;     // [2][BN][BK] ping-pong
+    __shared__ __attribute__((aligned(16))) bf16 lds[(BM + BN) * PB_BK];
+    bf16* A0 = lds;
+    bf16* B0 = lds + BM * PB_BK;
 
     const int tid = threadIdx.x;
     const int wave = tid / WAVE;
@@ -98,17 +97,8 @@ __global__ __launch_bounds__(PB_TPB, 4) void gemm_prebn_kernel(
             for (int ni = 0; ni < NI; ++ni)
                 acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
 
-        // software pipeline: A loads (registers) and B glds for tile
-        // kt+1 are issued before compute(kt), so their latency hides
-        // under the MFMA phase; LDS is double-buffered (A and B ping-
-        // pong), with a counted vmcnt leaving exactly the next tile's
-        // ops in flight. The A transform for tile kt consumes registers
-        // loaded one tile earlier.
-        bf16x8 areg[2][ACH];
-        f32x4 sreg[2][ACH][4];   // [s0 s1 c0 c1] per chunk
-        bool aval[2][ACH];
-
-        auto issue_ab = [&](int kt, int ph) {
+        for (int kt = 0; kt < nk; ++kt) {
+            // ---- B: direct global -> LDS (weights, no transform)
 #pragma unroll
             for (int i = 0; i < BCH; ++i) {
                 int n = n0 + b_row[i];
@@ -116,52 +106,41 @@ __global__ __launch_bounds__(PB_TPB, 4) void gemm_prebn_kernel(
                 const bf16* src = (n < Cout && k < K)
                                       ? Wt + (long)n * K + k
                                       : Z;
-                pb_glds16(src, B0 + ph * BN * PB_BK
-                                   + (wave * (BCH * 64) + i * 64) * 8);
+                pb_glds16(src, B0 + (wave * (BCH * 64) + i * 64) * 8);
             }
+            // ---- A: registers, bn+relu in fp32, bf16 round, ds_write
+            // (same rounding as the standalone bn_act kernel -> the
+            // fused path is bit-identical to the two-step one)
 #pragma unroll
             for (int i = 0; i < ACH; ++i) {
                 int m = m0 + a_row[i];
                 int k = kt * PB_BK + a_k8[i] * 8;
-                bool v = (m < M && k < K);
-                aval[ph][i] = v;
-                if (v) {
-                    areg[ph][i] = load_bf16x8(X + (long)m * K + k);
-                    sreg[ph][i][0] =
-                        *reinterpret_cast<const f32x4*>(PS + k);
-                    sreg[ph][i][1] =
-                        *reinterpret_cast<const f32x4*>(PS + k + 4);
-                    sreg[ph][i][2] =
-                        *reinterpret_cast<const f32x4*>(PBb + k);
-                    sreg[ph][i][3] =
-                        *reinterpret_cast<const f32x4*>(PBb + k + 4);
-                }
-            }
-        };
-        auto transform_a = [&](int ph) {
-#pragma unroll
-            for (int i = 0; i < ACH; ++i) {
                 bf16x8 o;
-                if (aval[ph][i]) {
+                if (m < M && k < K) {
+                    bf16x8 v = load_bf16x8(X + (long)m * K + k);
+                    f32x4 s0 = *reinterpret_cast<const f32x4*>(PS + k);
+                    f32x4 s1 = *reinterpret_cast<const f32x4*>(PS + k
+                                                               + 4);
+                    f32x4 c0 = *reinterpret_cast<const f32x4*>(PBb + k);
+                    f32x4 c1 = *reinterpret_cast<const f32x4*>(PBb + k
+                                                               + 4);
 #pragma unroll
                     for (int j = 0; j < 8; ++j) {
-                        float sj = j < 4 ? sreg[ph][i][0][j]
-                                         : sreg[ph][i][1][j - 4];
-                        float cj = j < 4 ? sreg[ph][i][2][j]
-                                         : sreg[ph][i][3][j - 4];
-                        o[j] = f2bf(fmaxf(
-                            fmaf(bf2f(areg[ph][i][j]), sj, cj), 0.f));
+                        float sj = j < 4 ? s0[j] : s1[j - 4];
+                        float cj = j < 4 ? c0[j] : c1[j - 4];
+                        o[j] = f2bf(fmaxf(fmaf(bf2f(v[j]), sj, cj),
+                                          0.f));
                     }
                 } else {
 #pragma unroll
                     for (int j = 0; j < 8; ++j) o[j] = f2bf(0.f);
                 }
-                store_bf16x8(A0 + ph * BM * PB_BK
-                                + (wave * (ACH * 64) + i * 64) * 8
-                                + lane * 8, o);
+                store_bf16x8(A0 + (wave * (ACH * 64) + i * 64) * 8
+                                 + lane * 8, o);
             }
-        };
-        auto compute = [&](int ph) {
+            asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+            // ---- compute (same fragment mapping + k order as igemm)
 #pragma unroll
             for (int ks = 0; ks < PB_BK / 32; ++ks) {
                 bf16x8 af[2], bfr[NI];
@@ -169,14 +148,14 @@ __global__ __launch_bounds__(PB_TPB, 4) void gemm_prebn_kernel(
                 for (int mi = 0; mi < 2; ++mi) {
                     int row = wm * 32 + mi * 16 + lo16;
                     af[mi] = *reinterpret_cast<bf16x8*>(
-                        A0 + ph * BM * PB_BK + row * PB_BK
+                        A0 + row * PB_BK
                            + pb_swz(row, ks * 4 + hi4) * 8);
                 }
 #pragma unroll
                 for (int ni = 0; ni < NI; ++ni) {
                     int row = wn * WN + ni * 16 + lo16;
                     bfr[ni] = *reinterpret_cast<bf16x8*>(
-                        B0 + ph * BN * PB_BK + row * PB_BK
+                        B0 + row * PB_BK
                            + pb_swz(row, ks * 4 + hi4) * 8);
                 }
 #pragma unroll
@@ -186,26 +165,8 @@ __global__ __launch_bounds__(PB_TPB, 4) void gemm_prebn_kernel(
                         acc[mi][ni] = MFMA_BF16_16x16x32(af[mi], bfr[ni],
                                                          acc[mi][ni]);
             }
-        };
-
-        issue_ab(0, 0);
-        for (int kt = 0; kt < nk; ++kt) {
-            const int ph = kt & 1;
-            // transform consumes registers loaded one iteration ago
-            // (the compiler's data deps wait exactly those loads; the
-            // previous compute phase covered their latency)
-            transform_a(ph);
-            // drain: tile kt's B glds were issued one iteration ago
-            // and have had a whole compute phase to land, so this is
-            // cheap — and it needs no fragile count of the compiler's
-            // A-load instructions
-            asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
-            __builtin_amdgcn_s_barrier();   // A/B of tile kt visible
-            // issue tile kt+1 NOW so its loads fly under compute(kt)
-            // (B glds target buffer ph^1, untouched until kt+1)
-            if (kt + 1 < nk) issue_ab(kt + 1, ph ^ 1);
-            compute(ph);
-            __builtin_amdgcn_s_barrier();   // buffer ph reused at kt+2
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();   // LDS reused next k-tile
         }
         // ---- store (direct; the outputs here are small vs the fused
         // input pass this kernel saves)
